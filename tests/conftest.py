@@ -1,0 +1,43 @@
+import os
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO_ROOT = Path(__file__).resolve().parent.parent
+if str(REPO_ROOT) not in sys.path:
+    sys.path.insert(0, str(REPO_ROOT))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test needs an MI355X GPU (run with -m gpu on a GPU box)"
+    )
+
+
+def pytest_collection_modifyitems(config, items):
+    """Skip gpu-marked tests automatically when no GPU is present."""
+    try:
+        import torch
+
+        has_gpu = torch.cuda.is_available()
+    except Exception:
+        has_gpu = False
+    if has_gpu:
+        return
+    skip = pytest.mark.skip(reason="no GPU available")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+@pytest.fixture()
+def tmp_config(tmp_path, monkeypatch):
+    """Point the config system at a throwaway file and clear its cache."""
+    from comfyui_distributed_amd.utils import config as config_mod
+
+    path = tmp_path / "gpu_config.json"
+    monkeypatch.setenv("DISTGPU_CONFIG", str(path))
+    config_mod._cache.update(path=None, mtime=None, data=None)
+    yield path
+    config_mod._cache.update(path=None, mtime=None, data=None)
