@@ -1,0 +1,237 @@
+"""Op dispatch: HIP kernels on GPU, reference Python implementations on CPU.
+
+On a GPU host the HIP extension is mandatory — a missing extension raises
+``KernelUnavailableError`` instead of silently running eager PyTorch. The
+CPU implementations exist for CPU-only test runs and for the reference
+numerics the GPU tests compare against; they implement *the same math* as
+the kernels (documented per-op).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from . import ext
+
+LANCZOS_A = 3
+ATTN_DPADS = (64, 96, 128, 160)
+
+
+def hip_available() -> bool:
+    return ext.get_ext(required=False) is not None
+
+
+def _on_gpu(*tensors: torch.Tensor) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# ---------------------------------------------------------------------------
+# norms / elementwise
+# ---------------------------------------------------------------------------
+
+
+def group_norm_silu(
+    x: torch.Tensor,
+    groups: int,
+    weight: torch.Tensor,
+    bias: torch.Tensor,
+    eps: float = 1e-6,
+    silu: bool = True,
+) -> torch.Tensor:
+    if _on_gpu(x):
+        return ext.get_ext(True).group_norm_fused(
+            x.to(torch.bfloat16), groups, weight, bias, eps, silu
+        )
+    y = F.group_norm(x.float(), groups, weight.float(), bias.float(), eps)
+    if silu:
+        y = F.silu(y)
+    return y.to(x.dtype)
+
+
+def layer_norm(
+    x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor, eps: float = 1e-5
+) -> torch.Tensor:
+    if _on_gpu(x):
+        return ext.get_ext(True).layer_norm(x.to(torch.bfloat16), weight, bias, eps)
+    y = F.layer_norm(x.float(), (x.shape[-1],), weight.float(), bias.float(), eps)
+    return y.to(x.dtype)
+
+
+def act_mul(a: torch.Tensor, b: torch.Tensor, gelu: bool = False) -> torch.Tensor:
+    """a * act(b): SiLU-mul / GEGLU gate."""
+    if _on_gpu(a):
+        return ext.get_ext(True).act_mul(a.to(torch.bfloat16), b.to(torch.bfloat16), gelu)
+    bf = b.float()
+    act = F.gelu(bf, approximate="tanh") if gelu else F.silu(bf)
+    return (a.float() * act).to(a.dtype)
+
+
+# ---------------------------------------------------------------------------
+# attention
+# ---------------------------------------------------------------------------
+
+
+def _dpad_for(d: int) -> int:
+    for p in ATTN_DPADS:
+        if d <= p:
+            return p
+    raise ValueError(f"head dim {d} unsupported (max {ATTN_DPADS[-1]})")
+
+
+def attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    heads: int,
+    kv_heads: int | None = None,
+    scale: float | None = None,
+) -> torch.Tensor:
+    """Multi-head attention over packed heads.
+
+    q: [B*H, Nq, D], k/v: [B*Hkv, Nk, D]; returns [B*H, Nq, D].
+    GPU path: the MFMA flash kernel (bf16, fp32 accumulation); dims are
+    zero-padded to the kernel contract (D -> D_PAD, Nq -> 64k, Nk -> 32k)
+    here, the kernel masks padded keys.
+    """
+    kv_heads = kv_heads or heads
+    d = q.shape[-1]
+    scale = scale if scale is not None else 1.0 / math.sqrt(d)
+    if _on_gpu(q):
+        dp = _dpad_for(d)
+        nq, nk = q.shape[1], k.shape[1]
+        nq_p = (nq + 63) // 64 * 64
+        nk_p = (nk + 31) // 32 * 32
+        qb = q.to(torch.bfloat16)
+        kb = k.to(torch.bfloat16)
+        vb = v.to(torch.bfloat16)
+        if dp != d or nq_p != nq:
+            qb = F.pad(qb, (0, dp - d, 0, nq_p - nq))
+        if dp != d or nk_p != nk:
+            kb = F.pad(kb, (0, dp - d, 0, nk_p - nk))
+            vb = F.pad(vb, (0, dp - d, 0, nk_p - nk))
+        o = ext.get_ext(True).attn_fwd(
+            qb.contiguous(), kb.contiguous(), vb.contiguous(), heads, kv_heads,
+            nk, scale,
+        )
+        return o[:, :nq, :d]
+    # CPU reference: fp32 math, GQA by repeating kv heads.
+    qf, kf, vf = q.float(), k.float(), v.float()
+    if kv_heads != heads:
+        rep = heads // kv_heads
+        b = q.shape[0] // heads
+        kf = kf.reshape(b, kv_heads, 1, *kf.shape[1:]).expand(-1, -1, rep, -1, -1)
+        kf = kf.reshape(b * heads, *k.shape[1:])
+        vf = vf.reshape(b, kv_heads, 1, *vf.shape[1:]).expand(-1, -1, rep, -1, -1)
+        vf = vf.reshape(b * heads, *v.shape[1:])
+    s = torch.bmm(qf, kf.transpose(1, 2)) * scale
+    p = torch.softmax(s, dim=-1)
+    return torch.bmm(p, vf).to(q.dtype)
+
+
+# ---------------------------------------------------------------------------
+# tile pipeline (Lanczos-3 resample + erf-mask blend)
+# ---------------------------------------------------------------------------
+
+
+def _lanczos3(x: torch.Tensor) -> torch.Tensor:
+    ax = x.abs()
+    out = torch.where(
+        ax < 1e-6,
+        torch.ones_like(x),
+        torch.sinc(x) * torch.sinc(x / LANCZOS_A),
+    )
+    return torch.where(ax >= LANCZOS_A, torch.zeros_like(x), out)
+
+
+def _lanczos_weight_matrix(
+    n_out: int, src_lo: float, scale: float, src_size: int
+) -> torch.Tensor:
+    """Dense [n_out, src_size] weight matrix matching the kernel's per-pixel
+    tap enumeration (inclusive [floor(c-s+.5), floor(c+s+.5)], max 16 taps,
+    edge clamp, renormalized)."""
+    centers = src_lo + (torch.arange(n_out, dtype=torch.float64) + 0.5) * scale - 0.5
+    fscale = max(scale, 1.0)
+    support = LANCZOS_A * fscale
+    x0 = torch.floor(centers - support + 0.5)
+    x1 = torch.floor(centers + support + 0.5)
+    max_taps = min(int((x1 - x0).max().item()) + 1, 16)
+    taps = x0[:, None] + torch.arange(max_taps, dtype=torch.float64)[None, :]
+    w = _lanczos3(((taps - centers[:, None]) / fscale).float()).double()
+    w = torch.where(taps <= x1[:, None], w, torch.zeros_like(w))
+    w = w / w.sum(dim=1, keepdim=True).clamp_min(1e-12)
+    idx = taps.long().clamp(0, src_size - 1)
+    dense = torch.zeros(n_out, src_size, dtype=torch.float64)
+    dense.scatter_add_(1, idx, w)
+    return dense.float()
+
+
+def _lanczos_resample_cpu(
+    src: torch.Tensor, x1: int, y1: int, x2: int, y2: int, ow: int, oh: int
+) -> torch.Tensor:
+    """[B,H,W,C] f32 -> [B,oh,ow,C]: separable Lanczos-3 of region."""
+    B, H, W, C = src.shape
+    wx = _lanczos_weight_matrix(ow, float(x1), (x2 - x1) / ow, W)  # [ow, W]
+    wy = _lanczos_weight_matrix(oh, float(y1), (y2 - y1) / oh, H)  # [oh, H]
+    t = src.permute(0, 3, 1, 2).reshape(B * C, H, W).float()
+    t = torch.einsum("oh,bhw->bow", wy, t)
+    t = torch.einsum("pw,bow->bop", wx, t)
+    return t.reshape(B, C, oh, ow).permute(0, 2, 3, 1).contiguous()
+
+
+def extract_resize(
+    src: torch.Tensor, region: tuple[int, int, int, int], ow: int, oh: int
+) -> torch.Tensor:
+    """Crop ``region`` of [B,H,W,C] f32 and Lanczos-3 resample to (ow, oh)."""
+    x1, y1, x2, y2 = (int(r) for r in region)
+    if _on_gpu(src):
+        return ext.get_ext(True).extract_resize(src.float(), x1, y1, x2, y2, ow, oh)
+    return _lanczos_resample_cpu(src.float(), x1, y1, x2, y2, ow, oh)
+
+
+def rect_mask_cpu(
+    h: int,
+    w: int,
+    rect: tuple[int, int, int, int],
+    sigma: float,
+    device=None,
+) -> torch.Tensor:
+    """Analytic Gaussian-blurred white-rect mask [h, w] (erf-product form)."""
+    rx1, ry1, rx2, ry2 = (float(r) for r in rect)
+    xs = torch.arange(w, dtype=torch.float32, device=device)
+    ys = torch.arange(h, dtype=torch.float32, device=device)
+    if sigma <= 0:
+        gx = ((xs >= rx1) & (xs < rx2)).float()
+        gy = ((ys >= ry1) & (ys < ry2)).float()
+    else:
+        inv_s = 1.0 / (sigma * math.sqrt(2.0))
+        gx = 0.5 * (torch.erf((xs + 0.5 - rx1) * inv_s) - torch.erf((xs + 0.5 - rx2) * inv_s))
+        gy = 0.5 * (torch.erf((ys + 0.5 - ry1) * inv_s) - torch.erf((ys + 0.5 - ry2) * inv_s))
+    return gy[:, None] * gx[None, :]
+
+
+def blend_tile(
+    canvas: torch.Tensor,
+    tile: torch.Tensor,
+    region: tuple[int, int, int, int],
+    mask_rect: tuple[int, int, int, int],
+    sigma: float,
+) -> None:
+    """In-place: resample ``tile`` to ``region`` size and composite into
+    ``canvas`` under the analytic blurred-rect mask."""
+    x1, y1, x2, y2 = (int(r) for r in region)
+    if _on_gpu(canvas):
+        ext.get_ext(True).blend_tile(
+            canvas, tile.float(), x1, y1, x2, y2,
+            int(mask_rect[0]), int(mask_rect[1]), int(mask_rect[2]),
+            int(mask_rect[3]), float(sigma),
+        )
+        return
+    rw, rh = x2 - x1, y2 - y1
+    t = _lanczos_resample_cpu(tile.float(), 0, 0, tile.shape[2], tile.shape[1], rw, rh)
+    m_full = rect_mask_cpu(canvas.shape[1], canvas.shape[2], mask_rect, sigma)
+    m = m_full[y1:y2, x1:x2][None, :, :, None]
+    patch = canvas[:, y1:y2, x1:x2, :]
+    canvas[:, y1:y2, x1:x2, :] = patch * (1 - m) + t * m

@@ -1,0 +1,32 @@
+// Python bindings for the gfx950 kernel extension.
+#include <torch/extension.h>
+
+torch::Tensor group_norm_fused(torch::Tensor x, int64_t groups,
+                               torch::Tensor weight, torch::Tensor bias,
+                               double eps, bool fuse_silu);
+torch::Tensor layer_norm_bf16(torch::Tensor x, torch::Tensor gamma,
+                              torch::Tensor beta, double eps);
+torch::Tensor act_mul_bf16(torch::Tensor a, torch::Tensor b, bool gelu);
+torch::Tensor extract_resize(torch::Tensor src, int64_t x1, int64_t y1,
+                             int64_t x2, int64_t y2, int64_t ow, int64_t oh);
+void blend_tile(torch::Tensor canvas, torch::Tensor tile, int64_t x1,
+                int64_t y1, int64_t x2, int64_t y2, int64_t mx1, int64_t my1,
+                int64_t mx2, int64_t my2, double sigma);
+torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                       int64_t heads, int64_t kv_heads, int64_t nk_real,
+                       double scale);
+torch::Tensor mfma_selftest(torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("group_norm_fused", &group_norm_fused,
+        "fused GroupNorm(+SiLU), NCHW bf16");
+  m.def("layer_norm", &layer_norm_bf16, "LayerNorm over last dim, bf16");
+  m.def("act_mul", &act_mul_bf16, "a * act(b) elementwise, bf16");
+  m.def("extract_resize", &extract_resize,
+        "crop region of [B,H,W,C] f32 -> Lanczos-3 resample");
+  m.def("blend_tile", &blend_tile,
+        "fused resample + blurred-rect mask + composite (in-place canvas)");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward, bf16 MFMA");
+  m.def("mfma_selftest", &mfma_selftest,
+        "single-wave 16x16x32 bf16 MFMA with the kernel fragment layouts");
+}

@@ -1,0 +1,199 @@
+// Tile pipeline kernels (MI355X/gfx950).
+//
+// The reference does tile extraction / resize / masking / blending on the
+// CPU with PIL (upscale/tile_ops.py:34-155, 289-349).
+// Here the whole pipeline stays in HBM: fused gather+Lanczos-3 resample for
+// tile extraction, and a fused resample+mask+composite kernel for the seam
+// blend. The tile mask is evaluated analytically per pixel as the closed
+// form of a Gaussian-blurred white rectangle (product of erf ramps), which
+// is exactly the separable limit of the reference's rect+GaussianBlur mask
+// without ever materializing a canvas-sized mask.
+//
+// Canonical image layout: [B, H, W, C] float32 in [0,1] (C innermost).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <cmath>
+#include "common.h"
+
+namespace {
+
+constexpr int LANCZOS_A = 3;
+
+__device__ __forceinline__ float sinc_f(float x) {
+  if (fabsf(x) < 1e-6f) return 1.0f;
+  float px = 3.14159265358979f * x;
+  return __sinf(px) / px;
+}
+
+__device__ __forceinline__ float lanczos3_f(float x) {
+  if (fabsf(x) >= (float)LANCZOS_A) return 0.0f;
+  return sinc_f(x) * sinc_f(x / (float)LANCZOS_A);
+}
+
+// One output pixel of a separable Lanczos-3 resample of the source window
+// [sx1,sx2)x[sy1,sy2) to an out_w x out_h grid. Filter is widened by the
+// scale factor when minifying (standard high-quality resample behavior).
+// Weights are renormalized over in-bounds taps (edge clamp).
+template <int MAX_TAPS>
+__device__ void lanczos_sample(const float* __restrict__ src, int src_h,
+                               int src_w, int channels, long long src_batch_off,
+                               float sx1, float sy1, float scale_x, float scale_y,
+                               int ox, int oy, float* out_px) {
+  const float fscale_x = fmaxf(scale_x, 1.0f);
+  const float fscale_y = fmaxf(scale_y, 1.0f);
+  const float cx = sx1 + (ox + 0.5f) * scale_x - 0.5f;
+  const float cy = sy1 + (oy + 0.5f) * scale_y - 0.5f;
+  const float support_x = LANCZOS_A * fscale_x;
+  const float support_y = LANCZOS_A * fscale_y;
+  int x0 = (int)floorf(cx - support_x + 0.5f);
+  int x1 = (int)floorf(cx + support_x + 0.5f);
+  int y0 = (int)floorf(cy - support_y + 0.5f);
+  int y1 = (int)floorf(cy + support_y + 0.5f);
+
+  float wx[MAX_TAPS], wy[MAX_TAPS];
+  int nx = 0, ny = 0;
+  float wxsum = 0.f, wysum = 0.f;
+  for (int x = x0; x <= x1 && nx < MAX_TAPS; ++x, ++nx) {
+    float w = lanczos3_f((x - cx) / fscale_x);
+    wx[nx] = w;
+    wxsum += w;
+  }
+  for (int y = y0; y <= y1 && ny < MAX_TAPS; ++y, ++ny) {
+    float w = lanczos3_f((y - cy) / fscale_y);
+    wy[ny] = w;
+    wysum += w;
+  }
+  const float norm = 1.0f / (wxsum * wysum + 1e-12f);
+  for (int c = 0; c < channels; ++c) out_px[c] = 0.f;
+  for (int iy = 0; iy < ny; ++iy) {
+    const int y = min(max(y0 + iy, 0), src_h - 1);
+    for (int ix = 0; ix < nx; ++ix) {
+      const int x = min(max(x0 + ix, 0), src_w - 1);
+      const float w = wx[ix] * wy[iy];
+      const float* p = src + src_batch_off + ((long long)y * src_w + x) * channels;
+      for (int c = 0; c < channels; ++c) out_px[c] += w * p[c];
+    }
+  }
+  for (int c = 0; c < channels; ++c) out_px[c] *= norm;
+}
+
+// Blurred-rect mask: product of two erf ramps per axis, sigma = mask_blur.
+__device__ __forceinline__ float erf_ramp(float x, float lo, float hi,
+                                          float inv_s) {
+  // integral of a unit box [lo,hi) against a Gaussian centered at pixel x.
+  return 0.5f * (erff((x + 0.5f - lo) * inv_s) - erff((x + 0.5f - hi) * inv_s));
+}
+
+__device__ __forceinline__ float rect_mask(float x, float y, float rx1,
+                                           float ry1, float rx2, float ry2,
+                                           float sigma) {
+  if (sigma <= 0.f) {
+    return (x >= rx1 && x < rx2 && y >= ry1 && y < ry2) ? 1.f : 0.f;
+  }
+  const float inv_s = 0.70710678f / sigma;  // 1/(sigma*sqrt(2))
+  return erf_ramp(x, rx1, rx2, inv_s) * erf_ramp(y, ry1, ry2, inv_s);
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// extract_resize: crop region (x1,y1,x2,y2) of src [B,H,W,C] -> [B,oh,ow,C].
+// ---------------------------------------------------------------------------
+
+__global__ void extract_resize_kernel(const float* __restrict__ src,
+                                      float* __restrict__ dst, int B, int H,
+                                      int W, int C, int x1, int y1, int x2,
+                                      int y2, int ow, int oh) {
+  const long long total = (long long)B * ow * oh;
+  const float scale_x = (float)(x2 - x1) / ow;
+  const float scale_y = (float)(y2 - y1) / oh;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int ox = (int)(i % ow);
+    const int oy = (int)((i / ow) % oh);
+    const int b = (int)(i / ((long long)ow * oh));
+    float px[8];
+    lanczos_sample<16>(src, H, W, C, (long long)b * H * W * C, (float)x1,
+                       (float)y1, scale_x, scale_y, ox, oy, px);
+    float* d = dst + ((long long)b * oh * ow + (long long)oy * ow + ox) * C;
+    for (int c = 0; c < C; ++c) d[c] = px[c];
+  }
+}
+
+torch::Tensor extract_resize(torch::Tensor src, int64_t x1, int64_t y1,
+                             int64_t x2, int64_t y2, int64_t ow, int64_t oh) {
+  TORCH_CHECK(src.is_cuda() && src.dim() == 4 && src.scalar_type() == at::kFloat,
+              "src must be [B,H,W,C] float32 on device");
+  TORCH_CHECK(src.size(3) <= 8, "C <= 8");
+  TORCH_CHECK(x2 > x1 && y2 > y1, "empty region");
+  auto s = src.contiguous();
+  const int B = s.size(0), H = s.size(1), W = s.size(2), C = s.size(3);
+  auto dst = torch::empty({B, oh, ow, C}, s.options());
+  const long long total = (long long)B * ow * oh;
+  int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(extract_resize_kernel, dim3(blocks), dim3(256), 0, stream,
+                     s.data_ptr<float>(), dst.data_ptr<float>(), B, H, W, C,
+                     (int)x1, (int)y1, (int)x2, (int)y2, (int)ow, (int)oh);
+  HIP_CHECK_LAUNCH();
+  return dst;
+}
+
+// ---------------------------------------------------------------------------
+// blend_tile: composite a processed tile back into the canvas (in place).
+//   canvas [B,H,W,C] f32; tile [B,th,tw,C] f32 (processing size);
+//   crop region (x1..y2) on the canvas; mask rect (rx1..ry2) + sigma.
+// For each canvas pixel in the region:
+//   t = lanczos(tile -> region size); m = blurred-rect mask(x, y)
+//   canvas = canvas*(1-m) + t*m
+// ---------------------------------------------------------------------------
+
+__global__ void blend_tile_kernel(float* __restrict__ canvas,
+                                  const float* __restrict__ tile, int B, int H,
+                                  int W, int C, int th, int tw, int x1, int y1,
+                                  int x2, int y2, float rx1, float ry1,
+                                  float rx2, float ry2, float sigma) {
+  const int rw = x2 - x1, rh = y2 - y1;
+  const long long total = (long long)B * rw * rh;
+  const float scale_x = (float)tw / rw;
+  const float scale_y = (float)th / rh;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int ox = (int)(i % rw);
+    const int oy = (int)((i / rw) % rh);
+    const int b = (int)(i / ((long long)rw * rh));
+    const int gx = x1 + ox, gy = y1 + oy;
+    const float m = rect_mask((float)gx, (float)gy, rx1, ry1, rx2, ry2, sigma);
+    if (m <= 1e-6f) continue;
+    float px[8];
+    lanczos_sample<16>(tile, th, tw, C, (long long)b * th * tw * C, 0.f, 0.f,
+                       scale_x, scale_y, ox, oy, px);
+    float* d = canvas + ((long long)b * H * W + (long long)gy * W + gx) * C;
+    for (int c = 0; c < C; ++c) d[c] = d[c] * (1.f - m) + px[c] * m;
+  }
+}
+
+void blend_tile(torch::Tensor canvas, torch::Tensor tile, int64_t x1,
+                int64_t y1, int64_t x2, int64_t y2, int64_t mx1, int64_t my1,
+                int64_t mx2, int64_t my2, double sigma) {
+  TORCH_CHECK(canvas.is_cuda() && canvas.dim() == 4 &&
+              canvas.scalar_type() == at::kFloat && canvas.is_contiguous());
+  TORCH_CHECK(tile.is_cuda() && tile.dim() == 4 &&
+              tile.scalar_type() == at::kFloat);
+  TORCH_CHECK(canvas.size(0) == tile.size(0) && canvas.size(3) == tile.size(3));
+  auto t = tile.contiguous();
+  const int B = canvas.size(0), H = canvas.size(1), W = canvas.size(2),
+            C = canvas.size(3);
+  TORCH_CHECK(x1 >= 0 && y1 >= 0 && x2 <= W && y2 <= H && x2 > x1 && y2 > y1,
+              "region out of bounds");
+  const long long total = (long long)B * (x2 - x1) * (y2 - y1);
+  int blocks = (int)std::min<long long>((total + 255) / 256, 4096);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(blend_tile_kernel, dim3(blocks), dim3(256), 0, stream,
+                     canvas.data_ptr<float>(), t.data_ptr<float>(), B, H, W, C,
+                     (int)t.size(1), (int)t.size(2), (int)x1, (int)y1, (int)x2,
+                     (int)y2, (float)mx1, (float)my1, (float)mx2, (float)my2,
+                     (float)sigma);
+  HIP_CHECK_LAUNCH();
+}
