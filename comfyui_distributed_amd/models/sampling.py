@@ -1,0 +1,182 @@
+"""Samplers + noise schedules (the common_ksampler equivalent).
+
+Reference counterpart: ComfyUI's ``common_ksampler`` invoked per tile at
+upscale/tile_ops.py:225-229. Implements the discrete
+eps-prediction parameterization (linear beta schedule), karras/normal sigma
+schedules, Euler / Euler-ancestral / DPM++ 2M samplers, classifier-free
+guidance and partial denoise (the USDU ``denoise`` fraction).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+SAMPLERS = ("euler", "euler_ancestral", "dpmpp_2m")
+SCHEDULERS = ("normal", "karras")
+
+
+def make_alphas_cumprod(n: int = 1000, beta_start: float = 0.00085,
+                        beta_end: float = 0.012) -> torch.Tensor:
+    betas = torch.linspace(beta_start**0.5, beta_end**0.5, n, dtype=torch.float64) ** 2
+    return torch.cumprod(1.0 - betas, dim=0)
+
+
+class NoiseSchedule:
+    """sigma(t) = sqrt((1 - a_t) / a_t) over the discrete training schedule."""
+
+    def __init__(self, n: int = 1000):
+        ac = make_alphas_cumprod(n)
+        self.sigmas_all = ((1 - ac) / ac).sqrt().float()  # ascending in t
+        self.log_sigmas = self.sigmas_all.log()
+
+    @property
+    def sigma_max(self) -> float:
+        return float(self.sigmas_all[-1])
+
+    @property
+    def sigma_min(self) -> float:
+        return float(self.sigmas_all[0])
+
+    def timestep(self, sigma: torch.Tensor) -> torch.Tensor:
+        """Continuous timestep for a sigma (interpolated in log space)."""
+        log_sigma = sigma.log()
+        dists = log_sigma.reshape(-1, 1) - self.log_sigmas.to(sigma.device)[None]
+        low_idx = dists.ge(0).cumsum(dim=1).argmax(dim=1).clamp(max=len(self.sigmas_all) - 2)
+        high_idx = low_idx + 1
+        low = self.log_sigmas.to(sigma.device)[low_idx]
+        high = self.log_sigmas.to(sigma.device)[high_idx]
+        w = ((low - log_sigma.reshape(-1)) / (low - high)).clamp(0, 1)
+        return ((1 - w) * low_idx + w * high_idx).reshape(sigma.shape)
+
+    def sigmas(self, steps: int, scheduler: str = "normal",
+               denoise: float = 1.0) -> torch.Tensor:
+        """Descending sigma sequence of length steps+1 ending at 0.
+
+        ``denoise < 1`` keeps the tail: compute the full schedule at
+        ``ceil(steps / denoise)`` and take the last steps+1 entries (USDU /
+        ComfyUI partial-denoise behavior)."""
+        if denoise <= 0:
+            return torch.zeros(1)
+        total = steps if denoise >= 1.0 else max(int(math.ceil(steps / denoise)), steps)
+        if scheduler == "karras":
+            rho = 7.0
+            ramp = torch.linspace(0, 1, total)
+            mn, mx = self.sigma_min, self.sigma_max
+            s = (mx ** (1 / rho) + ramp * (mn ** (1 / rho) - mx ** (1 / rho))) ** rho
+        elif scheduler == "normal":
+            t = torch.linspace(len(self.sigmas_all) - 1, 0, total)
+            s = self.sigmas_all[t.long().clamp(0, len(self.sigmas_all) - 1)]
+        else:
+            raise ValueError(f"unknown scheduler {scheduler!r}")
+        s = torch.cat([s, torch.zeros(1)])
+        return s[-(steps + 1):]
+
+
+class CFGDenoiser:
+    """eps-model + classifier-free guidance -> denoised prediction x0."""
+
+    def __init__(self, unet, schedule: NoiseSchedule, cond, uncond, cfg_scale: float):
+        self.unet = unet
+        self.schedule = schedule
+        self.cond = cond
+        self.uncond = uncond
+        self.cfg_scale = cfg_scale
+
+    def __call__(self, x: torch.Tensor, sigma: torch.Tensor) -> torch.Tensor:
+        b = x.shape[0]
+        sig = sigma.reshape(-1).to(x.device)
+        if sig.numel() == 1:
+            sig = sig.expand(b)
+        t = self.schedule.timestep(sig)
+        # eps parameterization: x_t = x0 + sigma * eps, model input is
+        # x_t / sqrt(1 + sigma^2) (the "v-scaling" of discrete eps models).
+        c_in = (1.0 / (1.0 + sig**2).sqrt()).reshape(-1, 1, 1, 1).to(x.dtype)
+        need_cfg = self.cfg_scale != 1.0 and self.uncond is not None
+        if need_cfg:
+            x_in = torch.cat([x * c_in] * 2)
+            t_in = torch.cat([t] * 2)
+            ctx = torch.cat([self.cond["context"].expand(b, -1, -1),
+                             self.uncond["context"].expand(b, -1, -1)])
+            y = None
+            if self.cond.get("y") is not None:
+                y = torch.cat([self.cond["y"].expand(b, -1),
+                               self.uncond["y"].expand(b, -1)])
+            eps = self.unet(x_in, t_in, ctx, y=y)
+            eps_c, eps_u = eps.chunk(2)
+            eps = eps_u + self.cfg_scale * (eps_c - eps_u)
+        else:
+            ctx = self.cond["context"].expand(b, -1, -1)
+            y = self.cond.get("y")
+            if y is not None:
+                y = y.expand(b, -1)
+            eps = self.unet(x * c_in, t, ctx, y=y)
+        return x - eps.float() * sig.reshape(-1, 1, 1, 1)
+
+
+def sample(denoiser, noise_or_latent: torch.Tensor, sigmas: torch.Tensor,
+           sampler: str = "euler", seed: int | None = None,
+           start_from_latent: torch.Tensor | None = None) -> torch.Tensor:
+    """Run the sampler loop. ``noise_or_latent`` is pure noise for txt2img;
+    for img2img pass ``start_from_latent`` and the noised start is formed
+    here as latent + noise * sigmas[0]."""
+    x = noise_or_latent.float() * sigmas[0]
+    if start_from_latent is not None:
+        x = start_from_latent.float() + noise_or_latent.float() * sigmas[0]
+    gen = None
+    if seed is not None:
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+    if sampler == "euler":
+        return _sample_euler(denoiser, x, sigmas)
+    if sampler == "euler_ancestral":
+        return _sample_euler_ancestral(denoiser, x, sigmas, gen)
+    if sampler == "dpmpp_2m":
+        return _sample_dpmpp_2m(denoiser, x, sigmas)
+    raise ValueError(f"unknown sampler {sampler!r}")
+
+
+def _sample_euler(denoiser, x, sigmas):
+    for i in range(len(sigmas) - 1):
+        sigma = sigmas[i]
+        denoised = denoiser(x, sigma)
+        d = (x - denoised) / sigma
+        x = x + d * (sigmas[i + 1] - sigma)
+    return x
+
+
+def _sample_euler_ancestral(denoiser, x, sigmas, gen):
+    for i in range(len(sigmas) - 1):
+        sigma, sigma_next = sigmas[i], sigmas[i + 1]
+        denoised = denoiser(x, sigma)
+        if sigma_next == 0:
+            x = denoised
+            continue
+        sigma_up = (sigma_next**2 * (sigma**2 - sigma_next**2) / sigma**2).sqrt()
+        sigma_down = (sigma_next**2 - sigma_up**2).sqrt()
+        d = (x - denoised) / sigma
+        x = x + d * (sigma_down - sigma)
+        noise = torch.randn(x.shape, generator=gen, dtype=torch.float32).to(x.device)
+        x = x + noise * sigma_up
+    return x
+
+
+def _sample_dpmpp_2m(denoiser, x, sigmas):
+    old_denoised = None
+    for i in range(len(sigmas) - 1):
+        sigma, sigma_next = sigmas[i], sigmas[i + 1]
+        denoised = denoiser(x, sigma)
+        t, t_next = -sigma.log(), -sigma_next.log() if sigma_next > 0 else None
+        if sigma_next == 0:
+            x = denoised
+        elif old_denoised is None:
+            h = t_next - t
+            x = (sigma_next / sigma) * x - (-h).expm1() * denoised
+        else:
+            h = t_next - t
+            h_last = t - (-sigmas[i - 1].log())
+            r = h_last / h
+            denoised_d = (1 + 1 / (2 * r)) * denoised - (1 / (2 * r)) * old_denoised
+            x = (sigma_next / sigma) * x - (-h).expm1() * denoised_d
+        old_denoised = denoised
+    return x
