@@ -1,0 +1,40 @@
+"""txt2img generation (the KSampler+VAEDecode path a seed-parallel workflow
+runs on every GPU; reference delegates this to ComfyUI — SURVEY.md §0)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+
+from ..models.sampling import CFGDenoiser, sample
+
+
+@dataclass
+class GenParams:
+    seed: int = 0
+    steps: int = 20
+    cfg: float = 7.5
+    sampler_name: str = "euler"
+    scheduler: str = "normal"
+    width: int = 512
+    height: int = 512
+    batch_size: int = 1
+
+
+def generate_latents(stack, cond, uncond, p: GenParams) -> torch.Tensor:
+    ds = stack.vae.downscale
+    shape = (p.batch_size, stack.cfg.unet.in_channels, p.height // ds, p.width // ds)
+    g = torch.Generator(device="cpu").manual_seed(p.seed)
+    noise = torch.randn(shape, generator=g).to(stack.device)
+    sigmas = stack.schedule.sigmas(p.steps, p.scheduler).to(stack.device)
+    denoiser = CFGDenoiser(stack.unet, stack.schedule, cond, uncond, p.cfg)
+    with torch.no_grad():
+        return sample(denoiser, noise, sigmas, sampler=p.sampler_name, seed=p.seed)
+
+
+def generate_images(stack, cond, uncond, p: GenParams) -> torch.Tensor:
+    """Returns [B, H, W, 3] float32 in [0,1] on the stack device."""
+    latents = generate_latents(stack, cond, uncond, p)
+    with torch.no_grad():
+        return stack.vae.decode(latents.to(stack.dtype)).float()

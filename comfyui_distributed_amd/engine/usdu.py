@@ -1,0 +1,168 @@
+"""Ultimate-SD-Upscale tile engine (the compute body of the USDU node).
+
+Reference counterpart: upscale/tile_ops.py process_tile/process_tiles_batch
++ upscale/modes/single_gpu.py. Differences by design:
+
+* The whole pipeline stays on-device: extract+resample, VAE encode, sampler
+  loop, VAE decode and seam blend are HIP kernels / device tensors — the
+  reference round-trips through PIL on the CPU for every tile.
+* Tiles are batched along a combined (image, tile) axis — the reference
+  batches only across the image batch, one tile at a time
+  (tile_ops.py:239-287). With 288 GB HBM per GPU there is no reason not to
+  fill the device.
+* Per-tile noise is derived from (seed, tile_index, batch_index), so results
+  are bit-identical no matter which GPU processes which tile — the
+  reference has the same property because every worker re-seeds per tile.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+import torch
+
+from ..models.sampling import CFGDenoiser, NoiseSchedule, sample
+from ..ops import dispatch as ops
+from ..utils import usdu_math
+
+
+@dataclass
+class USDUParams:
+    seed: int = 0
+    steps: int = 20
+    cfg: float = 8.0
+    sampler_name: str = "euler"
+    scheduler: str = "normal"
+    denoise: float = 0.35
+    tile_width: int = 512
+    tile_height: int = 512
+    padding: int = 32
+    mask_blur: int = 8
+    force_uniform_tiles: bool = True
+    tiled_decode: bool = False
+    tile_batch: int = 4  # tiles sampled together per sampler call
+
+
+def plan_for_image(width: int, height: int, p: USDUParams):
+    return usdu_math.plan_tiles(
+        width, height, p.tile_width, p.tile_height, p.padding,
+        uniform=p.force_uniform_tiles,
+    )
+
+
+def _tile_noise(seed: int, tile_idx: int, batch_idx: int, shape) -> torch.Tensor:
+    g = torch.Generator(device="cpu").manual_seed(
+        (seed * 1_000_003 + tile_idx * 1009 + batch_idx) & 0x7FFFFFFF
+    )
+    return torch.randn(shape, generator=g)
+
+
+def sample_tiles(
+    stack,
+    cond: dict,
+    uncond: dict | None,
+    params: USDUParams,
+    canvas: torch.Tensor,
+    plans: list,
+    tile_indices: list[int],
+) -> dict[tuple[int, int], torch.Tensor]:
+    """Sample the given tiles of ``canvas`` (all batch images) and return
+    {(tile_idx, batch_idx): processed tile [1,Ph,Pw,C] float32}.
+
+    Extraction always reads the ORIGINAL canvas, never partially-blended
+    state, so results are bit-identical regardless of which rank processes
+    which tiles (the reference's progressive local blend —
+    upscale/modes/static.py:268-280 — makes worker output depend on its
+    tile assignment; this framework deliberately drops that so the
+    distributed result equals the single-GPU result exactly).
+    """
+    if not tile_indices:
+        return {}
+    B = canvas.shape[0]
+    schedule: NoiseSchedule = stack.schedule
+    sigmas = schedule.sigmas(params.steps, params.scheduler, params.denoise).to(
+        canvas.device
+    )
+    denoiser = CFGDenoiser(stack.unet, schedule, cond, uncond, params.cfg)
+
+    work = [(t, b) for t in sorted(tile_indices) for b in range(B)]
+    step = max(params.tile_batch, 1)
+    results: dict[tuple[int, int], torch.Tensor] = {}
+    for i in range(0, len(work), step):
+        chunk = work[i : i + step]
+        # ---- extract + resample each (tile, batch) crop to process size ----
+        crops = []
+        for t, b in chunk:
+            plan = plans[t]
+            x1, y1, x2, y2 = plan.crop_region
+            pw, ph = plan.process_size
+            crops.append(ops.extract_resize(canvas[b : b + 1], (x1, y1, x2, y2), pw, ph))
+        batch_img = torch.cat(crops, dim=0)
+        # ---- encode -> img2img sample -> decode ----
+        with torch.no_grad():
+            latents = stack.vae.encode(batch_img)
+            noise = torch.stack(
+                [
+                    _tile_noise(params.seed, t, b, latents.shape[1:])
+                    for t, b in chunk
+                ]
+            ).to(latents.device)
+            latent_out = sample(
+                denoiser,
+                noise,
+                sigmas,
+                sampler=params.sampler_name,
+                seed=params.seed,
+                start_from_latent=latents.float(),
+            )
+            out_img = stack.vae.decode(latent_out.to(stack.dtype))
+        for j, (t, b) in enumerate(chunk):
+            results[(t, b)] = out_img[j : j + 1].float()
+    return results
+
+
+def blend_results(
+    canvas: torch.Tensor,
+    results: dict[tuple[int, int], torch.Tensor],
+    plans: list,
+    params: USDUParams,
+) -> None:
+    """One canonical blend pass: ascending (tile_idx, batch_idx) — the
+    deterministic order the reference enforces for worker tiles
+    (upscale/modes/static.py:521-527)."""
+    for (t, b) in sorted(results.keys()):
+        blend_processed_tile(canvas, results[(t, b)], plans[t], params, batch_index=b)
+
+
+def process_tiles(
+    stack, cond, uncond, params: USDUParams, canvas: torch.Tensor, plans: list,
+    tile_indices: list[int],
+) -> None:
+    """sample_tiles + blend_results in place (single-rank convenience)."""
+    results = sample_tiles(stack, cond, uncond, params, canvas, plans, tile_indices)
+    blend_results(canvas, results, plans, params)
+
+
+def blend_processed_tile(canvas, tile_img, plan, params: USDUParams, batch_index: int):
+    """Blend one processed tile (at process size) into one canvas image
+    (in place — canvas must be contiguous [B,H,W,C])."""
+    assert canvas.is_contiguous()
+    ops.blend_tile(
+        canvas[batch_index : batch_index + 1],
+        tile_img,
+        plan.crop_region,
+        plan.tile_rect,
+        float(params.mask_blur),
+    )
+
+
+def process_single_gpu(stack, cond, uncond, params: USDUParams,
+                       image: torch.Tensor) -> torch.Tensor:
+    """Single-device USDU: all tiles locally (reference
+    upscale/modes/single_gpu.py:8-72 semantics, batched-per-tile)."""
+    canvas = image.to(stack.device, torch.float32).clone()
+    B, H, W, _ = canvas.shape
+    plans = plan_for_image(W, H, params)
+    process_tiles(stack, cond, uncond, params, canvas, plans,
+                  list(range(len(plans))))
+    return canvas
