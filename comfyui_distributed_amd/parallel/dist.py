@@ -1,0 +1,158 @@
+"""torch.distributed context: one process per GPU, RCCL over xGMI.
+
+This replaces the reference's process-per-GPU HTTP mesh
+(api/orchestration/dispatch.py + utils/network.py) for the intra-node data
+plane: the control plane becomes a TCPStore (tile_queue.py), the data plane
+NCCL-API collectives (= RCCL on ROCm). On CPU test hosts the same code runs
+on the gloo backend.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+from ..utils import constants
+from ..utils.logging import log
+
+
+@dataclass
+class DistContext:
+    rank: int
+    world_size: int
+    device: torch.device
+    backend: str
+
+    @property
+    def is_master(self) -> bool:
+        return self.rank == 0
+
+    def barrier(self):
+        if self.world_size > 1:
+            dist.barrier()
+
+    def sync_device(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+
+def init_from_env(backend: str | None = None, timeout_s: float = 600.0) -> DistContext:
+    """Initialize from torchrun env (RANK/WORLD_SIZE/LOCAL_RANK/MASTER_*).
+
+    Single-process (no env / WORLD_SIZE=1) returns a degenerate context
+    without initializing a process group.
+    """
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    if world <= 1:
+        return DistContext(0, 1, device, "none")
+    if backend is None:
+        backend = "nccl" if use_cuda else "gloo"
+    os.environ.setdefault("MASTER_ADDR", constants.DEFAULT_MASTER_ADDR)
+    os.environ.setdefault("MASTER_PORT", "29500")
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend,
+            rank=rank,
+            world_size=world,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
+    log(f"rank {rank}/{world} initialized ({backend}, {device})")
+    return DistContext(rank, world, device, backend)
+
+
+def gather_tensor_lists(
+    ctx: DistContext,
+    tensors: list[torch.Tensor],
+    meta: list[tuple],
+) -> tuple[list[torch.Tensor], list[tuple]] | None:
+    """Gather variable-length tensor lists (+ metadata tuples) to rank 0.
+
+    Each rank contributes N_i tensors of IDENTICAL shape (tile results).
+    Returns (all_tensors, all_meta) on rank 0, None elsewhere. Uses
+    all_gather for counts then batched p2p send/recv — on RCCL each source
+    rank streams over its own xGMI link to rank 0 rather than serializing
+    through a symmetric collective (SURVEY.md §5.8 topology note).
+    """
+    if ctx.world_size == 1:
+        return tensors, meta
+    device = ctx.device if ctx.backend == "nccl" else torch.device("cpu")
+    count = torch.tensor([len(tensors)], dtype=torch.int64, device=device)
+    counts = [torch.zeros_like(count) for _ in range(ctx.world_size)]
+    dist.all_gather(counts, count)
+    counts = [int(c.item()) for c in counts]
+
+    # metadata: fixed-width int64 rows
+    meta_width = len(meta[0]) if meta else 2
+    if tensors:
+        shape = tensors[0].shape
+        payload = torch.stack([t.to(device) for t in tensors]).contiguous()
+        meta_t = torch.tensor(meta, dtype=torch.int64, device=device)
+    else:
+        shape = None
+        payload = None
+        meta_t = None
+
+    # shape agreement: broadcast reference shape from the first non-empty rank
+    shape_src = next((r for r, c in enumerate(counts) if c > 0), None)
+    if shape_src is None:
+        return ([], []) if ctx.is_master else None
+    shape_t = torch.zeros(8, dtype=torch.int64, device=device)
+    if ctx.rank == shape_src and shape is not None:
+        dims = torch.tensor(shape, dtype=torch.int64, device=device)
+        shape_t[0] = len(shape)
+        shape_t[1 : 1 + len(shape)] = dims
+    dist.broadcast(shape_t, src=shape_src)
+    ndim = int(shape_t[0].item())
+    ref_shape = tuple(int(x) for x in shape_t[1 : 1 + ndim])
+
+    if ctx.is_master:
+        all_tensors: list[torch.Tensor] = list(tensors)
+        all_meta: list[tuple] = list(meta)
+        for src in range(1, ctx.world_size):
+            n = counts[src]
+            if n == 0:
+                continue
+            buf = torch.empty((n, *ref_shape), dtype=torch.float32, device=device)
+            mbuf = torch.empty((n, meta_width), dtype=torch.int64, device=device)
+            dist.recv(buf, src=src)
+            dist.recv(mbuf, src=src)
+            for i in range(n):
+                all_tensors.append(buf[i])
+                all_meta.append(tuple(int(x) for x in mbuf[i]))
+        return all_tensors, all_meta
+    if counts[ctx.rank] > 0:
+        dist.send(payload.to(torch.float32), dst=0)
+        dist.send(meta_t, dst=0)
+    return None
+
+
+def broadcast_tensor(ctx: DistContext, t: torch.Tensor | None, src: int = 0):
+    """Broadcast a tensor (shape+dtype negotiated) from src to all ranks."""
+    if ctx.world_size == 1:
+        return t
+    device = ctx.device if ctx.backend == "nccl" else torch.device("cpu")
+    shape_t = torch.zeros(9, dtype=torch.int64, device=device)
+    if ctx.rank == src:
+        assert t is not None
+        shape_t[0] = t.dim()
+        for i, s in enumerate(t.shape):
+            shape_t[1 + i] = s
+    dist.broadcast(shape_t, src=src)
+    ndim = int(shape_t[0].item())
+    shape = tuple(int(x) for x in shape_t[1 : 1 + ndim])
+    if ctx.rank == src:
+        buf = t.to(device=device, dtype=torch.float32).contiguous()
+    else:
+        buf = torch.empty(shape, dtype=torch.float32, device=device)
+    dist.broadcast(buf, src=src)
+    return buf

@@ -1,0 +1,72 @@
+"""Distributed USDU: tile pull-queue over all ranks, RCCL result gather,
+canonical blend on rank 0.
+
+This is the intra-node re-architecture of the reference's static mode
+(upscale/modes/static.py): the HTTP request_image/submit_tiles loop becomes
+a TCPStore pull queue + one batched p2p gather over xGMI; the master
+participates exactly like the reference master does (pulls tiles from its
+own queue). Determinism: extraction reads the original canvas and blending
+runs in ascending (tile, batch) order on rank 0, so the result is
+bit-identical to the single-GPU run regardless of tile assignment.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ..engine.usdu import USDUParams, blend_results, plan_for_image, sample_tiles
+from ..utils.logging import debug_log
+from .dist import DistContext, gather_tensor_lists
+from .tile_queue import TileQueue
+
+
+def run_distributed_usdu(
+    ctx: DistContext,
+    store,
+    stack,
+    cond: dict,
+    uncond: dict | None,
+    params: USDUParams,
+    image: torch.Tensor,
+    job_id: str = "usdu",
+) -> torch.Tensor | None:
+    """All ranks call this with the same input image (broadcast upstream or
+    constructed identically). Returns the blended canvas on rank 0, None on
+    workers."""
+    # .clone(): .to() is a no-copy alias when dtype/device already match,
+    # and the blend pass mutates the canvas in place.
+    canvas = image.to(stack.device, torch.float32).clone().contiguous()
+    B, H, W, _ = canvas.shape
+    plans = plan_for_image(W, H, params)
+
+    queue = TileQueue(store, job_id, ctx.rank)
+    if ctx.is_master:
+        queue.init_job(len(plans))
+    ctx.barrier()
+
+    tensors: list[torch.Tensor] = []
+    meta: list[tuple[int, int]] = []
+    done = 0
+    while True:
+        idx = queue.pop()
+        if idx is None:
+            break
+        res = sample_tiles(stack, cond, uncond, params, canvas, plans, [idx])
+        for (t, b), img in sorted(res.items()):
+            tensors.append(img[0])
+            meta.append((t, b))
+        queue.mark_done(idx)
+        queue.heartbeat()
+        done += 1
+    debug_log(f"rank {ctx.rank}: processed {done} tiles")
+
+    gathered = gather_tensor_lists(ctx, tensors, meta)
+    if not ctx.is_master:
+        return None
+    all_tensors, all_meta = gathered
+    results = {
+        (int(t), int(b)): tensor[None].float().to(canvas.device)
+        for tensor, (t, b) in zip(all_tensors, all_meta)
+    }
+    blend_results(canvas, results, plans, params)
+    return canvas
