@@ -47,17 +47,26 @@ def run_distributed_usdu(
     tensors: list[torch.Tensor] = []
     meta: list[tuple[int, int]] = []
     done = 0
+    # pop enough tile ids per iteration to fill the sampler batch
+    # (each tile id covers all B batch images)
+    ids_per_iter = max(1, params.tile_batch // max(B, 1))
     while True:
-        idx = queue.pop()
-        if idx is None:
+        ids: list[int] = []
+        while len(ids) < ids_per_iter:
+            idx = queue.pop()
+            if idx is None:
+                break
+            ids.append(idx)
+        if not ids:
             break
-        res = sample_tiles(stack, cond, uncond, params, canvas, plans, [idx])
+        res = sample_tiles(stack, cond, uncond, params, canvas, plans, ids)
         for (t, b), img in sorted(res.items()):
             tensors.append(img[0])
             meta.append((t, b))
-        queue.mark_done(idx)
+        for idx in ids:
+            queue.mark_done(idx)
         queue.heartbeat()
-        done += 1
+        done += len(ids)
     debug_log(f"rank {ctx.rank}: processed {done} tiles")
 
     gathered = gather_tensor_lists(ctx, tensors, meta)
