@@ -93,6 +93,7 @@ class CFGDenoiser:
         # eps parameterization: x_t = x0 + sigma * eps, model input is
         # x_t / sqrt(1 + sigma^2) (the "v-scaling" of discrete eps models).
         c_in = (1.0 / (1.0 + sig**2).sqrt()).reshape(-1, 1, 1, 1).to(x.dtype)
+        # the UNet casts inputs to its own parameter dtype (bf16 on GPU)
         need_cfg = self.cfg_scale != 1.0 and self.uncond is not None
         if need_cfg:
             x_in = torch.cat([x * c_in] * 2)

@@ -283,7 +283,14 @@ class UNetModel(nn.Module):
         self.out_conv = nn.Conv2d(ch0, cfg.out_channels, 3, padding=1)
 
     def forward(self, x, timesteps, context, y=None):
-        emb = self.time_embed(timestep_embedding(timesteps, self.cfg.model_channels))
+        dtype = self.out_conv.weight.dtype
+        x = x.to(dtype)
+        context = context.to(dtype)
+        if y is not None:
+            y = y.to(dtype)
+        emb = self.time_embed(
+            timestep_embedding(timesteps, self.cfg.model_channels).to(dtype)
+        )
         if self.cfg.adm_in_channels and y is not None:
             emb = emb + self.label_emb(y)
         hs = []

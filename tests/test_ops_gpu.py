@@ -89,7 +89,9 @@ def test_groupnorm_silu_numerics(extmod, shape, groups):
         x.cuda().to(torch.bfloat16), groups, w.cuda(), b.cuda(), 1e-5, True
     ).float().cpu()
     ref = F.silu(F.group_norm(x, groups, w, b, 1e-5))
-    assert (y - ref).abs().max().item() < 0.05
+    # bf16 output quantization: |err| <= atol + bf16-relative term
+    bound = 0.02 + 0.01 * ref.abs()
+    assert ((y - ref).abs() <= bound).all(), (y - ref).abs().max().item()
 
 
 def test_groupnorm_no_silu(extmod):
