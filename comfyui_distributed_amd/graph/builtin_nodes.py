@@ -1,0 +1,250 @@
+"""Built-in compute nodes (the subset of ComfyUI L0 the distributed
+workflows need): model loading, conditioning, sampling, VAE, image IO.
+
+The reference delegates all of these to the host ComfyUI;
+this framework provides them natively on the gfx950 engine. Node/socket
+names follow the ComfyUI wire vocabulary so reference workflows map 1:1.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ..engine.generate import GenParams, generate_latents
+from ..models import create_diffusion_stack
+from ..models.sampling import CFGDenoiser, SAMPLERS, SCHEDULERS, sample
+from ..utils.image import decode_png_bytes, encode_png_bytes
+
+_STACK_CACHE: dict[tuple, object] = {}
+
+
+def _device_dtype():
+    if torch.cuda.is_available():
+        return "cuda:0", torch.bfloat16
+    return "cpu", torch.float32
+
+
+class _ContextNode:
+    _ctx: dict = {}
+
+    def set_context(self, ctx):
+        self._ctx = ctx
+
+
+class CheckpointLoader(_ContextNode):
+    """Loads a model family as a DiffusionStack (random-init weights; there
+    is no checkpoint file IO in this offline environment — the ``ckpt_name``
+    selects the architecture family, e.g. "sd15", "sdxl", "tiny")."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"ckpt_name": ("STRING", {"default": "sd15"})}}
+
+    RETURN_TYPES = ("MODEL", "CLIP", "VAE")
+    FUNCTION = "load"
+    CATEGORY = "loaders"
+
+    def load(self, ckpt_name="sd15"):
+        device = self._ctx.get("device")
+        if device is None:
+            device, dtype = _device_dtype()
+        else:
+            dtype = torch.bfloat16 if str(device).startswith("cuda") else torch.float32
+        key = (str(ckpt_name), str(device))
+        if key not in _STACK_CACHE:
+            _STACK_CACHE[key] = create_diffusion_stack(
+                str(ckpt_name), device=device, dtype=dtype
+            )
+        stack = _STACK_CACHE[key]
+        # MODEL and VAE are the stack handle + its VAE; CLIP is the stack
+        # too (conditioning is synthesized from it).
+        return (stack, stack, stack.vae)
+
+
+class CLIPTextEncode(_ContextNode):
+    """Synthesizes conditioning of the model's context shape from the text
+    (deterministic hash -> seed; no text encoder weights exist offline)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"text": ("STRING", {"default": ""}), "clip": ("CLIP",)}}
+
+    RETURN_TYPES = ("CONDITIONING",)
+    FUNCTION = "encode"
+    CATEGORY = "conditioning"
+
+    def encode(self, text="", clip=None):
+        seed = abs(hash(str(text))) % (2**31)
+        return (clip.make_conditioning(seed),)
+
+
+class EmptyLatentImage(_ContextNode):
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {
+            "required": {
+                "width": ("INT", {"default": 512}),
+                "height": ("INT", {"default": 512}),
+                "batch_size": ("INT", {"default": 1}),
+            }
+        }
+
+    RETURN_TYPES = ("LATENT",)
+    FUNCTION = "generate"
+    CATEGORY = "latent"
+
+    def generate(self, width=512, height=512, batch_size=1):
+        return ({"samples": torch.zeros(int(batch_size), 4, int(height) // 8,
+                                        int(width) // 8), "size": (width, height)},)
+
+
+class KSampler(_ContextNode):
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {
+            "required": {
+                "model": ("MODEL",),
+                "seed": ("INT", {"default": 0}),
+                "steps": ("INT", {"default": 20}),
+                "cfg": ("FLOAT", {"default": 8.0}),
+                "sampler_name": (list(SAMPLERS),),
+                "scheduler": (list(SCHEDULERS),),
+                "positive": ("CONDITIONING",),
+                "negative": ("CONDITIONING",),
+                "latent_image": ("LATENT",),
+                "denoise": ("FLOAT", {"default": 1.0}),
+            }
+        }
+
+    RETURN_TYPES = ("LATENT",)
+    FUNCTION = "sample"
+    CATEGORY = "sampling"
+
+    def sample(self, model, seed, steps, cfg, sampler_name, scheduler,
+               positive, negative, latent_image, denoise=1.0):
+        stack = model
+        shape = latent_image["samples"].shape
+        g = torch.Generator(device="cpu").manual_seed(int(seed))
+        noise = torch.randn(shape, generator=g).to(stack.device)
+        sigmas = stack.schedule.sigmas(int(steps), scheduler, float(denoise)).to(
+            stack.device
+        )
+        denoiser = CFGDenoiser(stack.unet, stack.schedule, positive, negative,
+                               float(cfg))
+        start = None
+        if denoise < 1.0 and latent_image["samples"].abs().sum() > 0:
+            start = latent_image["samples"].to(stack.device).float()
+        with torch.no_grad():
+            out = sample(denoiser, noise, sigmas, sampler=sampler_name,
+                         seed=int(seed), start_from_latent=start)
+        return ({"samples": out.cpu()},)
+
+
+class VAEDecode(_ContextNode):
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"samples": ("LATENT",), "vae": ("VAE",)}}
+
+    RETURN_TYPES = ("IMAGE",)
+    FUNCTION = "decode"
+    CATEGORY = "latent"
+
+    def decode(self, samples, vae):
+        with torch.no_grad():
+            img = vae.decode(samples["samples"].to(next(vae.parameters()).device))
+        return (img.float().cpu(),)
+
+
+class VAEEncode(_ContextNode):
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"pixels": ("IMAGE",), "vae": ("VAE",)}}
+
+    RETURN_TYPES = ("LATENT",)
+    FUNCTION = "encode"
+    CATEGORY = "latent"
+
+    def encode(self, pixels, vae):
+        with torch.no_grad():
+            z = vae.encode(pixels)
+        return ({"samples": z.cpu()},)
+
+
+class LoadImage(_ContextNode):
+    """Loads an image from the input directory (or a synthetic one when the
+    name is "synthetic:<W>x<H>")."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"image": ("STRING", {"default": ""})}}
+
+    RETURN_TYPES = ("IMAGE",)
+    FUNCTION = "load"
+    CATEGORY = "image"
+
+    def load(self, image=""):
+        name = str(image)
+        if name.startswith("synthetic:"):
+            w, h = (int(v) for v in name.split(":", 1)[1].split("x"))
+            g = torch.Generator().manual_seed(0)
+            return (torch.rand(1, h, w, 3, generator=g),)
+        from pathlib import Path
+
+        base = Path(self._ctx.get("input_dir", "input"))
+        raw = (base / name).read_bytes()
+        return (decode_png_bytes(raw),)
+
+
+class SaveImage(_ContextNode):
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"images": ("IMAGE",),
+                             "filename_prefix": ("STRING", {"default": "out"})}}
+
+    RETURN_TYPES = ()
+    FUNCTION = "save"
+    OUTPUT_NODE = True
+    CATEGORY = "image"
+
+    def save(self, images, filename_prefix="out"):
+        from pathlib import Path
+
+        out_dir = Path(self._ctx.get("output_dir", "output"))
+        out_dir.mkdir(parents=True, exist_ok=True)
+        paths = []
+        for i in range(images.shape[0]):
+            p = out_dir / f"{filename_prefix}_{i:05d}.png"
+            p.write_bytes(encode_png_bytes(images[i : i + 1], compress_level=4))
+            paths.append(str(p))
+        sink = self._ctx.get("saved_images")
+        if isinstance(sink, list):
+            sink.extend(paths)
+        return ()
+
+
+class PreviewImage(SaveImage):
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"images": ("IMAGE",)}}
+
+    FUNCTION = "preview"
+
+    def preview(self, images):
+        sink = self._ctx.get("preview_images")
+        if isinstance(sink, list):
+            sink.append(images)
+        return ()
+
+
+BUILTIN_CLASS_MAPPINGS = {
+    "CheckpointLoader": CheckpointLoader,
+    "CheckpointLoaderSimple": CheckpointLoader,
+    "CLIPTextEncode": CLIPTextEncode,
+    "EmptyLatentImage": EmptyLatentImage,
+    "KSampler": KSampler,
+    "VAEDecode": VAEDecode,
+    "VAEEncode": VAEEncode,
+    "LoadImage": LoadImage,
+    "SaveImage": SaveImage,
+    "PreviewImage": PreviewImage,
+}

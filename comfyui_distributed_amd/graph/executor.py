@@ -1,0 +1,129 @@
+"""Workflow graph executor.
+
+The reference relies on ComfyUI's executor (L0 in SURVEY.md §1); this
+framework executes prompt graphs itself: topological evaluation of the
+``{node_id: {class_type, inputs}}`` wire format against a node registry
+(the distributed nodes from nodes/ plus the built-in compute nodes below).
+"""
+
+from __future__ import annotations
+
+from typing import Any
+
+import torch
+
+from ..utils.errors import PromptValidationError
+from .prompt import PromptGraph, is_link
+
+
+class NodeRegistry:
+    def __init__(self):
+        self._classes: dict[str, type] = {}
+
+    def register(self, name: str, cls: type) -> None:
+        self._classes[name] = cls
+
+    def register_map(self, mapping: dict[str, type]) -> None:
+        self._classes.update(mapping)
+
+    def get(self, name: str) -> type | None:
+        return self._classes.get(name)
+
+    def names(self) -> list[str]:
+        return list(self._classes)
+
+
+def default_registry() -> NodeRegistry:
+    from ..nodes import NODE_CLASS_MAPPINGS
+    from . import builtin_nodes
+
+    reg = NodeRegistry()
+    reg.register_map(NODE_CLASS_MAPPINGS)
+    reg.register_map(builtin_nodes.BUILTIN_CLASS_MAPPINGS)
+    return reg
+
+
+def validate_prompt(prompt: dict, registry: NodeRegistry) -> None:
+    """Structural validation: known classes, resolvable links, acyclic."""
+    errors: dict[str, list[str]] = {}
+    graph = PromptGraph(prompt)
+    for nid, node in graph.items():
+        cls_name = node.get("class_type", "")
+        if registry.get(cls_name) is None:
+            errors.setdefault(nid, []).append(f"unknown node class {cls_name!r}")
+            continue
+        for name, value in node.get("inputs", {}).items():
+            if is_link(value) and str(value[0]) not in prompt:
+                errors.setdefault(nid, []).append(
+                    f"input {name!r} links to missing node {value[0]!r}"
+                )
+    # cycle check via DFS
+    state: dict[str, int] = {}
+
+    def visit(nid: str):
+        if state.get(nid) == 1:
+            errors.setdefault(nid, []).append("cycle detected")
+            return
+        if state.get(nid) == 2:
+            return
+        state[nid] = 1
+        for _n, src in graph.input_links(nid):
+            if src in prompt:
+                visit(src)
+        state[nid] = 2
+
+    for nid in prompt:
+        visit(nid)
+    if errors:
+        raise PromptValidationError("prompt validation failed", errors)
+
+
+class Executor:
+    """Evaluates a prompt graph; caches per-node outputs within one run."""
+
+    def __init__(self, registry: NodeRegistry | None = None,
+                 context: dict[str, Any] | None = None):
+        self.registry = registry or default_registry()
+        #: shared objects node bodies may need (device, model cache, ...)
+        self.context = context or {}
+
+    def execute(self, prompt: dict) -> dict[str, tuple]:
+        """Run the graph; returns {node_id: outputs tuple} for OUTPUT_NODEs
+        and every executed node."""
+        validate_prompt(prompt, self.registry)
+        graph = PromptGraph(prompt)
+        cache: dict[str, tuple] = {}
+
+        def eval_node(nid: str) -> tuple:
+            if nid in cache:
+                return cache[nid]
+            node = graph.node(nid)
+            cls = self.registry.get(node["class_type"])
+            kwargs = {}
+            for name, value in node.get("inputs", {}).items():
+                if is_link(value):
+                    src_out = eval_node(str(value[0]))
+                    kwargs[name] = src_out[int(value[1])]
+                else:
+                    kwargs[name] = value
+            inst = cls()
+            if hasattr(inst, "set_context"):
+                inst.set_context(self.context)
+            fn = getattr(inst, getattr(cls, "FUNCTION", "run"))
+            out = fn(**kwargs)
+            if not isinstance(out, tuple):
+                out = (out,)
+            cache[nid] = out
+            return out
+
+        # evaluate every sink (output nodes and nodes nobody consumes)
+        consumed: set[str] = set()
+        for nid in graph.node_ids():
+            for _name, src in graph.input_links(nid):
+                consumed.add(src)
+        for nid in graph.node_ids():
+            cls = self.registry.get(graph.class_of(nid))
+            is_output = bool(getattr(cls, "OUTPUT_NODE", False))
+            if is_output or nid not in consumed:
+                eval_node(nid)
+        return cache

@@ -34,7 +34,8 @@ TINY_UNET = UNetConfig(
     context_dim=64,
     num_heads=2,
 )
-TINY_VAE = VAEConfig(base_channels=16, channel_mult=(1, 2), num_res_blocks=1)
+# 4 levels so the latent downscale matches the production VAEs (x8)
+TINY_VAE = VAEConfig(base_channels=8, channel_mult=(1, 1, 2, 2), num_res_blocks=1)
 
 MODEL_CONFIGS: dict[str, StackConfig] = {
     "sd15": StackConfig("sd15", SD15_UNET, SD_VAE, native_size=512),

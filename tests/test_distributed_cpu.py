@@ -102,7 +102,7 @@ def _body_usdu(ctx, port):
     stack = create_diffusion_stack("tiny", seed=7)
     cond = stack.make_conditioning(0)
     p = USDUParams(seed=3, steps=2, cfg=1.0, denoise=0.5, tile_width=16,
-                   tile_height=16, padding=4, mask_blur=2, tile_batch=2)
+                   tile_height=16, padding=16, mask_blur=2, tile_batch=2)
     g = torch.Generator().manual_seed(99)
     img = torch.rand(1, 32, 32, 3, generator=g)
     out = run_distributed_usdu(ctx, store, stack, cond, None, p, img)

@@ -1,0 +1,81 @@
+import pytest
+import torch
+
+from comfyui_distributed_amd.graph.executor import Executor, default_registry, validate_prompt
+from comfyui_distributed_amd.utils.errors import PromptValidationError
+
+
+def txt2img_prompt(model="tiny", w=16, h=16):
+    return {
+        "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": model}},
+        "2": {"class_type": "CLIPTextEncode", "inputs": {"text": "a cat", "clip": ["1", 1]}},
+        "3": {"class_type": "CLIPTextEncode", "inputs": {"text": "", "clip": ["1", 1]}},
+        "4": {"class_type": "EmptyLatentImage",
+              "inputs": {"width": w, "height": h, "batch_size": 1}},
+        "5": {"class_type": "KSampler", "inputs": {
+            "model": ["1", 0], "seed": 3, "steps": 2, "cfg": 2.0,
+            "sampler_name": "euler", "scheduler": "karras",
+            "positive": ["2", 0], "negative": ["3", 0],
+            "latent_image": ["4", 0], "denoise": 1.0}},
+        "6": {"class_type": "VAEDecode", "inputs": {"samples": ["5", 0], "vae": ["1", 2]}},
+        "7": {"class_type": "PreviewImage", "inputs": {"images": ["6", 0]}},
+    }
+
+
+def test_validate_rejects_unknown_class():
+    reg = default_registry()
+    with pytest.raises(PromptValidationError) as ei:
+        validate_prompt({"1": {"class_type": "NoSuchNode", "inputs": {}}}, reg)
+    assert "1" in ei.value.node_errors
+
+
+def test_validate_rejects_missing_link_and_cycle():
+    reg = default_registry()
+    with pytest.raises(PromptValidationError):
+        validate_prompt(
+            {"1": {"class_type": "PreviewImage", "inputs": {"images": ["9", 0]}}},
+            reg,
+        )
+    with pytest.raises(PromptValidationError):
+        validate_prompt(
+            {
+                "1": {"class_type": "VAEDecode", "inputs": {"samples": ["2", 0], "vae": ["2", 0]}},
+                "2": {"class_type": "VAEDecode", "inputs": {"samples": ["1", 0], "vae": ["1", 0]}},
+            },
+            reg,
+        )
+
+
+def test_txt2img_graph_executes():
+    previews = []
+    ex = Executor(context={"preview_images": previews, "device": "cpu"})
+    cache = ex.execute(txt2img_prompt())
+    assert previews and previews[0].shape == (1, 16, 16, 3)
+    assert torch.isfinite(previews[0]).all()
+    # KSampler output cached
+    assert "5" in cache
+
+
+def test_distributed_seed_node_in_graph():
+    prompt = {
+        "1": {"class_type": "DistributedSeed",
+              "inputs": {"seed": 100, "is_worker": True, "worker_id": "worker_2"}},
+    }
+    ex = Executor()
+    cache = ex.execute(prompt)
+    assert cache["1"] == (103,)  # 100 + index 2 + 1
+
+
+def test_batch_divider_in_graph():
+    import comfyui_distributed_amd.graph.builtin_nodes  # noqa: F401
+
+    prompt = {
+        "1": {"class_type": "LoadImage", "inputs": {"image": "synthetic:8x8"}},
+        "2": {"class_type": "ImageBatchDivider",
+              "inputs": {"images": ["1", 0], "divide_by": 2}},
+    }
+    ex = Executor()
+    cache = ex.execute(prompt)
+    outs = cache["2"]
+    assert len(outs) == 10
+    assert outs[0].shape[0] == 1 and outs[1].shape[0] == 0

@@ -22,7 +22,7 @@ def test_tiny_vae_roundtrip_shapes():
     img = torch.rand(1, 16, 16, 3)
     with torch.no_grad():
         z = stack.vae.encode(img)
-        assert z.shape == (1, 4, 8, 8)
+        assert z.shape == (1, 4, 2, 2)
         back = stack.vae.decode(z)
     assert back.shape == (1, 16, 16, 3)
     assert torch.isfinite(back).all()
