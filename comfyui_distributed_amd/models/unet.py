@@ -196,17 +196,26 @@ class Upsample(nn.Module):
         super().__init__()
         self.conv = nn.Conv2d(channels, channels, 3, padding=1)
 
-    def forward(self, x):
-        return self.conv(F.interpolate(x, scale_factor=2, mode="nearest"))
+    def forward(self, x, output_shape=None):
+        # odd latent sizes (e.g. 68->34->17->9) need the exact skip shape on
+        # the way back up, not blind 2x (ComfyUI passes output_shape the
+        # same way)
+        if output_shape is not None:
+            y = F.interpolate(x, size=tuple(output_shape), mode="nearest")
+        else:
+            y = F.interpolate(x, scale_factor=2, mode="nearest")
+        return self.conv(y)
 
 
 class _TimestepSequential(nn.ModuleList):
-    def forward(self, x, emb, context):
+    def forward(self, x, emb, context, output_shape=None):
         for layer in self:
             if isinstance(layer, ResBlock):
                 x = layer(x, emb)
             elif isinstance(layer, SpatialTransformer):
                 x = layer(x, context)
+            elif isinstance(layer, Upsample):
+                x = layer(x, output_shape=output_shape)
             else:
                 x = layer(x)
         return x
@@ -301,5 +310,6 @@ class UNetModel(nn.Module):
         h = self.middle_block(h, emb, context)
         for block in self.output_blocks:
             h = torch.cat([h, hs.pop()], dim=1)
-            h = block(h, emb, context)
+            output_shape = hs[-1].shape[2:] if hs else None
+            h = block(h, emb, context, output_shape=output_shape)
         return self.out_conv(self.out_norm(h))

@@ -1,0 +1,435 @@
+"""The REST server: wire-compatible /distributed/* + /prompt endpoints.
+
+Endpoint parity with the reference's aiohttp routes:
+job_routes.py (queue, job_complete, prepare_job, clear_memory, check_file),
+usdu_routes.py (heartbeat, submit_tiles, submit_image, request_image,
+job_status), config_routes.py (config CRUD), worker_routes.py
+(network_info, system_info, launch/stop/managed workers, log tail) and the
+ComfyUI-side GET/POST /prompt that probes and dispatch rely on.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import hashlib
+import os
+import platform
+import socket
+import time
+import uuid
+
+from aiohttp import web
+
+from ..graph.executor import Executor, default_registry, validate_prompt
+from ..nodes.collector import decode_job_complete_envelope
+from ..nodes.runtime import NodeRuntime, set_runtime
+from ..server.job_state import JobState
+from ..utils import constants
+from ..utils.async_bridge import ServerLoop
+from ..utils.config import (
+    WORKER_TYPES,
+    config_transaction,
+    get_worker_by_id,
+    load_config,
+)
+from ..utils.errors import PromptValidationError
+from ..utils.logging import log
+from . import network
+from .orchestration import orchestrate_distributed_execution
+from .queue_request import QueueRequestError, parse_queue_request_payload
+
+
+def _err(message: str, status: int = 400):
+    return web.json_response({"error": message}, status=status)
+
+
+class DistributedServer:
+    """Application state + routes. One instance per process (master or
+    worker role — the role only differs in config and env)."""
+
+    def __init__(self, is_worker: bool = False, device: str | None = None):
+        self.job_state = JobState()
+        self.is_worker = is_worker
+        self.device = device
+        self.prompt_queue: asyncio.Queue = asyncio.Queue()
+        self.executing = False
+        self.executor = Executor(context={"device": device} if device else {})
+        self.registry = self.executor.registry
+        self._exec_task: asyncio.Task | None = None
+        self._log_buffer: list[str] = []
+        self.managed = {}  # worker_id -> process handle
+
+        rt = NodeRuntime(self.job_state)
+        rt.probe_worker = self._probe_worker_by_id  # type: ignore
+        set_runtime(rt)
+
+    async def _probe_worker_by_id(self, worker_id: str):
+        cfg = load_config()
+        worker = get_worker_by_id(cfg, worker_id)
+        if worker is None:
+            return None
+        return await network.probe_worker(network.build_worker_url(worker))
+
+    # ------------------------------------------------------------------ app
+
+    def build_app(self) -> web.Application:
+        app = web.Application(client_max_size=constants.MAX_PAYLOAD_SIZE + 2**20)
+        r = app.router
+        r.add_get("/prompt", self.get_prompt)
+        r.add_post("/prompt", self.post_prompt)
+        r.add_post("/distributed/queue", self.post_queue)
+        r.add_post("/distributed/job_complete", self.post_job_complete)
+        r.add_post("/distributed/prepare_job", self.post_prepare_job)
+        r.add_post("/distributed/clear_memory", self.post_clear_memory)
+        r.add_post("/distributed/heartbeat", self.post_heartbeat)
+        r.add_post("/distributed/submit_tiles", self.post_submit_tiles)
+        r.add_post("/distributed/submit_image", self.post_submit_image)
+        r.add_post("/distributed/request_image", self.post_request_image)
+        r.add_get("/distributed/job_status/{job_id}", self.get_job_status)
+        r.add_post("/distributed/job_status", self.post_job_status)
+        r.add_get("/distributed/queue_status/{job_id}", self.get_queue_status)
+        r.add_get("/distributed/config", self.get_config)
+        r.add_post("/distributed/config/update_worker", self.post_update_worker)
+        r.add_post("/distributed/config/delete_worker", self.post_delete_worker)
+        r.add_post("/distributed/config/update_setting", self.post_update_setting)
+        r.add_post("/distributed/config/update_master", self.post_update_master)
+        r.add_post("/distributed/check_file", self.post_check_file)
+        r.add_get("/distributed/network_info", self.get_network_info)
+        r.add_get("/distributed/system_info", self.get_system_info)
+        r.add_post("/distributed/launch_worker", self.post_launch_worker)
+        r.add_post("/distributed/stop_worker", self.post_stop_worker)
+        r.add_get("/distributed/managed_workers", self.get_managed_workers)
+        app.on_startup.append(self._on_startup)
+        return app
+
+    async def _on_startup(self, app):
+        ServerLoop.set(asyncio.get_running_loop())
+        self._exec_task = asyncio.create_task(self._execution_loop())
+
+    # ----------------------------------------------------------- execution
+
+    async def _execution_loop(self):
+        loop = asyncio.get_running_loop()
+        while True:
+            prompt, client_id, prompt_id = await self.prompt_queue.get()
+            self.executing = True
+            try:
+                await loop.run_in_executor(None, self.executor.execute, prompt)
+                log(f"prompt {prompt_id} done (client {client_id})")
+            except Exception as exc:  # noqa: BLE001
+                log(f"prompt {prompt_id} FAILED: {exc!r}")
+            finally:
+                self.executing = False
+
+    async def enqueue_local(self, prompt: dict, client_id: str) -> str:
+        validate_prompt(prompt, self.registry)
+        prompt_id = uuid.uuid4().hex
+        await self.prompt_queue.put((prompt, client_id, prompt_id))
+        return prompt_id
+
+    def queue_remaining(self) -> int:
+        return self.prompt_queue.qsize() + (1 if self.executing else 0)
+
+    # -------------------------------------------------------------- routes
+
+    async def get_prompt(self, request):
+        return web.json_response(
+            {"exec_info": {"queue_remaining": self.queue_remaining()}}
+        )
+
+    async def post_prompt(self, request):
+        data = await request.json()
+        prompt = data.get("prompt")
+        if not isinstance(prompt, dict):
+            return _err("missing prompt")
+        try:
+            prompt_id = await self.enqueue_local(prompt, data.get("client_id", ""))
+        except PromptValidationError as exc:
+            return web.json_response(
+                {"error": str(exc), "node_errors": exc.node_errors}, status=400
+            )
+        return web.json_response({"prompt_id": prompt_id, "number": self.prompt_queue.qsize()})
+
+    async def post_queue(self, request):
+        try:
+            payload = parse_queue_request_payload(await request.json())
+        except QueueRequestError as exc:
+            return _err(str(exc))
+        result = await orchestrate_distributed_execution(
+            payload, self.job_state, self.enqueue_local
+        )
+        return web.json_response(result)
+
+    async def post_job_complete(self, request):
+        data = await request.json()
+        try:
+            item = decode_job_complete_envelope(data)
+        except (ValueError, KeyError) as exc:
+            return _err(str(exc))
+        q = await self.job_state.get_queue_waiting(str(data["job_id"]))
+        if q is None:
+            return _err("unknown job (queue never created)", status=404)
+        await q.put(item)
+        return web.json_response({"status": "ok"})
+
+    async def post_prepare_job(self, request):
+        data = await request.json()
+        job_id = data.get("job_id")
+        if not job_id:
+            return _err("missing job_id")
+        await self.job_state.ensure_queue(str(job_id))
+        return web.json_response({"status": "ready"})
+
+    async def post_clear_memory(self, request):
+        import gc
+
+        import torch
+
+        gc.collect()
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+        return web.json_response({"status": "ok"})
+
+    # ---- USDU tile endpoints ---------------------------------------------
+
+    async def post_heartbeat(self, request):
+        data = await request.json()
+        job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
+        if job is None:
+            return _err("unknown job", status=404)
+        job.worker_status[str(data.get("worker_id", ""))] = time.time()
+        return web.json_response({"status": "ok"})
+
+    async def post_submit_tiles(self, request):
+        from .usdu_http import decode_tile_submission
+
+        data = await request.json()
+        job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
+        if job is None:
+            return _err("unknown job", status=404)
+        items = decode_tile_submission(data)
+        for item in items:
+            await job.results.put(item)
+        if data.get("is_last"):
+            job.finished_workers.add(str(data.get("worker_id", "")))
+        job.worker_status[str(data.get("worker_id", ""))] = time.time()
+        return web.json_response({"status": "ok", "received": len(items)})
+
+    async def post_submit_image(self, request):
+        from ..utils.image import decode_png_base64
+
+        data = await request.json()
+        job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
+        if job is None:
+            return _err("unknown job", status=404)
+        await job.results.put({
+            "image_idx": int(data.get("image_idx", 0)),
+            "tensor": decode_png_base64(data["image"]),
+            "worker_id": str(data.get("worker_id", "")),
+            "is_last": bool(data.get("is_last", False)),
+        })
+        if data.get("is_last"):
+            job.finished_workers.add(str(data.get("worker_id", "")))
+        return web.json_response({"status": "ok"})
+
+    async def post_request_image(self, request):
+        """The pull scheduler (reference usdu_routes.py:168-215)."""
+        data = await request.json()
+        job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
+        if job is None:
+            return _err("unknown job", status=404)
+        wid = str(data.get("worker_id", ""))
+        job.worker_status[wid] = time.time()
+        pending = getattr(job, "pending_tasks", None)
+        key = "tile_idx"
+        if pending is None:
+            pending = job.pending_images
+            key = "image_idx"
+        try:
+            idx = await asyncio.wait_for(pending.get(), constants.QUEUE_POP_WAIT)
+        except asyncio.TimeoutError:
+            return web.json_response({key: None, "estimated_remaining": 0})
+        job.assigned_to_workers[idx] = wid
+        return web.json_response({
+            key: idx,
+            "estimated_remaining": pending.qsize(),
+            "batched_static": getattr(job, "batched_static", False),
+        })
+
+    async def get_job_status(self, request):
+        job_id = request.match_info["job_id"]
+        job = await self.job_state.get_tile_job(job_id)
+        return web.json_response({"ready": job is not None})
+
+    async def post_job_status(self, request):
+        data = await request.json()
+        job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
+        return web.json_response({"ready": job is not None})
+
+    async def get_queue_status(self, request):
+        job_id = request.match_info["job_id"]
+        async with self.job_state.jobs_lock:
+            q = self.job_state.pending_jobs.get(job_id)
+        return web.json_response(
+            {"exists": q is not None, "pending": q.qsize() if q else 0}
+        )
+
+    # ---- config CRUD ------------------------------------------------------
+
+    async def get_config(self, request):
+        return web.json_response(load_config())
+
+    async def post_update_worker(self, request):
+        data = await request.json()
+        wid = data.get("id")
+        if wid is None:
+            return _err("missing worker id")
+        async with config_transaction() as cfg:
+            worker = get_worker_by_id(cfg, wid)
+            fields = {k: v for k, v in data.items() if k in (
+                "name", "host", "port", "cuda_device", "enabled", "extra_args", "type")}
+            if worker is None:
+                required = {"name", "port"}
+                if not required.issubset(data):
+                    return _err("new worker needs name and port")
+                worker = {"id": str(wid), "enabled": False, "type": "local",
+                          "host": "", "cuda_device": 0, "extra_args": ""}
+                worker.update(fields)
+                if worker.get("type") not in WORKER_TYPES:
+                    return _err(f"bad worker type {worker.get('type')!r}")
+                cfg["workers"].append(worker)
+            else:
+                if "type" in fields and fields["type"] not in WORKER_TYPES:
+                    return _err(f"bad worker type {fields['type']!r}")
+                worker.update(fields)
+        return web.json_response({"status": "ok"})
+
+    async def post_delete_worker(self, request):
+        data = await request.json()
+        wid = str(data.get("id"))
+        async with config_transaction() as cfg:
+            before = len(cfg["workers"])
+            cfg["workers"] = [w for w in cfg["workers"] if str(w.get("id")) != wid]
+            if len(cfg["workers"]) == before:
+                return _err("unknown worker", status=404)
+        return web.json_response({"status": "ok"})
+
+    ALLOWED_SETTINGS = (
+        "debug", "auto_launch_workers", "stop_workers_on_master_exit",
+        "master_delegate_only", "websocket_orchestration",
+        "worker_timeout_seconds", "worker_probe_concurrency",
+        "worker_prep_concurrency", "media_sync_concurrency",
+        "media_sync_timeout_seconds", "has_auto_populated_workers",
+    )
+
+    async def post_update_setting(self, request):
+        data = await request.json()
+        key, value = data.get("key"), data.get("value")
+        if key not in self.ALLOWED_SETTINGS:
+            return _err(f"setting {key!r} not allowed")
+        async with config_transaction() as cfg:
+            cfg["settings"][key] = value
+        return web.json_response({"status": "ok"})
+
+    async def post_update_master(self, request):
+        data = await request.json()
+        async with config_transaction() as cfg:
+            for k in ("host", "port", "cuda_device", "extra_args"):
+                if k in data:
+                    cfg["master"][k] = data[k]
+        return web.json_response({"status": "ok"})
+
+    # ---- introspection ----------------------------------------------------
+
+    async def post_check_file(self, request):
+        data = await request.json()
+        path = data.get("path")
+        if not path or not os.path.isfile(path):
+            return web.json_response({"exists": False})
+        h = hashlib.md5()
+        with open(path, "rb") as fh:
+            for chunk in iter(lambda: fh.read(1 << 20), b""):
+                h.update(chunk)
+        return web.json_response({"exists": True, "md5": h.hexdigest()})
+
+    async def get_network_info(self, request):
+        import torch
+
+        hostname = socket.gethostname()
+        try:
+            ip = socket.gethostbyname(hostname)
+        except OSError:
+            ip = "127.0.0.1"
+        return web.json_response({
+            "hostname": hostname,
+            "ips": [ip],
+            "cuda_device_count": torch.cuda.device_count()
+            if torch.cuda.is_available() else 0,
+            "master_cuda_device": load_config()["master"].get("cuda_device", 0),
+        })
+
+    async def get_system_info(self, request):
+        return web.json_response({
+            "platform": platform.system().lower(),
+            "path_separator": os.sep,
+            "machine_id": hex(uuid.getnode()),
+            "is_docker": os.path.exists("/.dockerenv"),
+            "is_runpod": bool(os.environ.get("RUNPOD_POD_ID")),
+            "is_worker": self.is_worker,
+        })
+
+    # ---- worker process management ----------------------------------------
+
+    async def post_launch_worker(self, request):
+        from .workers import launch_worker
+
+        data = await request.json()
+        cfg = load_config()
+        worker = get_worker_by_id(cfg, data.get("id"))
+        if worker is None:
+            return _err("unknown worker", status=404)
+        try:
+            handle = launch_worker(worker)
+        except Exception as exc:  # noqa: BLE001
+            return _err(f"launch failed: {exc}", status=500)
+        self.managed[str(worker["id"])] = handle
+        return web.json_response({"status": "launched", "pid": handle.pid})
+
+    async def post_stop_worker(self, request):
+        from .workers import stop_worker
+
+        data = await request.json()
+        wid = str(data.get("id"))
+        handle = self.managed.pop(wid, None)
+        stop_worker(handle, wid)
+        return web.json_response({"status": "stopped"})
+
+    async def get_managed_workers(self, request):
+        out = {}
+        for wid, handle in self.managed.items():
+            out[wid] = {"pid": handle.pid, "alive": handle.poll() is None}
+        return web.json_response({"managed": out})
+
+
+def main():
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=8188)
+    ap.add_argument("--listen", default="0.0.0.0")
+    ap.add_argument("--worker", action="store_true",
+                    default=os.environ.get("DISTGPU_IS_WORKER") == "1")
+    args = ap.parse_args()
+
+    import torch
+
+    device = "cuda:0" if torch.cuda.is_available() else None
+    server = DistributedServer(is_worker=args.worker, device=device)
+    app = server.build_app()
+    role = "worker" if args.worker else "master"
+    log(f"starting {role} server on {args.listen}:{args.port} (device={device})")
+    web.run_app(app, host=args.listen, port=args.port, print=None)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,212 @@
+"""REST API tests (aiohttp test client, no sockets beyond loopback)."""
+
+import asyncio
+import json
+
+import pytest
+import torch
+from aiohttp.test_utils import TestClient, TestServer
+
+from comfyui_distributed_amd.server.app import DistributedServer
+from comfyui_distributed_amd.utils.image import encode_png_base64
+
+
+@pytest.fixture()
+def client(tmp_config, event_loop=None):
+    async def make():
+        srv = DistributedServer()
+        app = srv.build_app()
+        client = TestClient(TestServer(app))
+        await client.start_server()
+        return srv, client
+
+    loop = asyncio.new_event_loop()
+    srv, cl = loop.run_until_complete(make())
+    yield srv, cl, loop
+    loop.run_until_complete(cl.close())
+    loop.close()
+
+
+def run(loop, coro):
+    return loop.run_until_complete(coro)
+
+
+def test_get_prompt_probe(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.get("/prompt")
+        assert r.status == 200
+        data = await r.json()
+        assert "exec_info" in data and "queue_remaining" in data["exec_info"]
+
+    run(loop, go())
+
+
+def test_post_prompt_validation(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.post("/prompt", json={"prompt": {"1": {"class_type": "Nope", "inputs": {}}}})
+        assert r.status == 400
+        data = await r.json()
+        assert "node_errors" in data
+        r = await cl.post("/prompt", json={"prompt": {
+            "1": {"class_type": "DistributedSeed", "inputs": {"seed": 5}}}})
+        assert r.status == 200
+        assert "prompt_id" in await r.json()
+
+    run(loop, go())
+
+
+def test_queue_requires_fields(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.post("/distributed/queue", json={})
+        assert r.status == 400
+        r = await cl.post("/distributed/queue", json={
+            "prompt": {"1": {"class_type": "DistributedSeed", "inputs": {"seed": 1}}},
+            "client_id": "c1",
+        })
+        assert r.status == 400  # missing enabled_worker_ids
+
+    run(loop, go())
+
+
+def test_queue_local_fallback_no_workers(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.post("/distributed/queue", json={
+            "prompt": {"1": {"class_type": "DistributedSeed", "inputs": {"seed": 1}}},
+            "client_id": "c1",
+            "enabled_worker_ids": [],
+        })
+        assert r.status == 200
+        data = await r.json()
+        assert data["participants"] == ["master"]
+
+    run(loop, go())
+
+
+def test_job_complete_roundtrip(client):
+    srv, cl, loop = client
+
+    async def go():
+        await srv.job_state.ensure_queue("jobX")
+        img = torch.rand(1, 8, 8, 3)
+        r = await cl.post("/distributed/job_complete", json={
+            "job_id": "jobX", "worker_id": "w1", "batch_idx": 0,
+            "image": encode_png_base64(img), "is_last": True,
+        })
+        assert r.status == 200
+        q = srv.job_state.pending_jobs["jobX"]
+        item = q.get_nowait()
+        assert item["worker_id"] == "w1" and item["is_last"] is True
+        assert item["tensor"].shape == (1, 8, 8, 3)
+        # invalid envelope
+        r = await cl.post("/distributed/job_complete", json={"job_id": "jobX"})
+        assert r.status == 400
+
+    run(loop, go())
+
+
+def test_tile_job_endpoints(client):
+    srv, cl, loop = client
+
+    async def go():
+        await srv.job_state.init_static_job("tj", n_tiles=2, batch_size=1)
+        # worker polls ready
+        r = await cl.get("/distributed/job_status/tj")
+        assert (await r.json())["ready"] is True
+        r = await cl.get("/distributed/job_status/nope")
+        assert (await r.json())["ready"] is False
+        # pull both tiles then exhaustion
+        got = []
+        for _ in range(2):
+            r = await cl.post("/distributed/request_image",
+                              json={"job_id": "tj", "worker_id": "w1"})
+            got.append((await r.json())["tile_idx"])
+        assert sorted(got) == [0, 1]
+        r = await cl.post("/distributed/request_image",
+                          json={"job_id": "tj", "worker_id": "w1"})
+        assert (await r.json())["tile_idx"] is None
+        # heartbeat + submit
+        r = await cl.post("/distributed/heartbeat",
+                          json={"job_id": "tj", "worker_id": "w1"})
+        assert r.status == 200
+        tile_png = encode_png_base64(torch.rand(1, 8, 8, 3))
+        r = await cl.post("/distributed/submit_tiles", json={
+            "job_id": "tj", "worker_id": "w1", "is_last": True,
+            "tiles": [{"tile_idx": 0, "batch_idx": 0, "image": tile_png}],
+        })
+        assert (await r.json())["received"] == 1
+        job = await srv.job_state.get_tile_job("tj")
+        assert "w1" in job.finished_workers
+        item = job.results.get_nowait()
+        assert item["tile_idx"] == 0 and item["tensor"].shape == (1, 8, 8, 3)
+
+    run(loop, go())
+
+
+def test_config_crud(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.post("/distributed/config/update_worker", json={
+            "id": "w9", "name": "gpu9", "port": 8199, "cuda_device": 1,
+            "enabled": True, "type": "local",
+        })
+        assert r.status == 200
+        r = await cl.get("/distributed/config")
+        cfg = await r.json()
+        assert any(w["id"] == "w9" for w in cfg["workers"])
+        r = await cl.post("/distributed/config/update_setting",
+                          json={"key": "debug", "value": True})
+        assert r.status == 200
+        r = await cl.post("/distributed/config/update_setting",
+                          json={"key": "evil", "value": 1})
+        assert r.status == 400
+        r = await cl.post("/distributed/config/delete_worker", json={"id": "w9"})
+        assert r.status == 200
+        r = await cl.post("/distributed/config/delete_worker", json={"id": "w9"})
+        assert r.status == 404
+
+    run(loop, go())
+
+
+def test_info_endpoints(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.get("/distributed/network_info")
+        data = await r.json()
+        assert "hostname" in data and "cuda_device_count" in data
+        r = await cl.get("/distributed/system_info")
+        data = await r.json()
+        assert data["platform"] and "machine_id" in data
+        r = await cl.post("/distributed/check_file", json={"path": "/nonexistent"})
+        assert (await r.json())["exists"] is False
+
+    run(loop, go())
+
+
+def test_prompt_execution_end_to_end(client):
+    """POST /prompt with a DistributedSeed graph actually executes."""
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.post("/prompt", json={"prompt": {
+            "1": {"class_type": "DistributedSeed",
+                  "inputs": {"seed": 5, "is_worker": True, "worker_id": "worker_0"}},
+        }})
+        assert r.status == 200
+        # wait for the execution loop to drain
+        for _ in range(50):
+            if srv.prompt_queue.qsize() == 0 and not srv.executing:
+                break
+            await asyncio.sleep(0.05)
+        assert srv.prompt_queue.qsize() == 0
+
+    run(loop, go())
