@@ -83,8 +83,13 @@ class DiffusionStack:
         )
 
 
-def create_diffusion_stack(name: str, device="cpu", dtype=None,
-                           seed: int = 0) -> DiffusionStack:
+def create_diffusion_stack(name: str, device="cpu", dtype=None, seed: int = 0):
+    if name in ("wan14b", "wan_tiny"):
+        from .video import WAN_CONFIGS, WanStack
+
+        if dtype is None:
+            dtype = torch.bfloat16 if str(device).startswith("cuda") else torch.float32
+        return WanStack(WAN_CONFIGS[name], device=device, dtype=dtype, seed=seed)
     cfg = MODEL_CONFIGS[name]
     if dtype is None:
         dtype = torch.bfloat16 if (isinstance(device, str) and device.startswith("cuda")) or (
