@@ -65,6 +65,7 @@ def sample_tiles(
     canvas: torch.Tensor,
     plans: list,
     tile_indices: list[int],
+    batch_offset: int = 0,
 ) -> dict[tuple[int, int], torch.Tensor]:
     """Sample the given tiles of ``canvas`` (all batch images) and return
     {(tile_idx, batch_idx): processed tile [1,Ph,Pw,C] float32}.
@@ -103,7 +104,7 @@ def sample_tiles(
             latents = stack.vae.encode(batch_img)
             noise = torch.stack(
                 [
-                    _tile_noise(params.seed, t, b, latents.shape[1:])
+                    _tile_noise(params.seed, t, b + batch_offset, latents.shape[1:])
                     for t, b in chunk
                 ]
             ).to(latents.device)

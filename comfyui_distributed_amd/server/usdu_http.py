@@ -29,8 +29,17 @@ from ..utils.logging import debug_log, log
 def run_usdu_role(*, mode, params, stack, cond, uncond, image, job_id,
                   is_worker, master_url, enabled_workers, worker_id):
     if is_worker:
+        if mode == "dynamic":
+            return _worker_dynamic(params, stack, cond, uncond, image, job_id,
+                                   master_url, worker_id)
         return _worker_static(params, stack, cond, uncond, image, job_id,
                               master_url, worker_id)
+    if mode == "dynamic":
+        return run_async_in_server_loop(
+            _master_dynamic(params, stack, cond, uncond, image, job_id,
+                            enabled_workers),
+            timeout=None,
+        )
     return run_async_in_server_loop(
         _master_static(mode, params, stack, cond, uncond, image, job_id,
                        enabled_workers),
@@ -152,6 +161,124 @@ async def check_and_requeue_timed_out_workers(job, timeout: float):
             del job.assigned_to_workers[task]
         job.worker_status.pop(wid, None)
         log(f"usdu: worker {wid} timed out — requeued {requeued}")
+
+
+async def _master_dynamic(params: USDUParams, stack, cond, uncond, image,
+                          job_id, enabled_workers):
+    """Whole-image parallelism for big batches (reference
+    upscale/modes/dynamic.py:22-211): master participates by pulling image
+    indices from its own queue; worker results arrive as full images."""
+    rt = get_runtime()
+    state = rt.job_state
+    canvas = image.to(stack.device, torch.float32).clone().contiguous()
+    B, H, W, _ = canvas.shape
+    plans = plan_for_image(W, H, params)
+    job = await state.init_dynamic_job(job_id, B)
+    loop = asyncio.get_running_loop()
+    done_images: dict[int, torch.Tensor] = {}
+    timeout = get_worker_timeout_seconds()
+    last_check = time.monotonic()
+
+    def process_whole_image(b: int) -> torch.Tensor:
+        one = canvas[b : b + 1].clone().contiguous()
+        results = sample_tiles(stack, cond, uncond, params, one, plans,
+                               list(range(len(plans))), batch_offset=b)
+        blend_results(one, results, plans, params)
+        return one
+
+    while len(done_images) < B:
+        rt.throw_if_interrupted()
+        try:
+            idx = job.pending_images.get_nowait()
+            job.assigned_to_workers[idx] = "master"
+            done_images[idx] = await loop.run_in_executor(
+                None, process_whole_image, idx
+            )
+            job.completed_images[idx] = True
+            continue
+        except asyncio.QueueEmpty:
+            pass
+        try:
+            item = await asyncio.wait_for(job.results.get(),
+                                          constants.COLLECTOR_SLICE_TIMEOUT)
+            i = int(item["image_idx"])
+            done_images[i] = item["tensor"].to(canvas.device, torch.float32)
+            job.completed_images[i] = True
+            continue
+        except asyncio.TimeoutError:
+            pass
+        now = time.monotonic()
+        if now - last_check >= constants.HEARTBEAT_INTERVAL:
+            last_check = now
+            await _requeue_dynamic_timeouts(job, timeout)
+            if not job.worker_status and job.pending_images.empty():
+                missing = [i for i in range(B) if i not in done_images]
+                for m in missing:
+                    job.pending_images.put_nowait(m)
+                if not missing:
+                    break
+    for i, img in done_images.items():
+        canvas[i : i + 1] = img
+    await state.cleanup_job(job_id)
+    return canvas
+
+
+async def _requeue_dynamic_timeouts(job, timeout: float):
+    rt = get_runtime()
+    now = time.time()
+    for wid in [w for w, ts in job.worker_status.items() if now - ts > timeout]:
+        info = await rt.probe_worker(wid)
+        if info and info.get("exec_info", {}).get("queue_remaining", 0):
+            job.worker_status[wid] = time.time()
+            continue
+        for task, owner in list(job.assigned_to_workers.items()):
+            if owner == wid and task not in job.completed_images:
+                job.pending_images.put_nowait(task)
+                del job.assigned_to_workers[task]
+        job.worker_status.pop(wid, None)
+        log(f"usdu dynamic: worker {wid} timed out — dropped")
+
+
+def _worker_dynamic(params: USDUParams, stack, cond, uncond, image, job_id,
+                    master_url, worker_id):
+    """Reference dynamic.py:213-313: pull image indices, process every tile
+    locally, POST the finished full image."""
+    rt = get_runtime()
+    canvas = image.to(stack.device, torch.float32).clone().contiguous()
+    B, H, W, _ = canvas.shape
+    plans = plan_for_image(W, H, params)
+    for _ in range(constants.JOB_READY_POLL_ATTEMPTS):
+        status = run_async_in_server_loop(
+            rt.post_json(f"{master_url}/distributed/job_status",
+                         {"job_id": job_id}), timeout=30.0)
+        if status.get("ready"):
+            break
+        time.sleep(constants.JOB_READY_POLL_INTERVAL)
+    while True:
+        resp = run_async_in_server_loop(
+            rt.post_json(f"{master_url}/distributed/request_image",
+                         {"job_id": job_id, "worker_id": worker_id}),
+            timeout=60.0,
+        )
+        idx = resp.get("image_idx")
+        if idx is None:
+            break
+        idx = int(idx)
+        one = canvas[idx : idx + 1].clone().contiguous()
+        results = sample_tiles(stack, cond, uncond, params, one, plans,
+                               list(range(len(plans))), batch_offset=idx)
+        blend_results(one, results, plans, params)
+        run_async_in_server_loop(
+            rt.post_json(f"{master_url}/distributed/heartbeat",
+                         {"job_id": job_id, "worker_id": worker_id}),
+            timeout=30.0,
+        )
+        _post_with_retry(rt, f"{master_url}/distributed/submit_image", {
+            "job_id": job_id, "worker_id": worker_id, "image_idx": idx,
+            "image": encode_png_base64(one[0:1].cpu()),
+            "is_last": int(resp.get("estimated_remaining", 0)) == 0,
+        })
+    return None
 
 
 # ---------------------------------------------------------------------------
