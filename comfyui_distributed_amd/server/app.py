@@ -94,6 +94,7 @@ class DistributedServer:
         r.add_post("/distributed/config/update_setting", self.post_update_setting)
         r.add_post("/distributed/config/update_master", self.post_update_master)
         r.add_post("/distributed/check_file", self.post_check_file)
+        r.add_post("/upload/image", self.post_upload_image)
         r.add_get("/distributed/network_info", self.get_network_info)
         r.add_get("/distributed/system_info", self.get_system_info)
         r.add_post("/distributed/launch_worker", self.post_launch_worker)
@@ -351,6 +352,31 @@ class DistributedServer:
             for chunk in iter(lambda: fh.read(1 << 20), b""):
                 h.update(chunk)
         return web.json_response({"exists": True, "md5": h.hexdigest()})
+
+    async def post_upload_image(self, request):
+        """Multipart upload into the input dir (ComfyUI /upload/image
+        parity — media sync pushes files here)."""
+        from pathlib import Path
+
+        reader = await request.multipart()
+        input_dir = Path(self.executor.context.get("input_dir", "input"))
+        input_dir.mkdir(parents=True, exist_ok=True)
+        saved = []
+        while True:
+            part = await reader.next()
+            if part is None:
+                break
+            if part.name in ("image", "file"):
+                fname = os.path.basename(part.filename or "upload.bin")
+                dest = input_dir / fname
+                with open(dest, "wb") as fh:
+                    while True:
+                        chunk = await part.read_chunk(1 << 20)
+                        if not chunk:
+                            break
+                        fh.write(chunk)
+                saved.append(fname)
+        return web.json_response({"saved": saved})
 
     async def get_network_info(self, request):
         import torch

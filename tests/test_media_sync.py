@@ -1,0 +1,33 @@
+from comfyui_distributed_amd.server.media_sync import (
+    convert_path_for_platform,
+    find_media_references,
+)
+
+
+def test_find_media_references():
+    prompt = {
+        "1": {"class_type": "LoadImage", "inputs": {"image": "cat.png"}},
+        "2": {"class_type": "LoadVideo", "inputs": {"video": "clip.mp4"}},
+        "3": {"class_type": "KSampler", "inputs": {"image": ["1", 0], "seed": 3}},
+        "4": {"class_type": "LoadImage", "inputs": {"image": "synthetic:64x64"}},
+        "5": {"class_type": "Note", "inputs": {"text": "not_a_file.png_but_text"}},
+    }
+    refs = find_media_references(prompt)
+    assert ("1", "image", "cat.png") in refs
+    assert ("2", "video", "clip.mp4") in refs
+    assert all(nid not in ("3", "4", "5") for nid, _k, _f in refs)
+
+
+def test_convert_path_separators():
+    assert convert_path_for_platform("a/b/c.png", "\\") == "a\\b\\c.png"
+    assert convert_path_for_platform("a\\b\\c.png", "/") == "a/b/c.png"
+
+
+def test_sync_local_worker_is_noop():
+    import asyncio
+
+    from comfyui_distributed_amd.server.media_sync import sync_worker_media
+
+    prompt = {"1": {"class_type": "LoadImage", "inputs": {"image": "cat.png"}}}
+    out = asyncio.run(sync_worker_media(prompt, {"id": "w", "type": "local"}))
+    assert out == prompt
