@@ -74,11 +74,87 @@ class NoiseSchedule:
         return s[-(steps + 1):]
 
 
+class GraphedModel:
+    """hipGraph-captured model wrapper: one graph per input-shape set.
+
+    The UNet denoise step launches hundreds of small kernels; on MI355X the
+    step is launch-bound (see profiles/r01_usdu_kernel_stats.md: 66k
+    dispatches, wall >> kernel time). Capturing the forward in a hipGraph
+    replays the whole step as one submission. Inputs are copied into static
+    buffers; the returned tensor is the static output (consumed immediately
+    by the sampler math before the next replay).
+    """
+
+    def __init__(self, model):
+        self.model = model
+        self.graphs: dict = {}
+        self.enabled = torch.cuda.is_available()
+        self._failed = False
+
+    def __call__(self, x, t, ctx, y=None):
+        if not self.enabled or self._failed or not x.is_cuda:
+            return self.model(x, t, ctx, y=y) if y is not None else self.model(x, t, ctx)
+        key = (
+            tuple(x.shape), tuple(ctx.shape),
+            None if y is None else tuple(y.shape), x.dtype,
+        )
+        entry = self.graphs.get(key)
+        if entry is None:
+            try:
+                entry = self._capture(x, t, ctx, y)
+            except Exception as exc:  # noqa: BLE001 - fall back to eager
+                import warnings
+
+                warnings.warn(f"hipGraph capture failed, running eager: {exc!r}")
+                self._failed = True
+                return self.model(x, t, ctx, y=y) if y is not None else self.model(x, t, ctx)
+            self.graphs[key] = entry
+        sx, st, sc, sy, sout, g = entry
+        sx.copy_(x)
+        st.copy_(t)
+        sc.copy_(ctx)
+        if y is not None:
+            sy.copy_(y)
+        g.replay()
+        return sout
+
+    def _capture(self, x, t, ctx, y):
+        sx = x.detach().clone()
+        st = t.detach().clone()
+        sc = ctx.detach().clone()
+        sy = y.detach().clone() if y is not None else None
+
+        def fwd():
+            return (self.model(sx, st, sc, y=sy) if sy is not None
+                    else self.model(sx, st, sc))
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                out = fwd()
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            sout = fwd()
+        return (sx, st, sc, sy, sout, g)
+
+
 class CFGDenoiser:
     """eps-model + classifier-free guidance -> denoised prediction x0."""
 
-    def __init__(self, unet, schedule: NoiseSchedule, cond, uncond, cfg_scale: float):
-        self.unet = unet
+    def __init__(self, unet, schedule: NoiseSchedule, cond, uncond, cfg_scale: float,
+                 use_graph: bool = True):
+        if use_graph:
+            # one GraphedModel per underlying module, cached across jobs so
+            # capture cost is paid once per shape
+            wrapper = getattr(unet, "_graphed_wrapper", None)
+            if wrapper is None or wrapper.model is not unet:
+                wrapper = GraphedModel(unet)
+                object.__setattr__(unet, "_graphed_wrapper", wrapper)
+            self.unet = wrapper
+        else:
+            self.unet = unet
         self.schedule = schedule
         self.cond = cond
         self.uncond = uncond

@@ -97,6 +97,9 @@ class DistributedServer:
         r.add_post("/upload/image", self.post_upload_image)
         r.add_get("/distributed/network_info", self.get_network_info)
         r.add_get("/distributed/system_info", self.get_system_info)
+        r.add_post("/distributed/tunnel/start", self.post_tunnel_start)
+        r.add_post("/distributed/tunnel/stop", self.post_tunnel_stop)
+        r.add_get("/distributed/tunnel/status", self.get_tunnel_status)
         r.add_post("/distributed/launch_worker", self.post_launch_worker)
         r.add_post("/distributed/stop_worker", self.post_stop_worker)
         r.add_get("/distributed/managed_workers", self.get_managed_workers)
@@ -403,6 +406,34 @@ class DistributedServer:
             "is_runpod": bool(os.environ.get("RUNPOD_POD_ID")),
             "is_worker": self.is_worker,
         })
+
+    # ---- tunnel ------------------------------------------------------------
+
+    @property
+    def tunnel(self):
+        if not hasattr(self, "_tunnel"):
+            from .tunnel import TunnelManager
+
+            self._tunnel = TunnelManager()
+        return self._tunnel
+
+    async def post_tunnel_start(self, request):
+        from ..utils.errors import TunnelError
+
+        data = await request.json()
+        port = int(data.get("port", load_config()["master"].get("port", 8188)))
+        try:
+            url = await self.tunnel.start(port)
+        except TunnelError as exc:
+            return _err(str(exc), status=500)
+        return web.json_response({"status": "started", "url": url})
+
+    async def post_tunnel_stop(self, request):
+        await self.tunnel.stop()
+        return web.json_response({"status": "stopped"})
+
+    async def get_tunnel_status(self, request):
+        return web.json_response(self.tunnel.status())
 
     # ---- worker process management ----------------------------------------
 
