@@ -41,7 +41,28 @@ class VAEResBlock(nn.Module):
         self.conv2 = nn.Conv2d(cout, cout, 3, padding=1)
         self.skip = nn.Conv2d(cin, cout, 1) if cin != cout else nn.Identity()
 
+    def _mfma_ok(self, x):
+        return (
+            x.is_cuda
+            and ops.conv_supported(self.conv1)
+            and ops.conv_supported(self.conv2)
+            and (isinstance(self.skip, nn.Identity) or ops.conv_supported(self.skip))
+        )
+
     def forward(self, x):
+        if self._mfma_ok(x):
+            # hand-written NHWC path: fused GN+SiLU and implicit-GEMM convs,
+            # activations in channels_last storage end to end
+            if not x.is_contiguous(memory_format=torch.channels_last):
+                x = x.contiguous(memory_format=torch.channels_last)
+            h = ops.group_norm_silu_cl(x, self.norm1.groups, self.norm1.weight,
+                                       self.norm1.bias, 1e-5, True)
+            h = ops.conv2d_mfma(h, self.conv1)
+            h = ops.group_norm_silu_cl(h, self.norm2.groups, self.norm2.weight,
+                                       self.norm2.bias, 1e-5, True)
+            h = ops.conv2d_mfma(h, self.conv2)
+            skip = x if isinstance(self.skip, nn.Identity) else ops.conv2d_mfma(x, self.skip)
+            return h + skip
         h = self.conv2(self.norm2(self.conv1(self.norm1(x))))
         return h + self.skip(x)
 
@@ -169,6 +190,8 @@ class VAE(nn.Module):
     def encode(self, images: torch.Tensor) -> torch.Tensor:
         p = next(self.parameters())
         x = images.permute(0, 3, 1, 2).to(p.device, p.dtype) * 2.0 - 1.0
+        if x.is_cuda:
+            x = x.contiguous(memory_format=torch.channels_last)
         moments = self.encoder(x)
         mean = moments[:, : self.cfg.latent_channels]
         return mean * self.cfg.scale_factor
@@ -176,6 +199,9 @@ class VAE(nn.Module):
     def decode(self, latents: torch.Tensor) -> torch.Tensor:
         p = next(self.parameters())
         z = latents.to(p.device, p.dtype) / self.cfg.scale_factor
+        if z.is_cuda:
+            # channels_last so the decoder trunk runs the NHWC MFMA kernels
+            z = z.contiguous(memory_format=torch.channels_last)
         x = self.decoder(z)
         img = (x.float() + 1.0) / 2.0
         return img.clamp(0, 1).permute(0, 2, 3, 1)

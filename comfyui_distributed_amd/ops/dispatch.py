@@ -42,6 +42,12 @@ def group_norm_silu(
     silu: bool = True,
 ) -> torch.Tensor:
     if _on_gpu(x):
+        if (
+            x.dim() == 4
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and not x.is_contiguous()
+        ):
+            return group_norm_silu_cl(x, groups, weight, bias, eps, silu)
         return ext.get_ext(True).group_norm_fused(
             x.to(torch.bfloat16), groups, weight, bias, eps, silu
         )
@@ -67,6 +73,67 @@ def act_mul(a: torch.Tensor, b: torch.Tensor, gelu: bool = False) -> torch.Tenso
     bf = b.float()
     act = F.gelu(bf, approximate="tanh") if gelu else F.silu(bf)
     return (a.float() * act).to(a.dtype)
+
+
+# ---------------------------------------------------------------------------
+# channels-last conv path (hand-written implicit-GEMM MFMA kernels)
+# ---------------------------------------------------------------------------
+
+
+def conv_supported(conv) -> bool:
+    """True when the MFMA conv kernel covers this nn.Conv2d."""
+    k = conv.kernel_size
+    return (
+        conv.stride == (1, 1)
+        and conv.in_channels % 32 == 0
+        and conv.out_channels % 16 == 0
+        and ((k == (3, 3) and conv.padding == (1, 1))
+             or (k == (1, 1) and conv.padding == (0, 0)))
+        and conv.dilation == (1, 1)
+        and conv.groups == 1
+    )
+
+
+def _repacked_weight(conv) -> torch.Tensor:
+    """[K, R*S*C] bf16, (r,s,c) with c innermost — cached on the module."""
+    wt = getattr(conv, "_distgpu_wt", None)
+    if wt is None or wt.device != conv.weight.device:
+        w = conv.weight.detach()  # [K, C, R, S]
+        wt = w.permute(0, 2, 3, 1).reshape(w.shape[0], -1).contiguous()
+        wt = wt.to(torch.bfloat16)
+        conv._distgpu_wt = wt
+    return wt
+
+
+def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
+    """x: NCHW tensor in channels_last memory format (bf16, on GPU) ->
+    same layout. Dispatches to the implicit-GEMM NHWC kernel."""
+    assert x.is_cuda
+    b, c, h, w = x.shape
+    nhwc = x.permute(0, 2, 3, 1)  # view: contiguous when x is channels_last
+    if not nhwc.is_contiguous():
+        nhwc = nhwc.contiguous()
+    wt = _repacked_weight(conv)
+    rs = 9 if conv.kernel_size == (3, 3) else 1
+    bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
+    y = ext.get_ext(True).conv_nhwc(
+        nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
+        fuse_silu,
+    )
+    return y.permute(0, 3, 1, 2)  # NCHW semantic, channels_last storage
+
+
+def group_norm_silu_cl(x: torch.Tensor, groups: int, weight, bias,
+                       eps: float = 1e-5, silu: bool = True) -> torch.Tensor:
+    """GroupNorm(+SiLU) for channels_last NCHW tensors on GPU."""
+    assert x.is_cuda
+    nhwc = x.permute(0, 2, 3, 1)
+    if not nhwc.is_contiguous():
+        nhwc = nhwc.contiguous()
+    y = ext.get_ext(True).group_norm_nhwc(
+        nhwc.to(torch.bfloat16), groups, weight, bias, eps, silu
+    )
+    return y.permute(0, 3, 1, 2)
 
 
 # ---------------------------------------------------------------------------

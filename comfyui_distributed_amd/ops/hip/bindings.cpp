@@ -16,6 +16,12 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        int64_t heads, int64_t kv_heads, int64_t nk_real,
                        double scale);
 torch::Tensor mfma_selftest(torch::Tensor a, torch::Tensor b);
+torch::Tensor conv_nhwc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
+                        int64_t B, int64_t H, int64_t W, int64_t C, int64_t K,
+                        int64_t rs, bool fuse_silu);
+torch::Tensor group_norm_nhwc(torch::Tensor x, int64_t groups,
+                              torch::Tensor weight, torch::Tensor bias,
+                              double eps, bool fuse_silu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_fused", &group_norm_fused,
@@ -29,4 +35,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "flash attention forward, bf16 MFMA");
   m.def("mfma_selftest", &mfma_selftest,
         "single-wave 16x16x32 bf16 MFMA with the kernel fragment layouts");
+  m.def("conv_nhwc", &conv_nhwc,
+        "implicit-GEMM 3x3/1x1 NHWC bf16 conv on MFMA (+fused SiLU)");
+  m.def("group_norm_nhwc", &group_norm_nhwc,
+        "fused GroupNorm(+SiLU), NHWC bf16");
 }

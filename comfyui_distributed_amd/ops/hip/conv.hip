@@ -1,0 +1,278 @@
+// Hand-written CDNA4 convolution kernels for the VAE decode path.
+//
+// Implicit-GEMM formulation on MFMA 16x16x32 bf16: output pixels are GEMM
+// rows (M = B*H*W), output channels are GEMM columns (N = K), the reduction
+// runs over (r, s, c) with the channel dim innermost (NHWC activations, so
+// every A fragment is a contiguous 16-byte load; no im2col materialization,
+// no NCHW<->NHWC transposes). Weights are host-repacked once per module to
+// W_t[k_out][(r*3+s)*C + c] so B fragments are contiguous too.
+//
+// Covers: 3x3 stride-1 pad-1 and 1x1 convs with C % 32 == 0, K % 16 == 0 —
+// the whole VAE decoder trunk (512/256/128 channels). Odd-channel edges
+// (conv_in from 4 latents, conv_out to RGB) stay on MIOpen.
+//
+// Reference counterpart: the VAEDecode ComfyUI op the reference calls per
+// tile.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// Tile geometry: BM=64 pixels x BN=64 channels per workgroup, 4 waves in a
+// 2x2 grid (each wave owns a 32x32 sub-tile = 2x2 MFMA fragments).
+#define CBM 64
+#define CBN 64
+
+template <int RS>  // 9 for 3x3, 1 for 1x1
+__global__ __launch_bounds__(256) void conv_nhwc_kernel(
+    const uint16_t* __restrict__ x,   // [B, H, W, C]
+    const uint16_t* __restrict__ wt,  // [K, RS*C] repacked
+    const float* __restrict__ bias,   // [K] or nullptr
+    uint16_t* __restrict__ y,         // [B, H, W, K]
+    int B, int H, int W, int C, int K, int fuse_silu) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1;          // wave row (0..1)
+  const int wn = wave & 1;           // wave col (0..1)
+  const long long HW = (long long)H * W;
+  const long long M = (long long)B * HW;
+
+  const long long m_base = (long long)blockIdx.x * CBM + wm * 32;
+  const int n_base = blockIdx.y * CBN + wn * 32;
+
+  // accumulators: 2x2 fragments of 16x16
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // per-lane A row (pixel) and B row (output channel)
+  const int fr = lane & 15;          // fragment row/col index
+  const int kgrp = lane >> 4;        // k-group: elements kgrp*8..+8
+
+  // decompose the two pixel rows this lane loads for A
+  long long m_row[2];
+  int px_y[2], px_x[2], px_b[2];
+  bool m_ok[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    m_row[i] = m_base + i * 16 + fr;
+    m_ok[i] = m_row[i] < M;
+    long long mm = m_ok[i] ? m_row[i] : 0;
+    px_b[i] = (int)(mm / HW);
+    int rem = (int)(mm - (long long)px_b[i] * HW);
+    px_y[i] = rem / W;
+    px_x[i] = rem % W;
+  }
+  const int n_col[2] = {n_base + fr, n_base + 16 + fr};
+
+  const int csteps = C / 32;
+  for (int rs = 0; rs < RS; ++rs) {
+    const int r = RS == 1 ? 0 : rs / 3;
+    const int s = RS == 1 ? 0 : rs % 3;
+    // source pixel offsets for this tap (pad 1 for 3x3)
+    int sy[2], sx[2];
+    bool tap_ok[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      sy[i] = px_y[i] + (RS == 1 ? 0 : r - 1);
+      sx[i] = px_x[i] + (RS == 1 ? 0 : s - 1);
+      tap_ok[i] = m_ok[i] && sy[i] >= 0 && sy[i] < H && sx[i] >= 0 && sx[i] < W;
+    }
+    for (int cs = 0; cs < csteps; ++cs) {
+      const int c0 = cs * 32 + kgrp * 8;
+      // A fragments (2 pixel rows x 8 contiguous channels)
+      short8 a[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        if (tap_ok[i]) {
+          const uint16_t* p = x + (((long long)px_b[i] * H + sy[i]) * W + sx[i]) * C + c0;
+          a[i] = *reinterpret_cast<const short8*>(p);
+        } else {
+          a[i] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+      }
+      // B fragments (2 output channels x 8 contiguous k-elements)
+      const int kdim = rs * C + cs * 32 + kgrp * 8;
+      short8 bfr[2];
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int n = n_col[j];
+        bfr[j] = (n < K)
+            ? *reinterpret_cast<const short8*>(wt + (long long)n * (RS * C) + kdim)
+            : short8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], bfr[j],
+                                                              acc[i][j], 0, 0, 0);
+    }
+  }
+
+  // epilogue: bias + optional SiLU, store NHWC
+  const int out_col = lane & 15;
+  const int rgrp = lane >> 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int n = n_base + j * 16 + out_col;
+      if (n >= K) continue;
+      const float bval = bias ? bias[n] : 0.f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const long long m = m_base + i * 16 + rgrp * 4 + rr;
+        if (m >= M) continue;
+        float v = acc[i][j][rr] + bval;
+        if (fuse_silu) v = silu_f(v);
+        y[m * K + n] = f32_to_bf16_bits(v);
+      }
+    }
+  }
+}
+
+torch::Tensor conv_nhwc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
+                        int64_t B, int64_t H, int64_t W, int64_t C, int64_t K,
+                        int64_t rs, bool fuse_silu) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(wt.is_cuda() && wt.scalar_type() == at::kBFloat16 && wt.is_contiguous());
+  TORCH_CHECK(C % 32 == 0, "C must be a multiple of 32");
+  TORCH_CHECK(rs == 9 || rs == 1, "only 3x3 and 1x1");
+  auto y = torch::empty({B, H, W, K}, x.options());
+  const long long M = B * H * W;
+  dim3 grid((unsigned)((M + CBM - 1) / CBM), (unsigned)((K + CBN - 1) / CBN));
+  dim3 block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+  const float* bptr = nullptr;
+  torch::Tensor bf;
+  if (bias.defined() && bias.numel() > 0) {
+    bf = bias.contiguous().to(at::kFloat);
+    bptr = bf.data_ptr<float>();
+  }
+  if (rs == 9)
+    hipLaunchKernelGGL((conv_nhwc_kernel<9>), grid, block, 0, stream,
+                       (const uint16_t*)x.data_ptr(), (const uint16_t*)wt.data_ptr(),
+                       bptr, (uint16_t*)y.data_ptr(), (int)B, (int)H, (int)W,
+                       (int)C, (int)K, fuse_silu ? 1 : 0);
+  else
+    hipLaunchKernelGGL((conv_nhwc_kernel<1>), grid, block, 0, stream,
+                       (const uint16_t*)x.data_ptr(), (const uint16_t*)wt.data_ptr(),
+                       bptr, (uint16_t*)y.data_ptr(), (int)B, (int)H, (int)W,
+                       (int)C, (int)K, fuse_silu ? 1 : 0);
+  HIP_CHECK_LAUNCH();
+  return y;
+}
+
+// ---------------------------------------------------------------------------
+// NHWC fused GroupNorm(+SiLU): one block per (n, g); channels of a group are
+// contiguous within each pixel (span Cg*2 bytes).
+// ---------------------------------------------------------------------------
+
+template <bool FUSE_SILU>
+__global__ void groupnorm_nhwc_kernel(const uint16_t* __restrict__ x,
+                                      uint16_t* __restrict__ y,
+                                      const float* __restrict__ weight,
+                                      const float* __restrict__ bias,
+                                      int C, int G, long long HW, float eps) {
+  const int n = blockIdx.x / G;
+  const int g = blockIdx.x % G;
+  const int Cg = C / G;
+  const long long base = (long long)n * HW * C + (long long)g * Cg;
+
+  __shared__ float scratch[16];
+  __shared__ float s_mean, s_rstd;
+
+  float sum = 0.f, sumsq = 0.f;
+  // each thread walks pixels with stride blockDim; within a pixel reads the
+  // group's Cg contiguous channels (vector 8 where possible)
+  for (long long p = threadIdx.x; p < HW; p += blockDim.x) {
+    const uint16_t* px = x + base + p * C;
+    int c = 0;
+    for (; c + 7 < Cg; c += 8) {
+      ushort8_t v = *reinterpret_cast<const ushort8_t*>(px + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_bits_to_f32(v[j]);
+        sum += f;
+        sumsq += f * f;
+      }
+    }
+    for (; c < Cg; ++c) {
+      float f = bf16_bits_to_f32(px[c]);
+      sum += f;
+      sumsq += f * f;
+    }
+  }
+  sum = block_reduce(sum, scratch, SumOp{}, 0.f);
+  sumsq = block_reduce(sumsq, scratch, SumOp{}, 0.f);
+  if (threadIdx.x == 0) {
+    const float count = (float)HW * Cg;
+    float mean = sum / count;
+    float var = sumsq / count - mean * mean;
+    s_mean = mean;
+    s_rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+  }
+  __syncthreads();
+  const float mean = s_mean, rstd = s_rstd;
+
+  for (long long p = threadIdx.x; p < HW; p += blockDim.x) {
+    const uint16_t* px = x + base + p * C;
+    uint16_t* py = y + base + p * C;
+    int c = 0;
+    for (; c + 7 < Cg; c += 8) {
+      ushort8_t v = *reinterpret_cast<const ushort8_t*>(px + c);
+      ushort8_t o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int ch = g * Cg + c + j;
+        float f = (bf16_bits_to_f32(v[j]) - mean) * rstd;
+        f = f * weight[ch] + bias[ch];
+        if (FUSE_SILU) f = silu_f(f);
+        o[j] = f32_to_bf16_bits(f);
+      }
+      *reinterpret_cast<ushort8_t*>(py + c) = o;
+    }
+    for (; c < Cg; ++c) {
+      const int ch = g * Cg + c;
+      float f = (bf16_bits_to_f32(px[c]) - mean) * rstd;
+      f = f * weight[ch] + bias[ch];
+      if (FUSE_SILU) f = silu_f(f);
+      py[c] = f32_to_bf16_bits(f);
+    }
+  }
+}
+
+torch::Tensor group_norm_nhwc(torch::Tensor x, int64_t groups,
+                              torch::Tensor weight, torch::Tensor bias,
+                              double eps, bool fuse_silu) {
+  // x: [B, H, W, C] bf16 contiguous
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kBFloat16
+              && x.is_contiguous());
+  const int B = x.size(0), C = x.size(3);
+  const long long HW = (long long)x.size(1) * x.size(2);
+  TORCH_CHECK(C % groups == 0);
+  auto w = weight.contiguous().to(at::kFloat);
+  auto b = bias.contiguous().to(at::kFloat);
+  auto y = torch::empty_like(x);
+  dim3 grid(B * (int)groups);
+  dim3 block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (fuse_silu)
+    hipLaunchKernelGGL((groupnorm_nhwc_kernel<true>), grid, block, 0, stream,
+                       (const uint16_t*)x.data_ptr(), (uint16_t*)y.data_ptr(),
+                       w.data_ptr<float>(), b.data_ptr<float>(), C,
+                       (int)groups, HW, (float)eps);
+  else
+    hipLaunchKernelGGL((groupnorm_nhwc_kernel<false>), grid, block, 0, stream,
+                       (const uint16_t*)x.data_ptr(), (uint16_t*)y.data_ptr(),
+                       w.data_ptr<float>(), b.data_ptr<float>(), C,
+                       (int)groups, HW, (float)eps);
+  HIP_CHECK_LAUNCH();
+  return y;
+}

@@ -1,0 +1,85 @@
+"""GPU numerics for the implicit-GEMM NHWC conv + NHWC GroupNorm kernels."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def extmod():
+    from comfyui_distributed_amd.ops import ext
+
+    return ext.get_ext(required=True)
+
+
+@pytest.mark.parametrize("shape,k", [
+    ((2, 64, 17, 19), 32),     # odd spatial, C=64
+    ((1, 128, 32, 32), 128),
+    ((1, 512, 16, 16), 256),   # channel change
+])
+def test_conv3x3_matches_torch(extmod, shape, k):
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(shape[1])
+    b, c, h, w = shape
+    conv = torch.nn.Conv2d(c, k, 3, padding=1).cuda().to(torch.bfloat16)
+    x = (torch.randn(b, c, h, w) / 4).cuda().to(torch.bfloat16)
+    xcl = x.contiguous(memory_format=torch.channels_last)
+    y = dispatch.conv2d_mfma(xcl, conv).float()
+    ref = F.conv2d(x.float(), conv.weight.float(), conv.bias.float(), padding=1)
+    err = (y - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err / scale < 0.05, f"rel err {err/scale}"
+
+
+def test_conv1x1_matches_torch(extmod):
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(0)
+    conv = torch.nn.Conv2d(128, 256, 1).cuda().to(torch.bfloat16)
+    x = (torch.randn(2, 128, 24, 24) / 4).cuda().to(torch.bfloat16)
+    xcl = x.contiguous(memory_format=torch.channels_last)
+    y = dispatch.conv2d_mfma(xcl, conv).float()
+    ref = F.conv2d(x.float(), conv.weight.float(), conv.bias.float())
+    err = (y - ref).abs().max().item()
+    assert err / ref.abs().max().item() < 0.05
+
+
+def test_conv_fused_silu(extmod):
+    from comfyui_distributed_amd.ops import dispatch
+
+    conv = torch.nn.Conv2d(64, 64, 3, padding=1).cuda().to(torch.bfloat16)
+    x = (torch.randn(1, 64, 16, 16) / 4).cuda().to(torch.bfloat16)
+    xcl = x.contiguous(memory_format=torch.channels_last)
+    y = dispatch.conv2d_mfma(xcl, conv, fuse_silu=True).float()
+    ref = F.silu(F.conv2d(x.float(), conv.weight.float(), conv.bias.float(), padding=1))
+    assert (y - ref).abs().max().item() / (ref.abs().max().item() + 1e-6) < 0.05
+
+
+def test_groupnorm_nhwc_matches_torch(extmod):
+    torch.manual_seed(1)
+    x = torch.randn(2, 128, 20, 20)
+    w, b = torch.randn(128), torch.randn(128)
+    nhwc = x.permute(0, 2, 3, 1).contiguous().cuda().to(torch.bfloat16)
+    y = extmod.group_norm_nhwc(nhwc, 32, w.cuda(), b.cuda(), 1e-5, True)
+    ref = F.silu(F.group_norm(x, 32, w, b, 1e-5)).permute(0, 2, 3, 1)
+    err = (y.float().cpu() - ref).abs()
+    assert (err <= 0.02 + 0.01 * ref.abs()).all(), err.max().item()
+
+
+def test_vae_decode_nhwc_path_consistent(extmod):
+    """GPU VAE decode (MFMA conv path) vs CPU fp32 decode: bf16-level
+    agreement proves the hand-written conv path computes the same model."""
+    from comfyui_distributed_amd.models import create_diffusion_stack
+
+    gpu = create_diffusion_stack("sd15", device="cuda:0", dtype=torch.bfloat16, seed=3)
+    cpu = create_diffusion_stack("sd15", device="cpu", dtype=torch.float32, seed=3)
+    z = torch.randn(1, 4, 16, 16)
+    with torch.no_grad():
+        img_gpu = gpu.vae.decode(z.cuda()).cpu()
+        img_cpu = cpu.vae.decode(z)
+    err = (img_gpu - img_cpu).abs().max().item()
+    assert err < 0.12, f"decode mismatch {err}"  # bf16 conv chain tolerance
+    assert torch.isfinite(img_gpu).all()
