@@ -170,7 +170,7 @@ def attention(
         dp = _dpad_for(d)
         nq, nk = q.shape[1], k.shape[1]
         nq_p = (nq + 63) // 64 * 64
-        nk_p = (nk + 31) // 32 * 32
+        nk_p = (nk + 63) // 64 * 64  # kernel K-tile is 64 keys
         qb = q.to(torch.bfloat16)
         kb = k.to(torch.bfloat16)
         vb = v.to(torch.bfloat16)

@@ -261,7 +261,8 @@ torch::Tensor group_norm_nhwc(torch::Tensor x, int64_t groups,
   auto b = bias.contiguous().to(at::kFloat);
   auto y = torch::empty_like(x);
   dim3 grid(B * (int)groups);
-  dim3 block(256);
+  const long long per_group = (long long)(C / groups) * HW;
+  dim3 block(B * groups >= 128 ? 256 : (per_group > (1 << 20) ? 1024 : 256));
   auto stream = at::hip::getCurrentHIPStream();
   if (fuse_silu)
     hipLaunchKernelGGL((groupnorm_nhwc_kernel<true>), grid, block, 0, stream,

@@ -106,7 +106,10 @@ torch::Tensor group_norm_fused(torch::Tensor x, int64_t groups,
   TORCH_CHECK(C % groups == 0, "C must divide groups");
   auto y = torch::empty_like(xc);
   dim3 grid(N * groups);
-  dim3 block(256);
+  // few groups + huge spatial (VAE decode output) starves the chip at 256
+  // threads x few blocks; scale the block up instead
+  const long long per_group = (long long)(C / groups) * HW;
+  dim3 block(N * groups >= 128 ? 256 : (per_group > (1 << 20) ? 1024 : 256));
   auto stream = at::hip::getCurrentHIPStream();
   if (fuse_silu)
     hipLaunchKernelGGL((groupnorm_kernel<true>), grid, block, 0, stream,

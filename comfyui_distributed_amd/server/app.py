@@ -103,6 +103,9 @@ class DistributedServer:
         r.add_post("/distributed/launch_worker", self.post_launch_worker)
         r.add_post("/distributed/stop_worker", self.post_stop_worker)
         r.add_get("/distributed/managed_workers", self.get_managed_workers)
+        r.add_get("/distributed/worker_log", self.get_worker_log)
+        r.add_post("/interrupt", self.post_interrupt)
+        r.add_get("/distributed/ws", self.ws_handler)
         app.on_startup.append(self._on_startup)
         return app
 
@@ -467,6 +470,61 @@ class DistributedServer:
             out[wid] = {"pid": handle.pid, "alive": handle.poll() is None}
         return web.json_response({"managed": out})
 
+    async def get_worker_log(self, request):
+        """Tail a managed worker's log file (reference
+        worker_routes.py:606-646 behavior: efficient tail-seek)."""
+        from .workers import worker_log_path
+
+        wid = request.query.get("id", "")
+        lines = int(request.query.get("lines", "100"))
+        path = worker_log_path(wid)
+        if not path.exists():
+            return _err("no log for worker", status=404)
+        with open(path, "rb") as fh:
+            fh.seek(0, os.SEEK_END)
+            size = fh.tell()
+            fh.seek(max(0, size - 64 * 1024))
+            tail = fh.read().decode(errors="replace").splitlines()[-lines:]
+        return web.json_response({"log": "\n".join(tail)})
+
+    async def post_interrupt(self, request):
+        """User interrupt: flags the node runtime so every wait loop raises
+        (reference checks comfy.model_management interrupts in each loop)."""
+        from ..nodes.runtime import get_runtime
+
+        get_runtime().interrupt()
+        return web.json_response({"status": "interrupted"})
+
+    async def ws_handler(self, request):
+        """WebSocket orchestration endpoint: accepts dispatch_prompt
+        messages, replies dispatch_ack with the request id (reference
+        worker_routes.py:43-112 + dispatch.py:62-95)."""
+        import json as _json
+
+        ws = web.WebSocketResponse()
+        await ws.prepare(request)
+        async for msg in ws:
+            if msg.type != web.WSMsgType.TEXT:
+                continue
+            try:
+                data = _json.loads(msg.data)
+            except ValueError:
+                continue
+            if data.get("type") == "dispatch_prompt":
+                rid = data.get("request_id")
+                try:
+                    await self.enqueue_local(data.get("prompt") or {},
+                                             data.get("client_id", ""))
+                    await ws.send_json({"type": "dispatch_ack",
+                                        "request_id": rid, "ok": True})
+                except PromptValidationError as exc:
+                    await ws.send_json({"type": "dispatch_ack",
+                                        "request_id": rid, "ok": False,
+                                        "error": str(exc)})
+            elif data.get("type") == "ping":
+                await ws.send_json({"type": "pong"})
+        return ws
+
 
 def main():
     import argparse
@@ -484,6 +542,45 @@ def main():
     server = DistributedServer(is_worker=args.worker, device=device)
     app = server.build_app()
     role = "worker" if args.worker else "master"
+
+    if not args.worker:
+        # adopt/clean managed workers from a previous run, then auto-launch
+        # (reference workers/startup.py: delayed timer + signal/atexit)
+        from .workers import adopt_or_cleanup_managed, launch_worker, stop_worker
+
+        adopted = adopt_or_cleanup_managed()
+        if adopted:
+            log(f"re-adopted managed workers: {adopted}")
+        cfg = load_config()
+        if cfg["settings"].get("auto_launch_workers"):
+            import threading
+
+            def delayed():
+                for w in cfg.get("workers", []):
+                    if w.get("enabled") and w.get("type", "local") == "local" \
+                            and not w.get("host"):
+                        try:
+                            server.managed[str(w["id"])] = launch_worker(w)
+                        except Exception as exc:  # noqa: BLE001
+                            log(f"auto-launch of {w.get('id')} failed: {exc}")
+
+            threading.Timer(2.0, delayed).start()
+
+        import atexit
+        import signal as _signal
+
+        def cleanup(*_a):
+            if load_config()["settings"].get("stop_workers_on_master_exit", True):
+                for wid, handle in list(server.managed.items()):
+                    stop_worker(handle, wid)
+
+        atexit.register(cleanup)
+        for sig in (_signal.SIGTERM, _signal.SIGINT, _signal.SIGHUP):
+            try:
+                _signal.signal(sig, lambda *_a: (cleanup(), os._exit(0)))
+            except (ValueError, OSError):
+                pass
+
     log(f"starting {role} server on {args.listen}:{args.port} (device={device})")
     web.run_app(app, host=args.listen, port=args.port, print=None)
 

@@ -50,11 +50,39 @@ def select_least_busy(candidates: list[tuple[str, dict | None]]) -> str:
 
 
 async def dispatch_worker_prompt(worker: dict, prompt: dict, client_id: str,
-                                 timeout: float = 30.0) -> bool:
-    """POST /prompt to a worker (reference dispatch.py:98-141)."""
+                                 timeout: float = 30.0,
+                                 use_websocket: bool | None = None) -> bool:
+    """Dispatch to a worker: WS dispatch_prompt/dispatch_ack when
+    websocket_orchestration is on (reference dispatch.py:62-95), falling
+    back to POST /prompt (:98-141)."""
+    import secrets
+
     import aiohttp
 
     url = network.build_worker_url(worker)
+    if use_websocket is None:
+        use_websocket = bool(
+            load_config().get("settings", {}).get("websocket_orchestration", True)
+        )
+    if use_websocket:
+        try:
+            session = await network.get_client_session()
+            rid = secrets.token_hex(8)
+            async with session.ws_connect(
+                f"{url}/distributed/ws",
+                timeout=aiohttp.ClientWSTimeout(ws_close=timeout),
+            ) as ws:
+                await ws.send_json({"type": "dispatch_prompt", "request_id": rid,
+                                    "prompt": prompt, "client_id": client_id})
+                async for msg in ws:
+                    if msg.type != aiohttp.WSMsgType.TEXT:
+                        break
+                    data = msg.json()
+                    if (data.get("type") == "dispatch_ack"
+                            and data.get("request_id") == rid):
+                        return bool(data.get("ok"))
+        except Exception as exc:  # noqa: BLE001
+            log(f"WS dispatch to {url} failed ({exc}); falling back to POST")
     try:
         session = await network.get_client_session()
         async with session.post(

@@ -48,7 +48,7 @@ def build_launch_command(worker: dict) -> list[str]:
     return cmd
 
 
-def launch_worker(worker: dict) -> subprocess.Popen:
+def launch_worker(worker: dict, monitor: bool = True) -> subprocess.Popen:
     wid = str(worker.get("id"))
     env = dict(os.environ)
     env["HIP_VISIBLE_DEVICES"] = str(worker.get("cuda_device", 0))
@@ -57,6 +57,12 @@ def launch_worker(worker: dict) -> subprocess.Popen:
     env["DISTGPU_MASTER_PID"] = str(os.getpid())
     env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     cmd = build_launch_command(worker)
+    if monitor:
+        # wrap in the watchdog so orphaned workers die with the master
+        # (reference lifecycle.py:67-76)
+        cmd = [sys.executable, "-m",
+               "comfyui_distributed_amd.server.worker_monitor",
+               "--master-pid", str(os.getpid()), "--", *cmd]
     logf = open(worker_log_path(wid), "ab")
     proc = subprocess.Popen(
         cmd, env=env, stdout=logf, stderr=subprocess.STDOUT,
