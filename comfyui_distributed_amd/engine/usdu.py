@@ -24,6 +24,7 @@ import torch
 from ..models.sampling import CFGDenoiser, NoiseSchedule, sample
 from ..ops import dispatch as ops
 from ..utils import usdu_math
+from ..utils.trace import trace_range
 
 
 @dataclass
@@ -101,22 +102,25 @@ def sample_tiles(
         batch_img = torch.cat(crops, dim=0)
         # ---- encode -> img2img sample -> decode ----
         with torch.no_grad():
-            latents = stack.vae.encode(batch_img)
+            with trace_range("usdu.vae_encode"):
+                latents = stack.vae.encode(batch_img)
             noise = torch.stack(
                 [
                     _tile_noise(params.seed, t, b + batch_offset, latents.shape[1:])
                     for t, b in chunk
                 ]
             ).to(latents.device)
-            latent_out = sample(
-                denoiser,
-                noise,
-                sigmas,
-                sampler=params.sampler_name,
-                seed=params.seed,
-                start_from_latent=latents.float(),
-            )
-            out_img = stack.vae.decode(latent_out.to(stack.dtype))
+            with trace_range("usdu.sample"):
+                latent_out = sample(
+                    denoiser,
+                    noise,
+                    sigmas,
+                    sampler=params.sampler_name,
+                    seed=params.seed,
+                    start_from_latent=latents.float(),
+                )
+            with trace_range("usdu.vae_decode"):
+                out_img = stack.vae.decode(latent_out.to(stack.dtype))
         for j, (t, b) in enumerate(chunk):
             results[(t, b)] = out_img[j : j + 1].float()
     return results

@@ -210,7 +210,12 @@ class DistributedServer:
     async def post_submit_tiles(self, request):
         from .usdu_http import decode_tile_submission
 
-        data = await request.json()
+        if request.content_type.startswith("multipart/"):
+            # reference wire format: PNG file parts + tiles_metadata JSON
+            # (upscale/payload_parsers.py:7-64)
+            data = await self._parse_multipart_tiles(request)
+        else:
+            data = await request.json()
         job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
         if job is None:
             return _err("unknown job", status=404)
@@ -221,6 +226,37 @@ class DistributedServer:
             job.finished_workers.add(str(data.get("worker_id", "")))
         job.worker_status[str(data.get("worker_id", ""))] = time.time()
         return web.json_response({"status": "ok", "received": len(items)})
+
+    async def _parse_multipart_tiles(self, request) -> dict:
+        import base64
+        import json as _json
+
+        reader = await request.multipart()
+        fields: dict = {}
+        blobs: list[bytes] = []
+        while True:
+            part = await reader.next()
+            if part is None:
+                break
+            if part.filename:
+                blobs.append(await part.read())
+            else:
+                fields[part.name] = (await part.read()).decode()
+        meta = _json.loads(fields.get("tiles_metadata", "[]"))
+        tiles = []
+        for i, m in enumerate(meta):
+            if i < len(blobs):
+                tiles.append({
+                    "tile_idx": m.get("tile_idx", m.get("global_idx", i)),
+                    "batch_idx": m.get("batch_idx", 0),
+                    "image": base64.b64encode(blobs[i]).decode("ascii"),
+                })
+        return {
+            "job_id": fields.get("job_id", ""),
+            "worker_id": fields.get("worker_id", ""),
+            "is_last": fields.get("is_last", "").lower() in ("1", "true"),
+            "tiles": tiles,
+        }
 
     async def post_submit_image(self, request):
         from ..utils.image import decode_png_base64
