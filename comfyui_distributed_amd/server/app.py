@@ -75,6 +75,8 @@ class DistributedServer:
     def build_app(self) -> web.Application:
         app = web.Application(client_max_size=constants.MAX_PAYLOAD_SIZE + 2**20)
         r = app.router
+        r.add_get("/", self.get_panel)
+        r.add_get("/panel", self.get_panel)
         r.add_get("/prompt", self.get_prompt)
         r.add_post("/prompt", self.post_prompt)
         r.add_post("/distributed/queue", self.post_queue)
@@ -138,6 +140,12 @@ class DistributedServer:
         return self.prompt_queue.qsize() + (1 if self.executing else 0)
 
     # -------------------------------------------------------------- routes
+
+    async def get_panel(self, request):
+        from pathlib import Path
+
+        path = Path(__file__).parent / "static" / "panel.html"
+        return web.Response(text=path.read_text(), content_type="text/html")
 
     async def get_prompt(self, request):
         return web.json_response(

@@ -78,6 +78,34 @@ def bench_gn(iters):
               f"eager {te*1e3:7.3f} ms ({te/t:4.1f}x)")
 
 
+def bench_conv(iters):
+    print("== conv3x3 bf16: ours (NHWC MFMA) vs MIOpen NCHW vs MIOpen CL ==")
+    from comfyui_distributed_amd.ops import dispatch
+
+    shapes = [
+        # (B, C, H, W, K) — UNet/VAE hot shapes at tile batch 16
+        (16, 320, 68, 68, 320),
+        (16, 640, 34, 34, 640),
+        (16, 1280, 17, 17, 1280),
+        (1, 512, 136, 136, 512),
+        (1, 256, 544, 544, 256),
+        (1, 128, 1088, 1088, 128),
+    ]
+    for b, c, h, w, k in shapes:
+        conv = torch.nn.Conv2d(c, k, 3, padding=1).cuda().to(torch.bfloat16)
+        x = (torch.randn(b, c, h, w) / 4).cuda().to(torch.bfloat16)
+        xcl = x.contiguous(memory_format=torch.channels_last)
+        conv_cl = torch.nn.Conv2d(c, k, 3, padding=1).cuda().to(torch.bfloat16) \
+            .to(memory_format=torch.channels_last)
+        t_ours = timeit(lambda: dispatch.conv2d_mfma(xcl, conv), iters)
+        t_nchw = timeit(lambda: conv(x), iters)
+        t_cl = timeit(lambda: conv_cl(xcl), iters)
+        flops = 2 * b * h * w * k * c * 9
+        print(f"B{b} C{c} {h}x{w} K{k}: ours {t_ours*1e3:7.3f} ms "
+              f"({flops/t_ours/1e12:6.1f} TF) | miopen-nchw {t_nchw*1e3:7.3f} "
+              f"| miopen-cl {t_cl*1e3:7.3f}")
+
+
 def bench_tiles(iters):
     print("== tile ops (f32) ==")
     from comfyui_distributed_amd.ops import ext
@@ -103,6 +131,8 @@ def main():
         bench_attn(args.iters)
     if args.op in ("gn", "all"):
         bench_gn(args.iters)
+    if args.op in ("conv", "all"):
+        bench_conv(args.iters)
     if args.op in ("tiles", "all"):
         bench_tiles(args.iters)
 
