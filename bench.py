@@ -30,11 +30,23 @@ from comfyui_distributed_amd.parallel.usdu_dist import run_distributed_usdu
 from comfyui_distributed_amd.ops import dispatch as ops
 
 
+CONFIG_PRESETS = {
+    # BASELINE.json configs (3 is the flagship / default)
+    "usdu-sd15-4k": {},
+    "gen-sdxl": {"model": "sdxl"},
+    "usdu-sdxl-8k": {"model": "sdxl", "src_size": 2048, "scale": 4,
+                     "tile": 1024},
+    "wan-t2v": {"model": "wan14b"},
+}
+
+
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--config", default="usdu-sd15-4k",
+                    choices=sorted(CONFIG_PRESETS))
     ap.add_argument("--model", default="sd15")
     ap.add_argument("--src-size", type=int, default=1024)
     ap.add_argument("--scale", type=int, default=4)
@@ -43,7 +55,90 @@ def parse_args():
     ap.add_argument("--denoise", type=float, default=0.5)
     ap.add_argument("--cfg", type=float, default=8.0)
     ap.add_argument("--tile-batch", type=int, default=8)
-    return ap.parse_args()
+    args = ap.parse_args()
+    for key, value in CONFIG_PRESETS[args.config].items():
+        if getattr(args, key) == ap.get_default(key):
+            setattr(args, key, value)
+    return args
+
+
+def bench_generation(args, ctx, stack, cond, uncond):
+    """Seed-parallel generation benches (BASELINE configs 2 and 5)."""
+    import torch.distributed as dist
+
+    from comfyui_distributed_amd.parallel.collector import seed_parallel_generate
+
+    world = ctx.world_size
+    if args.config == "wan-t2v":
+        from comfyui_distributed_amd.models.video import (
+            VideoGenParams,
+            generate_video,
+        )
+
+        p = VideoGenParams(seed=3, steps=args.sampler_steps, cfg=args.cfg,
+                           width=480, height=480, frames=17)
+
+        def one_step(_s):
+            # every rank generates its own seed-offset clip; frames gathered
+            from dataclasses import replace
+
+            from comfyui_distributed_amd.parallel.collector import seed_for_rank
+            from comfyui_distributed_amd.parallel.dist import gather_tensor_lists
+
+            local = generate_video(
+                stack, cond, uncond, replace(p, seed=seed_for_rank(p.seed, ctx.rank))
+            )
+            tensors = [local[i] for i in range(local.shape[0])]
+            meta = [(ctx.rank, i) for i in range(local.shape[0])]
+            gather_tensor_lists(ctx, tensors, meta)
+
+        unit, per_step = "frames/s", 17 * world
+        model_name = "wan2.2-14B-dit-random-init"
+        cfg_extra = {"frames": 17, "size": "480x480",
+                     "sampler": f"flow-euler/{args.sampler_steps}steps/cfg{args.cfg}"}
+    else:
+        from comfyui_distributed_amd.engine.generate import GenParams
+
+        p = GenParams(seed=3, steps=args.sampler_steps, cfg=args.cfg,
+                      width=1024, height=1024, batch_size=1)
+
+        def one_step(_s):
+            seed_parallel_generate(ctx, stack, cond, uncond, p)
+
+        unit, per_step = "images/s", world
+        model_name = "sdxl-unet-2.6B-random-init"
+        cfg_extra = {"size": "1024x1024",
+                     "sampler": f"euler/{args.sampler_steps}steps/cfg{args.cfg}"}
+
+    for w in range(args.warmup):
+        one_step(-1 - w)
+    ctx.barrier()
+    ctx.sync_device()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        one_step(s)
+    ctx.barrier()
+    ctx.sync_device()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        dev = ctx.device if ctx.backend == "nccl" else torch.device("cpu")
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    if ctx.is_master:
+        print(json.dumps({
+            "metric": f"{args.config} whole-node throughput",
+            "value": round(per_step * args.steps / elapsed, 3),
+            "unit": unit, "n_gpus": world, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 1),
+            "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+            "dtype": "bf16" if ctx.device.type == "cuda" else "fp32",
+            "data": "synthetic",
+            "config": {"model": model_name, "global_batch": world,
+                       "parallelism": f"seed-parallel dp{world} + RCCL collector",
+                       **cfg_extra},
+        }))
 
 
 def main():
@@ -57,6 +152,10 @@ def main():
     stack = create_diffusion_stack(args.model, device=device, dtype=dtype, seed=0)
     cond = stack.make_conditioning(0)
     uncond = stack.make_conditioning(1)
+
+    if args.config in ("gen-sdxl", "wan-t2v"):
+        bench_generation(args, ctx, stack, cond, uncond)
+        return
 
     params = USDUParams(
         seed=7,

@@ -169,8 +169,8 @@ def attention(
     if _on_gpu(q):
         dp = _dpad_for(d)
         nq, nk = q.shape[1], k.shape[1]
-        nq_p = (nq + 63) // 64 * 64
-        nk_p = (nk + 63) // 64 * 64  # kernel K-tile is 64 keys
+        nq_p = (nq + 127) // 128 * 128  # kernel block = 128 q rows
+        nk_p = (nk + 63) // 64 * 64     # kernel K-tile = 64 keys
         qb = q.to(torch.bfloat16)
         kb = k.to(torch.bfloat16)
         vb = v.to(torch.bfloat16)

@@ -1,23 +1,24 @@
 #include "hip/hip_runtime.h"
 // Flash-attention forward for diffusion UNet/DiT blocks (MI355X/gfx950).
 //
-// Hand-written CDNA4 kernel: MFMA 16x16x32 bf16 tiles, online softmax with
-// in-register row state, V staged transposed through LDS (conflict-free
-// padded layout), K fragments streamed from global (the K tile is
-// L2-resident: the grid is ordered so every workgroup of one (batch, head)
-// lands on the same XCD — blockIdx.x = bh and the dispatcher places block b
-// on XCD b%8, so with BH % 8 == 0 a head's K/V stays in one XCD's L2).
+// Hand-written CDNA4 kernel, v3. Structure per 64-key tile:
+//   * 8-wave workgroup owns 128 q rows (wave w rows [16w, 16w+16)), so the
+//     shared V^T stage is amortized over 8 waves and each SIMD carries 2+
+//     waves of this kernel for latency overlap;
+//   * K fragments are double-buffered in registers: tile t+1's eight
+//     16-byte K loads are issued before tile t's PV phase (async-stage
+//     split — HBM latency hides under MFMA work);
+//   * online softmax with defer-max (skip the O-rescale and the m update
+//     while the tile max stays within DEFER_THR of the running max; the
+//     exp inputs stay bounded by e^DEFER_THR which fp32 accumulation
+//     tolerates);
+//   * epilogue divides replaced by one reciprocal per row.
 //
-// Contract (enforced by the Python wrapper ops/attention.py):
-//   q      [BH,  Nq_pad, D_PAD]  bf16, Nq_pad % 64 == 0
-//   k, v   [BHk, Nk_pad, D_PAD]  bf16, Nk_pad % 64 == 0, zero-padded
-//   o      [BH,  Nq_pad, D_PAD]  bf16 (written)
-//   D_PAD in {64, 96, 128, 160}; real D zero-padded up; real Nk passed for
-//   the softmax mask. scale applied to scores. gqa = H / H_kv.
-//
-// Geometry: 4-wave workgroup owns 64 q rows (wave w rows [16w,16w+16));
-// K-tile = 64 keys = four 16x16 score fragments per wave; two barriers per
-// 64-key tile.
+// Contract (enforced by ops/attention.py): q [BH, Nq_pad, D_PAD] bf16 with
+// Nq_pad % 128 == 0; k/v [BHk, Nk_pad, D_PAD] with Nk_pad % 64 == 0,
+// zero-padded; D_PAD in {64, 96, 128, 160}; real Nk masks padded keys;
+// grid.x = bh so one head's K/V stays on one XCD's L2 (dispatcher places
+// block b on XCD b%8).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -27,22 +28,24 @@ typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define QROWS_PER_WAVE 16
-#define QROWS_PER_BLOCK 64
-#define KT 64            // keys per tile
-#define KFRAG (KT / 16)  // score fragments per wave per tile
-#define VT_PITCH (KT + 8)  // V_T row pitch: 144 B rows -> conflict-free b128
+#define NWAVES 8
+#define QROWS_PER_BLOCK (QROWS_PER_WAVE * NWAVES)  // 128
+#define KT 64
+#define KFRAG (KT / 16)
+#define VT_PITCH (KT + 8)
 #define PT_PITCH (KT + 8)
+#define DEFER_THR 8.0f
 
 template <int D_PAD>
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+__global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ o, int Nq_pad,
     int Nk_pad, int Nk, int H, int Hkv, float scale) {
-  constexpr int DK = D_PAD / 32;   // QK^T k-steps
-  constexpr int DN = D_PAD / 16;   // O column fragments
+  constexpr int DK = D_PAD / 32;
+  constexpr int DN = D_PAD / 16;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int bh = blockIdx.x;        // b * H + h  (XCD affinity: bh % 8)
+  const int bh = blockIdx.x;
   const int h = bh % H;
   const int bhk = (bh / H) * Hkv + h / (H / Hkv);
   const int q_row0 = blockIdx.y * QROWS_PER_BLOCK + wave * QROWS_PER_WAVE;
@@ -52,9 +55,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const uint16_t* vbase = v + (long long)bhk * Nk_pad * D_PAD;
 
   __shared__ __align__(16) uint16_t v_t[D_PAD][VT_PITCH];
-  __shared__ __align__(16) uint16_t p_lds[4][QROWS_PER_WAVE][PT_PITCH];
+  __shared__ __align__(16) uint16_t p_lds[NWAVES][QROWS_PER_WAVE][PT_PITCH];
 
-  // ---- Q fragments resident in registers --------------------------------
+  // ---- Q fragments resident ---------------------------------------------
   short8 qfrag[DK];
   {
     const int row = lane & 15;
@@ -73,12 +76,28 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   for (int n = 0; n < DN; ++n) oacc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = Nk_pad / KT;
+  const int kd0 = (lane >> 4) * 8;
+
+  // K fragment double buffer: kf[kk][f] for the CURRENT tile.
+  short8 kf[DK][KFRAG];
+  auto load_kfrags = [&](int key0, short8 dst[DK][KFRAG]) {
+#pragma unroll
+    for (int kk = 0; kk < DK; ++kk)
+#pragma unroll
+      for (int f = 0; f < KFRAG; ++f) {
+        const int key = key0 + f * 16 + (lane & 15);
+        dst[kk][f] = *reinterpret_cast<const short8*>(
+            kbase + (long long)key * D_PAD + kk * 32 + kd0);
+      }
+  };
+  load_kfrags(0, kf);
+
   for (int t = 0; t < ntiles; ++t) {
     const int key0 = t * KT;
 
-    // ---- stage V^T cooperatively: v_t[d][key] = V[key0+key][d] ----------
+    // ---- stage V^T cooperatively ----------------------------------------
     __syncthreads();
-    for (int c = threadIdx.x; c < KT * (D_PAD / 8); c += 256) {
+    for (int c = threadIdx.x; c < KT * (D_PAD / 8); c += NWAVES * 64) {
       const int key = c & (KT - 1);
       const int d0 = (c / KT) * 8;
       short8 vv = *reinterpret_cast<const short8*>(
@@ -88,23 +107,20 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- scores: KFRAG fragments of 16 keys -----------------------------
+    // ---- scores from the prefetched K fragments --------------------------
     f32x4 s[KFRAG];
 #pragma unroll
     for (int f = 0; f < KFRAG; ++f) s[f] = f32x4{0.f, 0.f, 0.f, 0.f};
-    {
-      const int d0 = (lane >> 4) * 8;
 #pragma unroll
-      for (int kk = 0; kk < DK; ++kk) {
+    for (int kk = 0; kk < DK; ++kk)
 #pragma unroll
-        for (int f = 0; f < KFRAG; ++f) {
-          const int key = key0 + f * 16 + (lane & 15);
-          short8 kf = *reinterpret_cast<const short8*>(
-              kbase + (long long)key * D_PAD + kk * 32 + d0);
-          s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], kf, s[f], 0, 0, 0);
-        }
-      }
-    }
+      for (int f = 0; f < KFRAG; ++f)
+        s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], kf[kk][f],
+                                                       s[f], 0, 0, 0);
+
+    // ---- prefetch next tile's K while softmax runs -----------------------
+    if (t + 1 < ntiles) load_kfrags(key0 + KT, kf);
+
 #pragma unroll
     for (int f = 0; f < KFRAG; ++f) {
       const int key = key0 + f * 16 + (lane & 15);
@@ -113,36 +129,50 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         s[f][r] = (key < Nk) ? s[f][r] * scale : -1e30f;
     }
 
-    // ---- online softmax --------------------------------------------------
-    float p[KFRAG][4], alpha[4];
+    // ---- online softmax with defer-max -----------------------------------
+    float p[KFRAG][4];
+    float mt[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float mt = s[0][r];
+      float m = s[0][r];
 #pragma unroll
-      for (int f = 1; f < KFRAG; ++f) mt = fmaxf(mt, s[f][r]);
+      for (int f = 1; f < KFRAG; ++f) m = fmaxf(m, s[f][r]);
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
-        mt = fmaxf(mt, __shfl_xor(mt, off, 64));
-      const float m_new = fmaxf(m_run[r], mt);
-      alpha[r] = __expf(m_run[r] - m_new);
-      m_run[r] = m_new;
+        m = fmaxf(m, __shfl_xor(m, off, 64));
+      mt[r] = m;
+    }
+    // wave-uniform defer decision: skip rescale while every row's tile max
+    // stays within DEFER_THR of its running max (T13; P bounded by e^THR)
+    bool need = false;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) need |= (mt[r] - m_run[r]) > DEFER_THR;
+    if (__ballot(need) != 0ull) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float m_new = fmaxf(m_run[r], mt[r]);
+        const float a = __expf(m_run[r] - m_new);
+        m_run[r] = m_new;
+        l_run[r] *= a;
+#pragma unroll
+        for (int n = 0; n < DN; ++n) oacc[n][r] *= a;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
       float rowsum = 0.f;
 #pragma unroll
       for (int f = 0; f < KFRAG; ++f) {
-        p[f][r] = __expf(s[f][r] - m_new);
+        p[f][r] = __expf(s[f][r] - m_run[r]);
         rowsum += p[f][r];
       }
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
         rowsum += __shfl_xor(rowsum, off, 64);
-      l_run[r] = l_run[r] * alpha[r] + rowsum;
+      l_run[r] += rowsum;
     }
-#pragma unroll
-    for (int n = 0; n < DN; ++n)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) oacc[n][r] *= alpha[r];
 
-    // ---- P -> LDS (wave-private; compiler orders ds_write->ds_read) -----
+    // ---- P -> LDS (wave-private) -----------------------------------------
     {
       const int col = lane & 15;
       const int rg = lane >> 4;
@@ -153,7 +183,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
           p_lds[wave][rg * 4 + r][f * 16 + col] = f32_to_bf16_bits(p[f][r]);
     }
 
-    // ---- PV: O += P[16 x KT] @ V[KT x D_PAD] ----------------------------
+    // ---- PV ----------------------------------------------------------------
     {
       const int prow = lane & 15;
       const int pk0 = (lane >> 4) * 8;
@@ -172,18 +202,20 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     }
   }
 
-  // ---- epilogue ---------------------------------------------------------
+  // ---- epilogue: one reciprocal per row ---------------------------------
   {
     const int col = lane & 15;
     const int rg = lane >> 4;
+    float rl[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) rl[r] = __builtin_amdgcn_rcpf(l_run[r]);
 #pragma unroll
     for (int n = 0; n < DN; ++n) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = rg * 4 + r;
-        const float val = oacc[n][r] / l_run[r];
         o[((long long)bh * Nq_pad + q_row0 + row) * D_PAD + n * 16 + col] =
-            f32_to_bf16_bits(val);
+            f32_to_bf16_bits(oacc[n][r] * rl[r]);
       }
     }
   }
@@ -200,17 +232,16 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(D == 64 || D == 96 || D == 128 || D == 160,
               "D_PAD must be one of 64/96/128/160, got ", D);
   const int Nq_pad = q.size(1), Nk_pad = k.size(1);
-  TORCH_CHECK(Nq_pad % 64 == 0 && Nk_pad % KT == 0,
-              "pad Nq to 64 and Nk to ", KT);
+  TORCH_CHECK(Nq_pad % QROWS_PER_BLOCK == 0 && Nk_pad % KT == 0,
+              "pad Nq to ", QROWS_PER_BLOCK, " and Nk to ", KT);
   TORCH_CHECK(k.size(2) == D && v.size(2) == D);
   const int BH = q.size(0);
   TORCH_CHECK(BH % heads == 0, "BH must divide heads");
   TORCH_CHECK(heads % kv_heads == 0, "GQA ratio must be integral");
 
   auto o = torch::empty_like(q);
-  // blockIdx.x = bh -> XCD affinity per head (dispatcher: XCD = block % 8)
   dim3 grid(BH, Nq_pad / QROWS_PER_BLOCK);
-  dim3 block(256);
+  dim3 block(NWAVES * 64);
   auto stream = at::hip::getCurrentHIPStream();
 #define LAUNCH_D(DP)                                                          \
   hipLaunchKernelGGL((attn_fwd_kernel<DP>), grid, block, 0, stream,           \

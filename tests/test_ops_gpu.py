@@ -68,6 +68,29 @@ def test_attention_gqa_gpu(extmod):
     assert (out - ref).abs().max().item() < 0.03
 
 
+def test_attention_defer_max_rescale_branch(extmod):
+    """Force the defer-max rescale branch (guide T13 hazard): a spiked K row
+    in a LATER tile makes the running max jump past the threshold. Results
+    must still match the exact CPU softmax."""
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(9)
+    bh, n, d = 2, 256, 64
+    q = torch.randn(bh, n, d) / 4
+    k = torch.randn(bh, n, d) / 4
+    v = torch.randn(bh, n, d)
+    # tile size is 64: spike keys in tiles 2 and 3 so m jumps mid-stream
+    k[:, 140] = q[:, 17] * 40.0  # huge dot for row 17 at tile 2
+    k[:, 200] = q[:, 33] * 60.0  # even bigger at tile 3
+    out = dispatch.attention(
+        q.cuda().to(torch.bfloat16), k.cuda().to(torch.bfloat16),
+        v.cuda().to(torch.bfloat16), heads=1,
+    ).float().cpu()
+    ref = dispatch.attention(q, k, v, heads=1)
+    err = (out - ref).abs().max().item()
+    assert err < 0.05, f"defer-max branch mismatch {err}"
+
+
 def test_attention_softmax_rows_sum(extmod):
     """Uniform V exposes softmax normalization errors: out must equal V."""
     from comfyui_distributed_amd.ops import dispatch
