@@ -222,10 +222,13 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    model_names = {"sd15": "sd15-unet-860M-random-init",
+                   "sdxl": "sdxl-unet-2.6B-random-init"}
     if ctx.is_master:
         value = tiles_per_step * args.steps / elapsed
         result = {
-            "metric": "tiles/sec (whole node), SD1.5 USDU 4x->4K 512px tiles",
+            "metric": f"tiles/sec (whole node), {args.model} USDU "
+                      f"{args.scale}x->{canvas_size} {args.tile}px tiles",
             "value": round(value, 3),
             "unit": "tiles/s",
             "n_gpus": world,
@@ -238,7 +241,7 @@ def main():
             "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "sd15-unet-860M-random-init",
+                "model": model_names.get(args.model, args.model + "-random-init"),
                 "global_batch": world,
                 "canvas": f"{canvas_size}x{canvas_size}",
                 "tile": f"{args.tile}+pad32->544",
