@@ -57,6 +57,10 @@ class DiffusionStack:
         torch.manual_seed(seed)
         self.unet = UNetModel(cfg.unet).to(device=device, dtype=dtype).eval()
         self.vae = VAE(cfg.vae).to(device=device, dtype=dtype).eval()
+        if torch.device(device).type == "cuda":
+            # conv weights in channels_last so MIOpen runs pure-NHWC paths
+            self.unet = self.unet.to(memory_format=torch.channels_last)
+            self.vae = self.vae.to(memory_format=torch.channels_last)
         self.schedule = sampling.NoiseSchedule()
         self.device = torch.device(device)
         self.dtype = dtype

@@ -294,6 +294,11 @@ class UNetModel(nn.Module):
     def forward(self, x, timesteps, context, y=None):
         dtype = self.out_conv.weight.dtype
         x = x.to(dtype)
+        if x.is_cuda:
+            # channels_last end to end: MIOpen picks NHWC igemm kernels with
+            # no batched_transpose pre/post passes, and the fused GroupNorm
+            # dispatches to its NHWC variant (profiles/r01: transposes ~5%)
+            x = x.contiguous(memory_format=torch.channels_last)
         context = context.to(dtype)
         if y is not None:
             y = y.to(dtype)
