@@ -170,6 +170,40 @@ class VAEEncode(_ContextNode):
         return ({"samples": z.cpu()},)
 
 
+class WanVideoGenerate(_ContextNode):
+    """t2v generation with the WAN stack; outputs frames as an IMAGE batch
+    (frames along the batch dim, so batch dividers segment them)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {
+            "required": {
+                "model": ("MODEL",),
+                "positive": ("CONDITIONING",),
+                "seed": ("INT", {"default": 0}),
+                "steps": ("INT", {"default": 20}),
+                "cfg": ("FLOAT", {"default": 5.0}),
+                "width": ("INT", {"default": 480}),
+                "height": ("INT", {"default": 480}),
+                "frames": ("INT", {"default": 17}),
+            },
+            "optional": {"negative": ("CONDITIONING",)},
+        }
+
+    RETURN_TYPES = ("IMAGE",)
+    FUNCTION = "generate"
+    CATEGORY = "video"
+
+    def generate(self, model, positive, seed, steps, cfg, width, height,
+                 frames, negative=None):
+        from ..models.video import VideoGenParams, generate_video
+
+        p = VideoGenParams(seed=int(seed), steps=int(steps), cfg=float(cfg),
+                           width=int(width), height=int(height),
+                           frames=int(frames))
+        return (generate_video(model, positive, negative, p).cpu(),)
+
+
 class LoadImage(_ContextNode):
     """Loads an image from the input directory (or a synthetic one when the
     name is "synthetic:<W>x<H>")."""
@@ -247,4 +281,5 @@ BUILTIN_CLASS_MAPPINGS = {
     "LoadImage": LoadImage,
     "SaveImage": SaveImage,
     "PreviewImage": PreviewImage,
+    "WanVideoGenerate": WanVideoGenerate,
 }
