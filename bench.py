@@ -244,7 +244,7 @@ def main():
                 "model": model_names.get(args.model, args.model + "-random-init"),
                 "global_batch": world,
                 "canvas": f"{canvas_size}x{canvas_size}",
-                "tile": f"{args.tile}+pad32->544",
+                "tile": f"{args.tile}+pad32->{plans[0].process_size[0]}",
                 "tiles_per_image": len(plans),
                 "sampler": f"euler/{args.sampler_steps}steps/denoise{args.denoise}/cfg{args.cfg}",
                 "parallelism": f"tile-pull-queue dp{world} over RCCL/xGMI",
