@@ -117,17 +117,12 @@ class CrossAttention(nn.Module):
 
     def forward(self, x, context=None):
         context = x if context is None else context
-        b, n, _ = x.shape
-        nk = context.shape[1]
-        q = self.to_q(x).reshape(b, n, self.heads, self.head_dim)
-        k = self.to_k(context).reshape(b, nk, self.heads, self.head_dim)
-        v = self.to_v(context).reshape(b, nk, self.heads, self.head_dim)
-        # pack heads: [B*H, N, D]
-        q = q.permute(0, 2, 1, 3).reshape(b * self.heads, n, self.head_dim)
-        k = k.permute(0, 2, 1, 3).reshape(b * self.heads, nk, self.head_dim)
-        v = v.permute(0, 2, 1, 3).reshape(b * self.heads, nk, self.head_dim)
-        o = ops.attention(q, k, v, heads=self.heads)
-        o = o.reshape(b, self.heads, n, self.head_dim).permute(0, 2, 1, 3).reshape(b, n, -1)
+        # packed layout [B, N, H*D] straight from the projections — the
+        # strided kernel consumes it with zero reshapes/copies
+        o = ops.attention_packed(
+            self.to_q(x), self.to_k(context), self.to_v(context),
+            heads=self.heads,
+        )
         return self.to_out(o)
 
 
@@ -180,6 +175,20 @@ class SpatialTransformer(nn.Module):
             t = blk(t, context)
         t = self.proj_out(t)
         return residual + t.reshape(b, h, w, c).permute(0, 3, 1, 2)
+
+
+class StemConv(nn.Module):
+    """conv_in (C<=8): dedicated HIP kernel on GPU channels_last inputs
+    (MIOpen/CK have no fast NHWC solver for tiny C — see profiles/)."""
+
+    def __init__(self, cin: int, cout: int):
+        super().__init__()
+        self.conv = nn.Conv2d(cin, cout, 3, padding=1)
+
+    def forward(self, x):
+        if x.is_cuda and x.is_contiguous(memory_format=torch.channels_last):
+            return ops.conv2d_smallc(x, self.conv)
+        return self.conv(x)
 
 
 class Downsample(nn.Module):
@@ -243,7 +252,7 @@ class UNetModel(nn.Module):
             return cfg.num_heads, ch // cfg.num_heads
 
         self.input_blocks = nn.ModuleList(
-            [_TimestepSequential([nn.Conv2d(cfg.in_channels, ch0, 3, padding=1)])]
+            [_TimestepSequential([StemConv(cfg.in_channels, ch0)])]
         )
         skip_chans = [ch0]
         ch = ch0

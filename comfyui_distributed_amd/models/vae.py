@@ -86,20 +86,8 @@ class VAEAttention(nn.Module):
         b, c, h, w = x.shape
         t = self.norm(x).permute(0, 2, 3, 1).reshape(b, h * w, c)
         q, k, v = self.qkv(t).chunk(3, dim=-1)
-
-        def split(u):
-            return (
-                u.reshape(b, h * w, self.heads, self.head_dim)
-                .permute(0, 2, 1, 3)
-                .reshape(b * self.heads, h * w, self.head_dim)
-            )
-
-        o = ops.attention(split(q), split(k), split(v), heads=self.heads)
-        o = (
-            o.reshape(b, self.heads, h * w, self.head_dim)
-            .permute(0, 2, 1, 3)
-            .reshape(b, h * w, c)
-        )
+        o = ops.attention_packed(q.contiguous(), k.contiguous(), v.contiguous(),
+                                 heads=self.heads)
         return x + self.proj(o).reshape(b, h, w, c).permute(0, 3, 1, 2)
 
 

@@ -1,5 +1,6 @@
 from .dispatch import (
     attention,
+    attention_packed,
     act_mul,
     blend_tile,
     extract_resize,

@@ -123,6 +123,24 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
     return y.permute(0, 3, 1, 2)  # NCHW semantic, channels_last storage
 
 
+def conv2d_smallc(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
+    """Stem conv (in_channels <= 8) on the dedicated HIP kernel — MIOpen/CK
+    fall back to very slow paths for NHWC C=4. x: channels_last NCHW."""
+    assert x.is_cuda and conv.in_channels <= 8
+    b, c, h, w = x.shape
+    nhwc = x.permute(0, 2, 3, 1)
+    if not nhwc.is_contiguous():
+        nhwc = nhwc.contiguous()
+    wt = _repacked_weight(conv)
+    rs = 9 if conv.kernel_size == (3, 3) else 1
+    bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
+    y = ext.get_ext(True).conv_smallc(
+        nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
+        fuse_silu,
+    )
+    return y.permute(0, 3, 1, 2)
+
+
 def group_norm_silu_cl(x: torch.Tensor, groups: int, weight, bias,
                        eps: float = 1e-5, silu: bool = True) -> torch.Tensor:
     """GroupNorm(+SiLU) for channels_last NCHW tensors on GPU."""
@@ -167,23 +185,14 @@ def attention(
     d = q.shape[-1]
     scale = scale if scale is not None else 1.0 / math.sqrt(d)
     if _on_gpu(q):
-        dp = _dpad_for(d)
-        nq, nk = q.shape[1], k.shape[1]
-        nq_p = (nq + 127) // 128 * 128  # kernel block = 128 q rows
-        nk_p = (nk + 63) // 64 * 64     # kernel K-tile = 64 keys
-        qb = q.to(torch.bfloat16)
-        kb = k.to(torch.bfloat16)
-        vb = v.to(torch.bfloat16)
-        if dp != d or nq_p != nq:
-            qb = F.pad(qb, (0, dp - d, 0, nq_p - nq))
-        if dp != d or nk_p != nk:
-            kb = F.pad(kb, (0, dp - d, 0, nk_p - nk))
-            vb = F.pad(vb, (0, dp - d, 0, nk_p - nk))
+        _dpad_for(d)  # validates the supported range
         o = ext.get_ext(True).attn_fwd(
-            qb.contiguous(), kb.contiguous(), vb.contiguous(), heads, kv_heads,
-            nk, scale,
+            q.to(torch.bfloat16).contiguous(),
+            k.to(torch.bfloat16).contiguous(),
+            v.to(torch.bfloat16).contiguous(),
+            heads, kv_heads, k.shape[1], scale,
         )
-        return o[:, :nq, :d]
+        return o
     # CPU reference: fp32 math, GQA by repeating kv heads.
     qf, kf, vf = q.float(), k.float(), v.float()
     if kv_heads != heads:
@@ -196,6 +205,41 @@ def attention(
     s = torch.bmm(qf, kf.transpose(1, 2)) * scale
     p = torch.softmax(s, dim=-1)
     return torch.bmm(p, vf).to(q.dtype)
+
+
+def attention_packed(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    heads: int,
+    kv_heads: int | None = None,
+    scale: float | None = None,
+) -> torch.Tensor:
+    """Attention on the packed projection layout: q [B, Nq, H*D],
+    k/v [B, Nk, Hkv*D] -> [B, Nq, H*D]. On GPU this feeds the strided
+    kernel directly (zero reshapes/copies); on CPU it unpacks and reuses
+    the reference path."""
+    kv_heads = kv_heads or heads
+    d = q.shape[-1] // heads
+    scale = scale if scale is not None else 1.0 / math.sqrt(d)
+    if _on_gpu(q):
+        return ext.get_ext(True).attn_fwd_packed(
+            q.to(torch.bfloat16).contiguous(),
+            k.to(torch.bfloat16).contiguous(),
+            v.to(torch.bfloat16).contiguous(),
+            heads, kv_heads, scale,
+        )
+    b, nq, _ = q.shape
+    nk = k.shape[1]
+
+    def split(u, h):
+        return (u.reshape(b, -1, h, d).permute(0, 2, 1, 3)
+                .reshape(b * h, -1, d))
+
+    o = attention(split(q, heads), split(k, kv_heads), split(v, kv_heads),
+                  heads=heads, kv_heads=kv_heads, scale=scale)
+    return (o.reshape(b, heads, nq, d).permute(0, 2, 1, 3)
+            .reshape(b, nq, heads * d))
 
 
 # ---------------------------------------------------------------------------

@@ -1,23 +1,18 @@
 // Flash-attention forward for diffusion UNet/DiT blocks (MI355X/gfx950).
 //
-// Hand-written CDNA4 kernel, v3. Structure per 64-key tile:
-//   * 8-wave workgroup owns 128 q rows (wave w rows [16w, 16w+16)), so the
-//     shared V^T stage is amortized over 8 waves and each SIMD carries 2+
-//     waves of this kernel for latency overlap;
-//   * K fragments are double-buffered in registers: tile t+1's eight
-//     16-byte K loads are issued before tile t's PV phase (async-stage
-//     split — HBM latency hides under MFMA work);
-//   * online softmax with defer-max (skip the O-rescale and the m update
-//     while the tile max stays within DEFER_THR of the running max; the
-//     exp inputs stay bounded by e^DEFER_THR which fp32 accumulation
-//     tolerates);
-//   * epilogue divides replaced by one reciprocal per row.
+// Hand-written CDNA4 kernel, v4: generalized strided addressing so the
+// kernel consumes the QKV projections' natural packed layout
+// [B, N, H*D] directly — no pad, no permute, no copies on the host side.
+// The head dim D only needs to be a multiple of 8: fragment loads are
+// masked at 8-element granularity against the padded compute width D_PAD.
 //
-// Contract (enforced by ops/attention.py): q [BH, Nq_pad, D_PAD] bf16 with
-// Nq_pad % 128 == 0; k/v [BHk, Nk_pad, D_PAD] with Nk_pad % 64 == 0,
-// zero-padded; D_PAD in {64, 96, 128, 160}; real Nk masks padded keys;
-// grid.x = bh so one head's K/V stays on one XCD's L2 (dispatcher places
-// block b on XCD b%8).
+// Structure per 64-key tile (see profiles/ for measurements):
+//   * 8-wave workgroup owns 128 q rows; shared V^T stage amortized over
+//     all 8 waves; grid.x = b*H + h so one head's K/V stays on one XCD's
+//     L2 (dispatcher places block b on XCD b%8);
+//   * K fragments double-buffered in registers (tile t+1's loads issue
+//     before tile t's PV phase);
+//   * online softmax with defer-max (T13) and a reciprocal epilogue.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -35,36 +30,47 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define PT_PITCH (KT + 8)
 #define DEFER_THR 8.0f
 
+#define ZERO8 short8{0, 0, 0, 0, 0, 0, 0, 0}
+
 template <int D_PAD>
 __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
-    const uint16_t* __restrict__ v, uint16_t* __restrict__ o, int Nq_pad,
-    int Nk_pad, int Nk, int H, int Hkv, float scale) {
+    const uint16_t* __restrict__ v, uint16_t* __restrict__ o, int Nq, int Nk,
+    int D, int H, int Hkv, float scale, long long q_bstride,
+    long long q_hstride, long long q_rstride, long long k_bstride,
+    long long k_hstride, long long k_rstride) {
   constexpr int DK = D_PAD / 32;
   constexpr int DN = D_PAD / 16;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int bh = blockIdx.x;
+  const int b = bh / H;
   const int h = bh % H;
-  const int bhk = (bh / H) * Hkv + h / (H / Hkv);
+  const int hk = h / (H / Hkv);
   const int q_row0 = blockIdx.y * QROWS_PER_BLOCK + wave * QROWS_PER_WAVE;
 
-  const uint16_t* qbase = q + ((long long)bh * Nq_pad + q_row0) * D_PAD;
-  const uint16_t* kbase = k + (long long)bhk * Nk_pad * D_PAD;
-  const uint16_t* vbase = v + (long long)bhk * Nk_pad * D_PAD;
+  const uint16_t* qbase = q + b * q_bstride + h * q_hstride;
+  const uint16_t* kbase = k + b * k_bstride + hk * k_hstride;
+  const uint16_t* vbase = v + b * k_bstride + hk * k_hstride;
+  uint16_t* obase = o + b * q_bstride + h * q_hstride;
 
   __shared__ __align__(16) uint16_t v_t[D_PAD][VT_PITCH];
   __shared__ __align__(16) uint16_t p_lds[NWAVES][QROWS_PER_WAVE][PT_PITCH];
 
-  // ---- Q fragments resident ---------------------------------------------
+  const int kd0 = (lane >> 4) * 8;
+
+  // ---- Q fragments resident (masked rows / masked d) --------------------
   short8 qfrag[DK];
   {
-    const int row = lane & 15;
-    const int d0 = (lane >> 4) * 8;
+    const int row = q_row0 + (lane & 15);
+    const bool row_ok = row < Nq;
 #pragma unroll
-    for (int kk = 0; kk < DK; ++kk)
-      qfrag[kk] = *reinterpret_cast<const short8*>(
-          qbase + (long long)row * D_PAD + kk * 32 + d0);
+    for (int kk = 0; kk < DK; ++kk) {
+      const int d = kk * 32 + kd0;
+      qfrag[kk] = (row_ok && d + 8 <= D)
+          ? *reinterpret_cast<const short8*>(qbase + (long long)row * q_rstride + d)
+          : ZERO8;
+    }
   }
 
   float m_run[4], l_run[4];
@@ -74,39 +80,43 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
 #pragma unroll
   for (int n = 0; n < DN; ++n) oacc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int ntiles = Nk_pad / KT;
-  const int kd0 = (lane >> 4) * 8;
+  const int ntiles = (Nk + KT - 1) / KT;
 
-  // K fragment double buffer: kf[kk][f] for the CURRENT tile.
   short8 kf[DK][KFRAG];
   auto load_kfrags = [&](int key0, short8 dst[DK][KFRAG]) {
 #pragma unroll
-    for (int kk = 0; kk < DK; ++kk)
+    for (int kk = 0; kk < DK; ++kk) {
+      const int d = kk * 32 + kd0;
+      const bool d_ok = d + 8 <= D;
 #pragma unroll
       for (int f = 0; f < KFRAG; ++f) {
         const int key = key0 + f * 16 + (lane & 15);
-        dst[kk][f] = *reinterpret_cast<const short8*>(
-            kbase + (long long)key * D_PAD + kk * 32 + kd0);
+        dst[kk][f] = (d_ok && key < Nk)
+            ? *reinterpret_cast<const short8*>(kbase + (long long)key * k_rstride + d)
+            : ZERO8;
       }
+    }
   };
   load_kfrags(0, kf);
 
   for (int t = 0; t < ntiles; ++t) {
     const int key0 = t * KT;
 
-    // ---- stage V^T cooperatively ----------------------------------------
+    // ---- stage V^T cooperatively (masked) --------------------------------
     __syncthreads();
     for (int c = threadIdx.x; c < KT * (D_PAD / 8); c += NWAVES * 64) {
       const int key = c & (KT - 1);
       const int d0 = (c / KT) * 8;
-      short8 vv = *reinterpret_cast<const short8*>(
-          vbase + (long long)(key0 + key) * D_PAD + d0);
+      short8 vv = (key0 + key < Nk && d0 + 8 <= D)
+          ? *reinterpret_cast<const short8*>(
+                vbase + (long long)(key0 + key) * k_rstride + d0)
+          : ZERO8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) v_t[d0 + j][key] = (uint16_t)vv[j];
     }
     __syncthreads();
 
-    // ---- scores from the prefetched K fragments --------------------------
+    // ---- scores ----------------------------------------------------------
     f32x4 s[KFRAG];
 #pragma unroll
     for (int f = 0; f < KFRAG; ++f) s[f] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -117,7 +127,6 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
         s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], kf[kk][f],
                                                        s[f], 0, 0, 0);
 
-    // ---- prefetch next tile's K while softmax runs -----------------------
     if (t + 1 < ntiles) load_kfrags(key0 + KT, kf);
 
 #pragma unroll
@@ -141,8 +150,6 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
         m = fmaxf(m, __shfl_xor(m, off, 64));
       mt[r] = m;
     }
-    // wave-uniform defer decision: skip rescale while every row's tile max
-    // stays within DEFER_THR of its running max (T13; P bounded by e^THR)
     bool need = false;
 #pragma unroll
     for (int r = 0; r < 4; ++r) need |= (mt[r] - m_run[r]) > DEFER_THR;
@@ -182,7 +189,7 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
           p_lds[wave][rg * 4 + r][f * 16 + col] = f32_to_bf16_bits(p[f][r]);
     }
 
-    // ---- PV ----------------------------------------------------------------
+    // ---- PV --------------------------------------------------------------
     {
       const int prow = lane & 15;
       const int pk0 = (lane >> 4) * 8;
@@ -201,7 +208,7 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     }
   }
 
-  // ---- epilogue: one reciprocal per row ---------------------------------
+  // ---- epilogue ---------------------------------------------------------
   {
     const int col = lane & 15;
     const int rg = lane >> 4;
@@ -210,36 +217,29 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     for (int r = 0; r < 4; ++r) rl[r] = __builtin_amdgcn_rcpf(l_run[r]);
 #pragma unroll
     for (int n = 0; n < DN; ++n) {
+      const int d = n * 16 + col;
+      if (d >= D) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = rg * 4 + r;
-        o[((long long)bh * Nq_pad + q_row0 + row) * D_PAD + n * 16 + col] =
+        const int row = q_row0 + rg * 4 + r;
+        if (row >= Nq) continue;
+        obase[(long long)row * q_rstride + d] =
             f32_to_bf16_bits(oacc[n][r] * rl[r]);
       }
     }
   }
 }
 
-torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                       int64_t heads, int64_t kv_heads, int64_t nk_real,
-                       double scale) {
-  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
-  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3,
-              "q/k/v must be [B*H, N, D_PAD]");
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
-  const int D = q.size(2);
-  TORCH_CHECK(D == 64 || D == 96 || D == 128 || D == 160,
-              "D_PAD must be one of 64/96/128/160, got ", D);
-  const int Nq_pad = q.size(1), Nk_pad = k.size(1);
-  TORCH_CHECK(Nq_pad % QROWS_PER_BLOCK == 0 && Nk_pad % KT == 0,
-              "pad Nq to ", QROWS_PER_BLOCK, " and Nk to ", KT);
-  TORCH_CHECK(k.size(2) == D && v.size(2) == D);
-  const int BH = q.size(0);
-  TORCH_CHECK(BH % heads == 0, "BH must divide heads");
-  TORCH_CHECK(heads % kv_heads == 0, "GQA ratio must be integral");
-
+static torch::Tensor launch_attn(const torch::Tensor& q, const torch::Tensor& k,
+                                 const torch::Tensor& v, int H, int Hkv,
+                                 int Nq, int Nk, int D, float scale,
+                                 long long qb, long long qh, long long qr,
+                                 long long kb, long long kh, long long kr,
+                                 int batch) {
   auto o = torch::empty_like(q);
-  dim3 grid(BH, Nq_pad / QROWS_PER_BLOCK);
+  const int dpad = D <= 64 ? 64 : (D <= 96 ? 96 : (D <= 128 ? 128 : 160));
+  TORCH_CHECK(D % 8 == 0 && D <= 160, "head dim must be %8 and <=160, got ", D);
+  dim3 grid(batch * H, (Nq + QROWS_PER_BLOCK - 1) / QROWS_PER_BLOCK);
   dim3 block(NWAVES * 64);
   auto stream = at::hip::getCurrentHIPStream();
 #define LAUNCH_D(DP)                                                          \
@@ -247,9 +247,8 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                      (const uint16_t*)q.data_ptr(),                           \
                      (const uint16_t*)k.data_ptr(),                           \
                      (const uint16_t*)v.data_ptr(), (uint16_t*)o.data_ptr(),  \
-                     Nq_pad, Nk_pad, (int)nk_real, (int)heads, (int)kv_heads, \
-                     (float)scale)
-  switch (D) {
+                     Nq, Nk, D, H, Hkv, scale, qb, qh, qr, kb, kh, kr)
+  switch (dpad) {
     case 64: LAUNCH_D(64); break;
     case 96: LAUNCH_D(96); break;
     case 128: LAUNCH_D(128); break;
@@ -258,4 +257,45 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 #undef LAUNCH_D
   HIP_CHECK_LAUNCH();
   return o;
+}
+
+// Legacy layout: q [B*H, N, D] contiguous (one head per batch row).
+torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                       int64_t heads, int64_t kv_heads, int64_t nk_real,
+                       double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.dim() == 3 && q.is_contiguous() && k.is_contiguous()
+              && v.is_contiguous());
+  const int D = q.size(2);
+  const int Nq = q.size(1), Nk = (int)nk_real;
+  const int BH = q.size(0);
+  TORCH_CHECK(BH % heads == 0);
+  const int batch = BH / (int)heads;
+  // strides in elements for the [B*H, N, D] layout
+  const long long qr = D, qh = (long long)Nq * D,
+                  qb = (long long)heads * Nq * D;
+  const long long kr = D, kh = (long long)k.size(1) * D,
+                  kb = (long long)kv_heads * k.size(1) * D;
+  return launch_attn(q, k, v, (int)heads, (int)kv_heads, Nq, Nk, D,
+                     (float)scale, qb, qh, qr, kb, kh, kr, batch);
+}
+
+// Packed layout: q [B, Nq, H*D], k/v [B, Nk, Hkv*D] — the natural output of
+// fused QKV projections; zero host-side reshapes.
+torch::Tensor attn_fwd_packed(torch::Tensor q, torch::Tensor k,
+                              torch::Tensor v, int64_t heads,
+                              int64_t kv_heads, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.dim() == 3 && q.is_contiguous() && k.is_contiguous()
+              && v.is_contiguous());
+  const int B = q.size(0), Nq = q.size(1), Nk = k.size(1);
+  const int D = q.size(2) / (int)heads;
+  TORCH_CHECK((int)(heads * D) == q.size(2), "q last dim must be H*D");
+  TORCH_CHECK((int)(kv_heads * D) == k.size(2), "k last dim must be Hkv*D");
+  const long long qr = (long long)heads * D, qh = D,
+                  qb = (long long)Nq * heads * D;
+  const long long kr = (long long)kv_heads * D, kh = D,
+                  kb = (long long)Nk * kv_heads * D;
+  return launch_attn(q, k, v, (int)heads, (int)kv_heads, Nq, Nk, D,
+                     (float)scale, qb, qh, qr, kb, kh, kr, B);
 }

@@ -15,6 +15,9 @@ void blend_tile(torch::Tensor canvas, torch::Tensor tile, int64_t x1,
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        int64_t heads, int64_t kv_heads, int64_t nk_real,
                        double scale);
+torch::Tensor attn_fwd_packed(torch::Tensor q, torch::Tensor k,
+                              torch::Tensor v, int64_t heads,
+                              int64_t kv_heads, double scale);
 torch::Tensor mfma_selftest(torch::Tensor a, torch::Tensor b);
 torch::Tensor conv_nhwc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
                         int64_t B, int64_t H, int64_t W, int64_t C, int64_t K,
@@ -22,6 +25,9 @@ torch::Tensor conv_nhwc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
 torch::Tensor group_norm_nhwc(torch::Tensor x, int64_t groups,
                               torch::Tensor weight, torch::Tensor bias,
                               double eps, bool fuse_silu);
+torch::Tensor conv_smallc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
+                          int64_t B, int64_t H, int64_t W, int64_t C,
+                          int64_t K, int64_t rs, bool fuse_silu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_fused", &group_norm_fused,
@@ -33,10 +39,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("blend_tile", &blend_tile,
         "fused resample + blurred-rect mask + composite (in-place canvas)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward, bf16 MFMA");
+  m.def("attn_fwd_packed", &attn_fwd_packed,
+        "flash attention on packed [B,N,H*D] QKV (no host reshapes)");
   m.def("mfma_selftest", &mfma_selftest,
         "single-wave 16x16x32 bf16 MFMA with the kernel fragment layouts");
   m.def("conv_nhwc", &conv_nhwc,
         "implicit-GEMM 3x3/1x1 NHWC bf16 conv on MFMA (+fused SiLU)");
   m.def("group_norm_nhwc", &group_norm_nhwc,
         "fused GroupNorm(+SiLU), NHWC bf16");
+  m.def("conv_smallc", &conv_smallc,
+        "small-C NHWC conv (stem convs, C <= 8)");
 }

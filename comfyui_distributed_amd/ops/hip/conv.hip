@@ -313,6 +313,98 @@ torch::Tensor conv_nhwc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
 }
 
 // ---------------------------------------------------------------------------
+// Small-C conv (the UNet/VAE stem: C_in <= 8, e.g. 4 latents -> 320).
+// MIOpen/CK fall back to very slow paths for NHWC C=4; this is a simple
+// HBM-bound kernel: 32 pixels x 8 k-slots per 256-thread block, weights
+// staged in LDS, per-pixel taps read once and reused across the k-slots
+// via L1.
+// ---------------------------------------------------------------------------
+
+#define SC_KSLOTS 8
+
+template <int RS>
+__global__ __launch_bounds__(256) void conv_smallc_kernel(
+    const uint16_t* __restrict__ x, const uint16_t* __restrict__ wt,
+    const float* __restrict__ bias, uint16_t* __restrict__ y, int B, int H,
+    int W, int C, int K, int fuse_silu) {
+  extern __shared__ __align__(16) uint16_t w_sm[];  // [K][RS*C]
+  const int kd = RS * C;
+  for (int i = threadIdx.x; i < K * kd; i += blockDim.x) w_sm[i] = wt[i];
+  __syncthreads();
+
+  const long long HW = (long long)H * W;
+  const long long M = (long long)B * HW;
+  const int kper = (K + SC_KSLOTS - 1) / SC_KSLOTS;
+  const int pix_local = threadIdx.x / SC_KSLOTS;   // 0..31
+  const int kslot = threadIdx.x % SC_KSLOTS;
+  const long long p = (long long)blockIdx.x * 32 + pix_local;
+  if (p >= M) return;
+  const int b = (int)(p / HW);
+  const int rem = (int)(p - (long long)b * HW);
+  const int py = rem / W, px = rem % W;
+
+  // gather the RS*C input taps for this pixel (zeros outside)
+  float taps[RS * 8];
+#pragma unroll
+  for (int rs = 0; rs < RS; ++rs) {
+    const int r = RS == 1 ? 0 : rs / 3;
+    const int s = RS == 1 ? 0 : rs % 3;
+    const int sy = py + (RS == 1 ? 0 : r - 1);
+    const int sx = px + (RS == 1 ? 0 : s - 1);
+    const bool ok = sy >= 0 && sy < H && sx >= 0 && sx < W;
+    const uint16_t* src =
+        x + (((long long)b * H + sy) * W + sx) * C;
+    for (int c = 0; c < C; ++c)
+      taps[rs * C + c] = ok ? bf16_bits_to_f32(src[c]) : 0.f;
+  }
+
+  const int k0 = kslot * kper;
+  const int k1 = min(k0 + kper, K);
+  uint16_t* dst = y + p * K;
+  for (int k = k0; k < k1; ++k) {
+    const uint16_t* wrow = &w_sm[k * kd];
+    float acc = bias ? bias[k] : 0.f;
+    for (int i = 0; i < kd; ++i)
+      acc += taps[i] * bf16_bits_to_f32(wrow[i]);
+    if (fuse_silu) acc = silu_f(acc);
+    dst[k] = f32_to_bf16_bits(acc);
+  }
+}
+
+torch::Tensor conv_smallc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
+                          int64_t B, int64_t H, int64_t W, int64_t C,
+                          int64_t K, int64_t rs, bool fuse_silu) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(C <= 8, "conv_smallc is for C <= 8");
+  TORCH_CHECK(rs == 9 || rs == 1);
+  auto y = torch::empty({B, H, W, K}, x.options());
+  const long long M = B * H * W;
+  dim3 grid((unsigned)((M + 31) / 32));
+  dim3 block(256);
+  const size_t smem = (size_t)K * rs * C * sizeof(uint16_t);
+  TORCH_CHECK(smem <= 64 * 1024, "weights exceed LDS budget");
+  auto stream = at::hip::getCurrentHIPStream();
+  const float* bptr = nullptr;
+  torch::Tensor bf;
+  if (bias.defined() && bias.numel() > 0) {
+    bf = bias.contiguous().to(at::kFloat);
+    bptr = bf.data_ptr<float>();
+  }
+  if (rs == 9)
+    hipLaunchKernelGGL((conv_smallc_kernel<9>), grid, block, smem, stream,
+                       (const uint16_t*)x.data_ptr(), (const uint16_t*)wt.data_ptr(),
+                       bptr, (uint16_t*)y.data_ptr(), (int)B, (int)H, (int)W,
+                       (int)C, (int)K, fuse_silu ? 1 : 0);
+  else
+    hipLaunchKernelGGL((conv_smallc_kernel<1>), grid, block, smem, stream,
+                       (const uint16_t*)x.data_ptr(), (const uint16_t*)wt.data_ptr(),
+                       bptr, (uint16_t*)y.data_ptr(), (int)B, (int)H, (int)W,
+                       (int)C, (int)K, fuse_silu ? 1 : 0);
+  HIP_CHECK_LAUNCH();
+  return y;
+}
+
+// ---------------------------------------------------------------------------
 // NHWC fused GroupNorm(+SiLU): one block per (n, g); channels of a group are
 // contiguous within each pixel (span Cg*2 bytes).
 // ---------------------------------------------------------------------------
@@ -332,22 +424,32 @@ __global__ void groupnorm_nhwc_kernel(const uint16_t* __restrict__ x,
   __shared__ float s_mean, s_rstd;
 
   float sum = 0.f, sumsq = 0.f;
-  // each thread walks pixels with stride blockDim; within a pixel reads the
-  // group's Cg contiguous channels (vector 8 where possible)
-  for (long long p = threadIdx.x; p < HW; p += blockDim.x) {
-    const uint16_t* px = x + base + p * C;
-    int c = 0;
-    for (; c + 7 < Cg; c += 8) {
-      ushort8_t v = *reinterpret_cast<const ushort8_t*>(px + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = bf16_bits_to_f32(v[j]);
-        sum += f;
-        sumsq += f * f;
-      }
+  // flat index over the group's (pixel, channel) plane: consecutive lanes
+  // touch consecutive channels of the same pixel, so a wave's accesses
+  // collapse into ~Cg-byte segments instead of one segment per lane
+  const long long total = HW * Cg;
+  for (long long i = (long long)threadIdx.x * 2; i + 1 < total;
+       i += (long long)blockDim.x * 2) {
+    const long long p = i / Cg;
+    const int c = (int)(i - p * Cg);
+    const uint16_t* px = x + base + p * C + c;
+    if (c + 1 < Cg) {
+      float f0 = bf16_bits_to_f32(px[0]);
+      float f1 = bf16_bits_to_f32(px[1]);
+      sum += f0 + f1;
+      sumsq += f0 * f0 + f1 * f1;
+    } else {  // pair straddles the pixel boundary
+      float f0 = bf16_bits_to_f32(px[0]);
+      float f1 = bf16_bits_to_f32(x[base + (p + 1) * C]);
+      sum += f0 + f1;
+      sumsq += f0 * f0 + f1 * f1;
     }
-    for (; c < Cg; ++c) {
-      float f = bf16_bits_to_f32(px[c]);
+  }
+  if (total & 1) {  // odd tail element, thread 0
+    if (threadIdx.x == 0) {
+      const long long i = total - 1;
+      const long long p = i / Cg;
+      float f = bf16_bits_to_f32(x[base + p * C + (int)(i - p * Cg)]);
       sum += f;
       sumsq += f * f;
     }
@@ -364,30 +466,14 @@ __global__ void groupnorm_nhwc_kernel(const uint16_t* __restrict__ x,
   __syncthreads();
   const float mean = s_mean, rstd = s_rstd;
 
-  for (long long p = threadIdx.x; p < HW; p += blockDim.x) {
-    const uint16_t* px = x + base + p * C;
-    uint16_t* py = y + base + p * C;
-    int c = 0;
-    for (; c + 7 < Cg; c += 8) {
-      ushort8_t v = *reinterpret_cast<const ushort8_t*>(px + c);
-      ushort8_t o;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int ch = g * Cg + c + j;
-        float f = (bf16_bits_to_f32(v[j]) - mean) * rstd;
-        f = f * weight[ch] + bias[ch];
-        if (FUSE_SILU) f = silu_f(f);
-        o[j] = f32_to_bf16_bits(f);
-      }
-      *reinterpret_cast<ushort8_t*>(py + c) = o;
-    }
-    for (; c < Cg; ++c) {
-      const int ch = g * Cg + c;
-      float f = (bf16_bits_to_f32(px[c]) - mean) * rstd;
-      f = f * weight[ch] + bias[ch];
-      if (FUSE_SILU) f = silu_f(f);
-      py[c] = f32_to_bf16_bits(f);
-    }
+  for (long long i = threadIdx.x; i < total; i += blockDim.x) {
+    const long long p = i / Cg;
+    const int c = (int)(i - p * Cg);
+    const int ch = g * Cg + c;
+    float f = (bf16_bits_to_f32(x[base + p * C + c]) - mean) * rstd;
+    f = f * weight[ch] + bias[ch];
+    if (FUSE_SILU) f = silu_f(f);
+    y[base + p * C + c] = f32_to_bf16_bits(f);
   }
 }
 

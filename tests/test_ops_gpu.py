@@ -68,6 +68,40 @@ def test_attention_gqa_gpu(extmod):
     assert (out - ref).abs().max().item() < 0.03
 
 
+@pytest.mark.parametrize("b,h,n,nk,d", [(2, 8, 640, 640, 40), (1, 4, 100, 77, 64),
+                                        (2, 2, 256, 256, 80)])
+def test_attention_packed_matches_reference(extmod, b, h, n, nk, d):
+    """Packed [B,N,H*D] strided path vs CPU reference."""
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(b * 100 + d)
+    q = torch.randn(b, n, h * d) / 2
+    k = torch.randn(b, nk, h * d) / 2
+    v = torch.randn(b, nk, h * d)
+    out = dispatch.attention_packed(
+        q.cuda().to(torch.bfloat16), k.cuda().to(torch.bfloat16),
+        v.cuda().to(torch.bfloat16), heads=h,
+    ).float().cpu()
+    ref = dispatch.attention_packed(q, k, v, heads=h)
+    err = (out - ref).abs().max().item()
+    assert err < 0.03, f"packed err {err} b={b} h={h} n={n} d={d}"
+
+
+def test_conv_smallc_matches_torch(extmod):
+    import torch.nn.functional as F
+
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(2)
+    conv = torch.nn.Conv2d(4, 320, 3, padding=1).cuda().to(torch.bfloat16)
+    x = (torch.randn(2, 4, 17, 19) / 2).cuda().to(torch.bfloat16)
+    xcl = x.contiguous(memory_format=torch.channels_last)
+    y = dispatch.conv2d_smallc(xcl, conv).float()
+    ref = F.conv2d(x.float(), conv.weight.float(), conv.bias.float(), padding=1)
+    err = (y - ref).abs().max().item()
+    assert err / ref.abs().max().item() < 0.05
+
+
 def test_attention_defer_max_rescale_branch(extmod):
     """Force the defer-max rescale branch (guide T13 hazard): a spiked K row
     in a LATER tile makes the running max jump past the threshold. Results
