@@ -85,13 +85,35 @@ def sample_tiles(
     sigmas = schedule.sigmas(params.steps, params.scheduler, params.denoise).to(
         canvas.device
     )
-    denoiser = CFGDenoiser(stack.unet, schedule, cond, uncond, params.cfg)
+    base_denoiser = CFGDenoiser(stack.unet, schedule, cond, uncond, params.cfg)
 
+    from .conditioning import SPATIAL_KEYS, crop_tile_conditioning
+
+    has_spatial = any(k in (cond or {}) for k in SPATIAL_KEYS) or any(
+        k in (uncond or {}) for k in SPATIAL_KEYS
+    )
     work = [(t, b) for t in sorted(tile_indices) for b in range(B)]
-    step = max(params.tile_batch, 1)
+    # spatial conditioning differs per tile -> tiles can't share a sampler
+    # batch (reference processes per tile for the same reason)
+    step = 1 if has_spatial else max(params.tile_batch, 1)
     results: dict[tuple[int, int], torch.Tensor] = {}
     for i in range(0, len(work), step):
         chunk = work[i : i + step]
+        chunk_cond, chunk_uncond = cond, uncond
+        if has_spatial:
+            plan = plans[chunk[0][0]]
+            chunk_cond = crop_tile_conditioning(
+                cond, plan.crop_region, plan.canvas_size, plan.process_size
+            )
+            chunk_uncond = crop_tile_conditioning(
+                uncond, plan.crop_region, plan.canvas_size, plan.process_size
+            )
+        denoiser = (
+            base_denoiser
+            if not has_spatial
+            else CFGDenoiser(stack.unet, schedule, chunk_cond, chunk_uncond,
+                             params.cfg)
+        )
         # ---- extract + resample each (tile, batch) crop to process size ----
         crops = []
         for t, b in chunk:
