@@ -186,7 +186,8 @@ class StemConv(nn.Module):
         self.conv = nn.Conv2d(cin, cout, 3, padding=1)
 
     def forward(self, x):
-        if x.is_cuda and x.is_contiguous(memory_format=torch.channels_last):
+        if (x.is_cuda and x.shape[1] <= 8
+                and x.is_contiguous(memory_format=torch.channels_last)):
             return ops.conv2d_smallc(x, self.conv)
         return self.conv(x)
 

@@ -15,7 +15,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import dispatch as ops
-from .unet import FusedGroupNorm
+from .unet import FusedGroupNorm, StemConv
 
 
 @dataclass
@@ -95,7 +95,7 @@ class VAEEncoder(nn.Module):
     def __init__(self, cfg: VAEConfig):
         super().__init__()
         ch = cfg.base_channels
-        self.conv_in = nn.Conv2d(cfg.in_channels, ch, 3, padding=1)
+        self.conv_in = StemConv(cfg.in_channels, ch)
         downs = []
         cin = ch
         for level, mult in enumerate(cfg.channel_mult):
@@ -126,7 +126,7 @@ class VAEDecoder(nn.Module):
         super().__init__()
         ch = cfg.base_channels
         cin = ch * cfg.channel_mult[-1]
-        self.conv_in = nn.Conv2d(cfg.latent_channels, cin, 3, padding=1)
+        self.conv_in = StemConv(cfg.latent_channels, cin)
         self.mid = nn.ModuleList(
             [VAEResBlock(cin, cin), VAEAttention(cin), VAEResBlock(cin, cin)]
         )
