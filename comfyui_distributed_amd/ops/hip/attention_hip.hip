@@ -74,10 +74,9 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     }
   }
 
-  float m_run[4], l_run[4];
+  // per-lane softmax row state: this lane's q row is (lane & 15)
+  float m_run = -1e30f, l_run = 0.f;
   f32x4 oacc[DN];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
 #pragma unroll
   for (int n = 0; n < DN; ++n) oacc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
@@ -117,7 +116,11 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- scores ----------------------------------------------------------
+    // ---- scores, SWAPPED: S^T = mfma(K, Q) so each lane's 16 score
+    // values all belong to ONE q row (col = lane&15) — softmax becomes
+    // in-register with only 2 cross-lane steps (guide's swapped-QK^T
+    // idiom).  s[f][r] = score(key = key0 + f*16 + (lane>>4)*4 + r,
+    //                          qrow = lane&15)
     f32x4 s[KFRAG];
 #pragma unroll
     for (int f = 0; f < KFRAG; ++f) s[f] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -125,69 +128,67 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     for (int kk = 0; kk < DK; ++kk)
 #pragma unroll
       for (int f = 0; f < KFRAG; ++f)
-        s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], kf[kk][f],
+        s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[kk][f], qfrag[kk],
                                                        s[f], 0, 0, 0);
 
     if (t + 1 < ntiles) load_kfrags(key0 + KT, kf);
 
 #pragma unroll
     for (int f = 0; f < KFRAG; ++f) {
-      const int key = key0 + f * 16 + (lane & 15);
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        s[f][r] = (key < Nk) ? s[f][r] * scale : -1e30f;
-    }
-
-    // ---- online softmax with defer-max -----------------------------------
-    float p[KFRAG][4];
-    float mt[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float m = s[0][r];
-#pragma unroll
-      for (int f = 1; f < KFRAG; ++f) m = fmaxf(m, s[f][r]);
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        m = fmaxf(m, __shfl_xor(m, off, 64));
-      mt[r] = m;
-    }
-    bool need = false;
-#pragma unroll
-    for (int r = 0; r < 4; ++r) need |= (mt[r] - m_run[r]) > DEFER_THR;
-    if (__ballot(need) != 0ull) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const float m_new = fmaxf(m_run[r], mt[r]);
-        const float a = __expf(m_run[r] - m_new);
-        m_run[r] = m_new;
-        l_run[r] *= a;
-#pragma unroll
-        for (int n = 0; n < DN; ++n) oacc[n][r] *= a;
+        const int key = key0 + f * 16 + (lane >> 4) * 4 + r;
+        s[f][r] = (key < Nk) ? s[f][r] * scale : -1e30f;
       }
-    }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float rowsum = 0.f;
-#pragma unroll
-      for (int f = 0; f < KFRAG; ++f) {
-        p[f][r] = __expf(s[f][r] - m_run[r]);
-        rowsum += p[f][r];
-      }
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        rowsum += __shfl_xor(rowsum, off, 64);
-      l_run[r] += rowsum;
     }
 
-    // ---- P -> LDS (wave-private) -----------------------------------------
+    // ---- online softmax, per-lane row state ------------------------------
+    float mt = s[0][0];
+#pragma unroll
+    for (int f = 0; f < KFRAG; ++f)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) mt = fmaxf(mt, s[f][r]);
+    mt = fmaxf(mt, __shfl_xor(mt, 16, 64));
+    mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+
+    const bool need = (mt - m_run) > DEFER_THR;
+    if (__ballot(need) != 0ull) {
+      const float m_new = fmaxf(m_run, mt);
+      const float a = __expf(m_run - m_new);
+      m_run = m_new;
+      l_run *= a;
+      // O fragment rows are (lane>>4)*4+r: fetch each row's alpha from a
+      // lane holding that row's state (same 16-lane group, col == row)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int src = (lane & 48) | (((lane >> 4) * 4 + r) & 15);
+        const float ar = __shfl(a, src, 64);
+#pragma unroll
+        for (int n = 0; n < DN; ++n) oacc[n][r] *= ar;
+      }
+    }
+    float p[KFRAG][4];
+    float rowsum = 0.f;
+#pragma unroll
+    for (int f = 0; f < KFRAG; ++f)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        p[f][r] = __expf(s[f][r] - m_run);
+        rowsum += p[f][r];
+      }
+    rowsum += __shfl_xor(rowsum, 16, 64);
+    rowsum += __shfl_xor(rowsum, 32, 64);
+    l_run += rowsum;
+
+    // ---- P -> LDS in the PV A-fragment layout ----------------------------
     {
-      const int col = lane & 15;
-      const int rg = lane >> 4;
+      const int qrow = lane & 15;
+      const int g = lane >> 4;
 #pragma unroll
       for (int f = 0; f < KFRAG; ++f)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          p_lds[wave][rg * 4 + r][f * 16 + col] = f32_to_bf16_bits(p[f][r]);
+          p_lds[wave][qrow][f * 16 + g * 4 + r] = f32_to_bf16_bits(p[f][r]);
     }
 
     // ---- PV --------------------------------------------------------------
@@ -213,9 +214,13 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
   {
     const int col = lane & 15;
     const int rg = lane >> 4;
+    const float rl_own = __builtin_amdgcn_rcpf(l_run);
     float rl[4];
 #pragma unroll
-    for (int r = 0; r < 4; ++r) rl[r] = __builtin_amdgcn_rcpf(l_run[r]);
+    for (int r = 0; r < 4; ++r) {
+      const int src = (lane & 48) | ((rg * 4 + r) & 15);
+      rl[r] = __shfl(rl_own, src, 64);
+    }
 #pragma unroll
     for (int n = 0; n < DN; ++n) {
       const int d = n * 16 + col;
