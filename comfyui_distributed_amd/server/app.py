@@ -106,6 +106,8 @@ class DistributedServer:
         r.add_post("/distributed/stop_worker", self.post_stop_worker)
         r.add_get("/distributed/managed_workers", self.get_managed_workers)
         r.add_get("/distributed/worker_log", self.get_worker_log)
+        r.add_get("/distributed/local_log", self.get_local_log)
+        r.add_post("/distributed/load_image", self.post_load_image)
         r.add_post("/interrupt", self.post_interrupt)
         r.add_get("/distributed/ws", self.ws_handler)
         app.on_startup.append(self._on_startup)
@@ -530,6 +532,27 @@ class DistributedServer:
             fh.seek(max(0, size - 64 * 1024))
             tail = fh.read().decode(errors="replace").splitlines()[-lines:]
         return web.json_response({"log": "\n".join(tail)})
+
+    async def get_local_log(self, request):
+        from ..utils.logging import LOG_BUFFER
+
+        lines = int(request.query.get("lines", "100"))
+        return web.json_response({"log": "\n".join(list(LOG_BUFFER)[-lines:])})
+
+    async def post_load_image(self, request):
+        """Return an input image as a base64 PNG (reference
+        job_routes.py:238-258: workers fetch inputs from the master)."""
+        from pathlib import Path
+
+        from ..utils.image import decode_png_bytes, encode_png_base64
+
+        data = await request.json()
+        name = os.path.basename(str(data.get("filename", "")))
+        path = Path(self.executor.context.get("input_dir", "input")) / name
+        if not path.is_file():
+            return _err("not found", status=404)
+        tensor = decode_png_bytes(path.read_bytes())
+        return web.json_response({"image": encode_png_base64(tensor)})
 
     async def post_interrupt(self, request):
         """User interrupt: flags the node runtime so every wait loop raises

@@ -12,6 +12,12 @@ import time
 
 PREFIX = "[Distributed]"
 
+#: in-memory ring buffer served by /distributed/local_log (reference keeps
+#: an app.logger buffer, worker_routes.py:348-390)
+from collections import deque as _deque
+
+LOG_BUFFER: "_deque[str]" = _deque(maxlen=500)
+
 _debug_cache = {"value": None, "ts": 0.0}
 _DEBUG_TTL = 5.0
 
@@ -34,7 +40,9 @@ def is_debug_enabled() -> bool:
 
 
 def log(*args) -> None:
-    print(PREFIX, *args, file=sys.stderr, flush=True)
+    line = " ".join(str(a) for a in args)
+    LOG_BUFFER.append(f"{PREFIX} {line}")
+    print(PREFIX, line, file=sys.stderr, flush=True)
 
 
 def debug_log(*args) -> None:
