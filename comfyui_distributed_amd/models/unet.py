@@ -115,14 +115,27 @@ class CrossAttention(nn.Module):
         self.to_v = nn.Linear(context_dim, inner, bias=False)
         self.to_out = nn.Linear(inner, dim)
 
+    def _fused_weight(self, name: str, parts: list[torch.Tensor]) -> torch.Tensor:
+        cached = getattr(self, name, None)
+        if cached is None or cached.device != parts[0].device \
+                or cached.dtype != parts[0].dtype:
+            cached = torch.cat([p.detach() for p in parts], dim=0)
+            object.__setattr__(self, name, cached)
+        return cached
+
     def forward(self, x, context=None):
-        context = x if context is None else context
-        # packed layout [B, N, H*D] straight from the projections — the
-        # strided kernel consumes it with zero reshapes/copies
-        o = ops.attention_packed(
-            self.to_q(x), self.to_k(context), self.to_v(context),
-            heads=self.heads,
-        )
+        # fused projections + strided attention: one GEMM feeds the kernel
+        # directly — zero splits, pads or permutes on the host
+        if context is None:  # self-attention: fused QKV
+            w = self._fused_weight(
+                "_wqkv", [self.to_q.weight, self.to_k.weight, self.to_v.weight]
+            )
+            o = ops.attention_qkv(F.linear(x, w), heads=self.heads)
+        else:  # cross-attention: fused KV
+            w = self._fused_weight("_wkv", [self.to_k.weight, self.to_v.weight])
+            o = ops.attention_q_kv(
+                self.to_q(x), F.linear(context, w), heads=self.heads
+            )
         return self.to_out(o)
 
 

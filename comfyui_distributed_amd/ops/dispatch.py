@@ -242,6 +242,35 @@ def attention_packed(
             .reshape(b, nq, heads * d))
 
 
+def attention_qkv(qkv: torch.Tensor, heads: int,
+                  scale: float | None = None) -> torch.Tensor:
+    """Self-attention straight off a fused projection [B, N, 3*H*D]."""
+    hd = qkv.shape[-1] // 3
+    d = hd // heads
+    scale = scale if scale is not None else 1.0 / math.sqrt(d)
+    if _on_gpu(qkv):
+        return ext.get_ext(True).attn_fwd_qkv(
+            qkv.to(torch.bfloat16).contiguous(), heads, scale)
+    q, k, v = qkv.split(hd, dim=-1)
+    return attention_packed(q.contiguous(), k.contiguous(), v.contiguous(),
+                            heads=heads, scale=scale)
+
+
+def attention_q_kv(q: torch.Tensor, kv: torch.Tensor, heads: int,
+                   scale: float | None = None) -> torch.Tensor:
+    """Cross-attention: q [B,Nq,H*D] + fused kv [B,Nk,2*H*D]."""
+    hd = q.shape[-1]
+    d = hd // heads
+    scale = scale if scale is not None else 1.0 / math.sqrt(d)
+    if _on_gpu(q):
+        return ext.get_ext(True).attn_fwd_q_kv(
+            q.to(torch.bfloat16).contiguous(),
+            kv.to(torch.bfloat16).contiguous(), heads, scale)
+    k, v = kv.split(hd, dim=-1)
+    return attention_packed(q, k.contiguous(), v.contiguous(), heads=heads,
+                            scale=scale)
+
+
 # ---------------------------------------------------------------------------
 # tile pipeline (Lanczos-3 resample + erf-mask blend)
 # ---------------------------------------------------------------------------

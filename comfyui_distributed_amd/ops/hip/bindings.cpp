@@ -18,6 +18,9 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 torch::Tensor attn_fwd_packed(torch::Tensor q, torch::Tensor k,
                               torch::Tensor v, int64_t heads,
                               int64_t kv_heads, double scale);
+torch::Tensor attn_fwd_qkv(torch::Tensor qkv, int64_t heads, double scale);
+torch::Tensor attn_fwd_q_kv(torch::Tensor q, torch::Tensor kv, int64_t heads,
+                            double scale);
 torch::Tensor mfma_selftest(torch::Tensor a, torch::Tensor b);
 torch::Tensor conv_nhwc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
                         int64_t B, int64_t H, int64_t W, int64_t C, int64_t K,
@@ -41,6 +44,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "flash attention forward, bf16 MFMA");
   m.def("attn_fwd_packed", &attn_fwd_packed,
         "flash attention on packed [B,N,H*D] QKV (no host reshapes)");
+  m.def("attn_fwd_qkv", &attn_fwd_qkv,
+        "self-attention on fused [B,N,3*H*D] projection output");
+  m.def("attn_fwd_q_kv", &attn_fwd_q_kv,
+        "cross-attention on q [B,Nq,H*D] + fused kv [B,Nk,2*H*D]");
   m.def("mfma_selftest", &mfma_selftest,
         "single-wave 16x16x32 bf16 MFMA with the kernel fragment layouts");
   m.def("conv_nhwc", &conv_nhwc,
