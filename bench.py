@@ -54,7 +54,7 @@ def parse_args():
     ap.add_argument("--sampler-steps", type=int, default=20)
     ap.add_argument("--denoise", type=float, default=0.5)
     ap.add_argument("--cfg", type=float, default=8.0)
-    ap.add_argument("--tile-batch", type=int, default=8)
+    ap.add_argument("--tile-batch", type=int, default=16)
     args = ap.parse_args()
     for key, value in CONFIG_PRESETS[args.config].items():
         if getattr(args, key) == ap.get_default(key):

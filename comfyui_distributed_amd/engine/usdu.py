@@ -41,7 +41,7 @@ class USDUParams:
     mask_blur: int = 8
     force_uniform_tiles: bool = True
     tiled_decode: bool = False
-    tile_batch: int = 4  # tiles sampled together per sampler call
+    tile_batch: int = 16  # tiles sampled together per sampler call
 
 
 def plan_for_image(width: int, height: int, p: USDUParams):
