@@ -87,6 +87,30 @@ def test_attention_packed_matches_reference(extmod, b, h, n, nk, d):
     assert err < 0.03, f"packed err {err} b={b} h={h} n={n} d={d}"
 
 
+def test_attention_fused_qkv_matches_reference(extmod):
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(11)
+    b, h, n, d = 2, 8, 200, 40
+    qkv = torch.randn(b, n, 3 * h * d) / 2
+    out = dispatch.attention_qkv(qkv.cuda().to(torch.bfloat16), heads=h)
+    ref = dispatch.attention_qkv(qkv, heads=h)
+    assert (out.float().cpu() - ref).abs().max().item() < 0.03
+
+
+def test_attention_fused_q_kv_matches_reference(extmod):
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(12)
+    b, h, nq, nk, d = 2, 4, 150, 77, 64
+    q = torch.randn(b, nq, h * d) / 2
+    kv = torch.randn(b, nk, 2 * h * d) / 2
+    out = dispatch.attention_q_kv(q.cuda().to(torch.bfloat16),
+                                  kv.cuda().to(torch.bfloat16), heads=h)
+    ref = dispatch.attention_q_kv(q, kv, heads=h)
+    assert (out.float().cpu() - ref).abs().max().item() < 0.03
+
+
 def test_conv_smallc_matches_torch(extmod):
     import torch.nn.functional as F
 
