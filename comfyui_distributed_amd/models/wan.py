@@ -79,10 +79,10 @@ def rope_3d(t: int, h: int, w: int, head_dim: int, device, dtype=torch.float32):
 
 
 def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor):
-    """x: [B*H, N, D]; rotate pairs (even, odd)."""
+    """x: [B, N, H, D] (head-split view); rotate (even, odd) pairs of D."""
     x1, x2 = x[..., 0::2], x[..., 1::2]
-    c = cos[None, : x.shape[1], :].to(x.dtype)
-    s = sin[None, : x.shape[1], :].to(x.dtype)
+    c = cos[None, : x.shape[1], None, :].to(x.dtype)
+    s = sin[None, : x.shape[1], None, :].to(x.dtype)
     out = torch.empty_like(x)
     out[..., 0::2] = x1 * c - x2 * s
     out[..., 1::2] = x1 * s + x2 * c
@@ -113,20 +113,6 @@ class WanBlock(nn.Module):
         # adaLN modulation: 6 gates/shifts/scales from the time embedding
         self.mod = nn.Parameter(torch.randn(6, d) / d**0.5)
 
-    def _split(self, x, b, n):
-        return (
-            x.reshape(b, n, self.heads, self.head_dim)
-            .permute(0, 2, 1, 3)
-            .reshape(b * self.heads, n, self.head_dim)
-        )
-
-    def _merge(self, x, b, n):
-        return (
-            x.reshape(b, self.heads, n, self.head_dim)
-            .permute(0, 2, 1, 3)
-            .reshape(b, n, -1)
-        )
-
     def forward(self, x, emb6, context, rope_cs):
         # emb6: [B, 6, dim] time modulation (per WAN: shared table + time MLP)
         b, n, _ = x.shape
@@ -134,26 +120,21 @@ class WanBlock(nn.Module):
         shift_a, scale_a, gate_a, shift_f, scale_f, gate_f = m.unbind(1)
 
         h = self.norm1(x) * (1 + scale_a[:, None]) + shift_a[:, None]
-        q = self._split(self.q(h), b, n)
-        k = self._split(self.k(h), b, n)
-        v = self._split(self.v(h), b, n)
-        q = self.norm_q(q)
-        k = self.norm_k(k)
+        # head-split VIEW for per-head RMSNorm + RoPE, then back to the
+        # packed [B, N, H*D] layout the strided attention kernel reads
+        hs = (b, n, self.heads, self.head_dim)
         cos, sin = rope_cs
-        q = apply_rope(q, cos, sin)
-        k = apply_rope(k, cos, sin)
-        attn = ops.attention(q, k, v, heads=self.heads)
-        x = x + gate_a[:, None] * self.o(self._merge(attn, b, n))
+        q = apply_rope(self.norm_q(self.q(h).reshape(hs)), cos, sin).reshape(b, n, -1)
+        k = apply_rope(self.norm_k(self.k(h).reshape(hs)), cos, sin).reshape(b, n, -1)
+        attn = ops.attention_packed(q.contiguous(), k.contiguous(), self.v(h),
+                                    heads=self.heads)
+        x = x + gate_a[:, None] * self.o(attn)
 
-        # cross attention (no modulation per WAN)
+        # cross attention (no modulation per WAN): packed, zero reshapes
         h = self.norm2(x)
-        nq = h.shape[1]
-        nk = context.shape[1]
-        q = self._split(self.cq(h), b, nq)
-        k = self._split(self.ck(context), b, nk)
-        v = self._split(self.cv(context), b, nk)
-        attn = ops.attention(q, k, v, heads=self.heads)
-        x = x + self.co(self._merge(attn, b, nq))
+        attn = ops.attention_packed(self.cq(h), self.ck(context),
+                                    self.cv(context), heads=self.heads)
+        x = x + self.co(attn)
 
         h = self.norm3(x) * (1 + scale_f[:, None]) + shift_f[:, None]
         x = x + gate_f[:, None] * self.ffn2(F.silu(self.ffn1(h)))
