@@ -17,7 +17,7 @@ Reference counterpart: upscale/tile_ops.py process_tile/process_tiles_batch
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 

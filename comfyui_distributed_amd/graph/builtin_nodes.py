@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import torch
 
-from ..engine.generate import GenParams, generate_latents
 from ..models import create_diffusion_stack
 from ..models.sampling import CFGDenoiser, SAMPLERS, SCHEDULERS, sample
 from ..utils.image import decode_png_bytes, encode_png_bytes

@@ -10,8 +10,6 @@ from __future__ import annotations
 
 from typing import Any
 
-import torch
-
 from ..utils.errors import PromptValidationError
 from .prompt import PromptGraph, is_link
 

@@ -27,7 +27,6 @@ from ..utils.async_bridge import run_async_in_server_loop
 from ..utils.image import decode_png_base64, encode_png_base64
 from ..utils.logging import debug_log, log
 from .runtime import get_runtime
-from .utilities import any_type  # noqa: F401  (re-export parity)
 
 EMPTY_AUDIO = {"waveform": torch.zeros(1, 2, 1), "sample_rate": 44100}
 

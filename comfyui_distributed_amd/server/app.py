@@ -20,7 +20,7 @@ import uuid
 
 from aiohttp import web
 
-from ..graph.executor import Executor, default_registry, validate_prompt
+from ..graph.executor import Executor, validate_prompt
 from ..nodes.collector import decode_job_complete_envelope
 from ..nodes.runtime import NodeRuntime, set_runtime
 from ..server.job_state import JobState

@@ -20,7 +20,7 @@ import threading
 import time
 from collections import deque
 
-from ..utils.config import config_transaction, load_config
+from ..utils.config import config_transaction
 from ..utils.errors import TunnelError
 from ..utils.logging import log
 
