@@ -29,6 +29,15 @@ async def get_client_session():
         return _session
 
 
+async def close_client_session():
+    """Close the shared session (server shutdown / test teardown)."""
+    global _session
+    async with _session_lock:
+        if _session is not None and not _session.closed:
+            await _session.close()
+        _session = None
+
+
 def normalize_host(host: str) -> str:
     host = (host or "").strip()
     for prefix in ("http://", "https://"):
