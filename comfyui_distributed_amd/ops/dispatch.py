@@ -51,7 +51,15 @@ def group_norm_silu(
         return ext.get_ext(True).group_norm_fused(
             x.to(torch.bfloat16), groups, weight, bias, eps, silu
         )
-    y = F.group_norm(x.float(), groups, weight.float(), bias.float(), eps)
+    # manual GN (F.group_norm rejects 1-value-per-group shapes, e.g. a
+    # batch-1 tensor at 1x1 spatial in deep tiny-config levels)
+    b, c = x.shape[0], x.shape[1]
+    xf = x.float().reshape(b, groups, -1)
+    mean = xf.mean(-1, keepdim=True)
+    var = xf.var(-1, unbiased=False, keepdim=True)
+    y = ((xf - mean) / (var + eps).sqrt()).reshape(x.shape)
+    shape = [1, c] + [1] * (x.dim() - 2)
+    y = y * weight.float().view(shape) + bias.float().view(shape)
     if silu:
         y = F.silu(y)
     return y.to(x.dtype)

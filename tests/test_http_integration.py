@@ -1,0 +1,102 @@
+"""Full master<->worker HTTP integration: one process, two real aiohttp
+servers. A seed-parallel workflow is queued on the master; the master
+prunes/overrides/dispatches the worker prompt over real HTTP, both sides
+execute the graph (tiny model), the worker's collector POSTs base64-PNG
+envelopes to the master's /distributed/job_complete, and the master's
+collector combines master+worker batches.
+
+The reference has no test like this (its suite stubs all
+transport); this is the end-to-end wire check.
+"""
+
+import asyncio
+import json
+
+import pytest
+import torch
+from aiohttp.test_utils import TestClient, TestServer
+
+from comfyui_distributed_amd.server.app import DistributedServer
+from comfyui_distributed_amd.utils.config import load_config, save_config
+
+
+@pytest.mark.timeout(120)
+def test_seed_parallel_roundtrip_over_http(tmp_config, monkeypatch):
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+
+    async def go():
+        # worker first, master second: the master's NodeRuntime must be the
+        # process-global one (its collector drains its own job_state)
+        worker_srv = DistributedServer(is_worker=True)
+        worker_client = TestClient(TestServer(worker_srv.build_app()))
+        await worker_client.start_server()
+
+        master_srv = DistributedServer()
+        master_previews: list = []
+        master_srv.executor.context["preview_images"] = master_previews
+        master_srv.executor.context["saved_images"] = []
+        master_client = TestClient(TestServer(master_srv.build_app()))
+        await master_client.start_server()
+
+        # register the worker in config with its real ephemeral port
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "worker1", "host": "127.0.0.1",
+            "port": worker_client.server.port, "cuda_device": 0,
+            "enabled": True, "type": "remote",  # remote => real HTTP callbacks
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = master_client.server.port
+        save_config(cfg)
+
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "tiny"}},
+            "2": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "cat", "clip": ["1", 1]}},
+            "3": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "", "clip": ["1", 1]}},
+            "4": {"class_type": "DistributedSeed", "inputs": {"seed": 5}},
+            "5": {"class_type": "EmptyLatentImage",
+                  "inputs": {"width": 16, "height": 16, "batch_size": 1}},
+            "6": {"class_type": "KSampler", "inputs": {
+                "model": ["1", 0], "seed": ["4", 0], "steps": 1, "cfg": 1.0,
+                "sampler_name": "euler", "scheduler": "karras",
+                "positive": ["2", 0], "negative": ["3", 0],
+                "latent_image": ["5", 0], "denoise": 1.0}},
+            "7": {"class_type": "VAEDecode",
+                  "inputs": {"samples": ["6", 0], "vae": ["1", 2]}},
+            "8": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["7", 0], "load_balance": False}},
+            "9": {"class_type": "PreviewImage", "inputs": {"images": ["8", 0]}},
+        }
+
+        resp = await master_client.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "it", "enabled_worker_ids": ["w1"],
+        })
+        assert resp.status == 200
+        body = await resp.json()
+        assert "w1" in body["participants"] and "master" in body["participants"]
+
+        # wait for both executors to drain
+        for _ in range(400):
+            if (master_srv.prompt_queue.qsize() == 0 and not master_srv.executing
+                    and worker_srv.prompt_queue.qsize() == 0
+                    and not worker_srv.executing and master_previews):
+                break
+            await asyncio.sleep(0.25)
+
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await worker_client.close()
+        await master_client.close()
+        return master_previews
+
+    previews = asyncio.run(go())
+    assert previews, "master collector never produced output"
+    combined = previews[0]
+    # master image + worker image, combined on the master
+    assert combined.shape == (2, 16, 16, 3)
+    assert torch.isfinite(combined).all()
