@@ -17,6 +17,15 @@ from ..utils.image import decode_png_bytes, encode_png_bytes
 _STACK_CACHE: dict[tuple, object] = {}
 
 
+def stable_text_seed(text: str) -> int:
+    """Deterministic across processes (Python's hash() is salted per
+    process, which would give master and workers different synthetic
+    conditioning for the same prompt text)."""
+    import hashlib
+
+    return int.from_bytes(hashlib.md5(str(text).encode()).digest()[:4], "little") % (2**31)
+
+
 def _device_dtype():
     if torch.cuda.is_available():
         return "cuda:0", torch.bfloat16
@@ -73,8 +82,7 @@ class CLIPTextEncode(_ContextNode):
     CATEGORY = "conditioning"
 
     def encode(self, text="", clip=None):
-        seed = abs(hash(str(text))) % (2**31)
-        return (clip.make_conditioning(seed),)
+        return (clip.make_conditioning(stable_text_seed(text)),)
 
 
 class EmptyLatentImage(_ContextNode):
