@@ -16,6 +16,14 @@ from ..utils.image import decode_png_bytes, encode_png_bytes
 
 _STACK_CACHE: dict[tuple, object] = {}
 
+# Creation must be serialized: random init consumes the GLOBAL torch RNG
+# stream, so two threads racing through manual_seed(0)+init interleave
+# their draws and produce two different "identical" models (observed as a
+# master/worker weight mismatch in the HTTP integration test).
+import threading
+
+_STACK_LOCK = threading.Lock()
+
 
 def stable_text_seed(text: str) -> int:
     """Deterministic across processes (Python's hash() is salted per
@@ -59,10 +67,11 @@ class CheckpointLoader(_ContextNode):
         else:
             dtype = torch.bfloat16 if str(device).startswith("cuda") else torch.float32
         key = (str(ckpt_name), str(device))
-        if key not in _STACK_CACHE:
-            _STACK_CACHE[key] = create_diffusion_stack(
-                str(ckpt_name), device=device, dtype=dtype
-            )
+        with _STACK_LOCK:
+            if key not in _STACK_CACHE:
+                _STACK_CACHE[key] = create_diffusion_stack(
+                    str(ckpt_name), device=device, dtype=dtype
+                )
         stack = _STACK_CACHE[key]
         # MODEL and VAE are the stack handle + its VAE; CLIP is the stack
         # too (conditioning is synthesized from it).

@@ -98,6 +98,9 @@ def test_distributed_usdu_static_over_http(tmp_config, monkeypatch):
     g = torch.Generator().manual_seed(0)
     img = torch.rand(1, 32, 32, 3, generator=g)  # synthetic:32x32 seed 0
     ref = process_single_gpu(stack, cond, uncond, p, img)
-    assert torch.allclose(canvas, ref, atol=1e-4), (
+    # worker tiles cross the wire as PNG (uint8): tolerance covers that
+    # quantization, and nothing else (a conditioning/weight/seed mismatch
+    # shows up as O(0.5) diffs)
+    assert torch.allclose(canvas, ref, atol=0.02), (
         (canvas - ref).abs().max().item()
     )
