@@ -1,0 +1,125 @@
+"""Further end-to-end flows over real HTTP: load-balance routing and
+seed-parallel video with frame gathering."""
+
+import asyncio
+
+import pytest
+import torch
+from aiohttp.test_utils import TestClient, TestServer
+
+from comfyui_distributed_amd.server.app import DistributedServer
+from comfyui_distributed_amd.utils.config import load_config, save_config
+
+
+async def _pair(tmp_cfg):
+    worker_srv = DistributedServer(is_worker=True)
+    worker_client = TestClient(TestServer(worker_srv.build_app()))
+    await worker_client.start_server()
+    master_srv = DistributedServer()
+    master_srv.executor.context["preview_images"] = []
+    master_client = TestClient(TestServer(master_srv.build_app()))
+    await master_client.start_server()
+    cfg = load_config()
+    cfg["workers"] = [{
+        "id": "w1", "name": "worker1", "host": "127.0.0.1",
+        "port": worker_client.server.port, "cuda_device": 0,
+        "enabled": True, "type": "remote",
+    }]
+    cfg["master"]["host"] = "127.0.0.1"
+    cfg["master"]["port"] = master_client.server.port
+    save_config(cfg)
+    return worker_srv, worker_client, master_srv, master_client
+
+
+async def _teardown(wc, mc):
+    from comfyui_distributed_amd.server.network import close_client_session
+
+    await close_client_session()
+    await wc.close()
+    await mc.close()
+
+
+@pytest.mark.timeout(120)
+def test_load_balance_routes_whole_job_to_worker(tmp_config, monkeypatch):
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+
+    async def go():
+        wsrv, wc, msrv, mc = await _pair(tmp_config)
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "tiny"}},
+            "2": {"class_type": "CLIPTextEncode", "inputs": {"text": "x", "clip": ["1", 1]}},
+            "3": {"class_type": "EmptyLatentImage",
+                  "inputs": {"width": 16, "height": 16, "batch_size": 1}},
+            "4": {"class_type": "KSampler", "inputs": {
+                "model": ["1", 0], "seed": 1, "steps": 1, "cfg": 1.0,
+                "sampler_name": "euler", "scheduler": "karras",
+                "positive": ["2", 0], "negative": ["2", 0],
+                "latent_image": ["3", 0], "denoise": 1.0}},
+            "5": {"class_type": "VAEDecode", "inputs": {"samples": ["4", 0], "vae": ["1", 2]}},
+            "6": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["5", 0], "load_balance": True}},
+            "7": {"class_type": "PreviewImage", "inputs": {"images": ["6", 0]}},
+        }
+        # make the master look busy so the idle worker wins the route
+        await msrv.prompt_queue.put(({}, "pad", "pad1"))
+        msrv.prompt_queue.get_nowait()  # leave queue empty but master counts 0
+        resp = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "c", "enabled_worker_ids": ["w1"],
+        })
+        body = await resp.json()
+        previews = msrv.executor.context["preview_images"]
+        for _ in range(400):
+            if previews:
+                break
+            await asyncio.sleep(0.2)
+        await _teardown(wc, mc)
+        return body, previews
+
+    body, previews = asyncio.run(go())
+    # one participant won the whole job (master idle too -> round robin may
+    # pick either; both are valid routings)
+    assert body["participants"] in (["master"], ["w1"])
+    if body["participants"] == ["w1"]:
+        # master ran a delegate prompt and collected the worker's image
+        assert previews and previews[0].shape == (1, 16, 16, 3)
+
+
+@pytest.mark.timeout(180)
+def test_seed_parallel_video_over_http(tmp_config, monkeypatch):
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+
+    async def go():
+        wsrv, wc, msrv, mc = await _pair(tmp_config)
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "wan_tiny"}},
+            "2": {"class_type": "CLIPTextEncode", "inputs": {"text": "v", "clip": ["1", 1]}},
+            "3": {"class_type": "DistributedSeed", "inputs": {"seed": 11}},
+            "4": {"class_type": "WanVideoGenerate", "inputs": {
+                "model": ["1", 0], "positive": ["2", 0], "seed": ["3", 0],
+                "steps": 1, "cfg": 1.0, "width": 16, "height": 16, "frames": 5}},
+            "5": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["4", 0], "load_balance": False}},
+            "6": {"class_type": "ImageBatchDivider",
+                  "inputs": {"images": ["5", 0], "divide_by": 2}},
+            "7": {"class_type": "PreviewImage", "inputs": {"images": ["6", 0]}},
+        }
+        resp = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "c", "enabled_worker_ids": ["w1"],
+        })
+        assert resp.status == 200
+        previews = msrv.executor.context["preview_images"]
+        for _ in range(600):
+            if previews:
+                break
+            await asyncio.sleep(0.2)
+        await _teardown(wc, mc)
+        return previews
+
+    previews = asyncio.run(go())
+    assert previews, "video collect never completed"
+    # 5 master frames + 5 worker frames = 10, divider takes first half
+    assert previews[0].shape == (5, 16, 16, 3)
