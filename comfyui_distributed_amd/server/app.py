@@ -79,6 +79,7 @@ class DistributedServer:
         r.add_get("/panel", self.get_panel)
         r.add_get("/prompt", self.get_prompt)
         r.add_post("/prompt", self.post_prompt)
+        r.add_get("/object_info", self.get_object_info)
         r.add_post("/distributed/queue", self.post_queue)
         r.add_post("/distributed/job_complete", self.post_job_complete)
         r.add_post("/distributed/prepare_job", self.post_prepare_job)
@@ -154,6 +155,25 @@ class DistributedServer:
             {"exec_info": {"queue_remaining": self.queue_remaining()}}
         )
 
+    async def get_object_info(self, request):
+        """Node schemas (ComfyUI /object_info parity: the UI and remote
+        tooling discover node classes + input types here)."""
+        out = {}
+        for name in self.registry.names():
+            cls = self.registry.get(name)
+            try:
+                inputs = cls.INPUT_TYPES() if hasattr(cls, "INPUT_TYPES") else {}
+            except Exception:  # noqa: BLE001
+                inputs = {}
+            out[name] = {
+                "input": inputs,
+                "output": [str(t) for t in getattr(cls, "RETURN_TYPES", ())],
+                "output_name": [str(n) for n in getattr(cls, "RETURN_NAMES", ())],
+                "category": getattr(cls, "CATEGORY", ""),
+                "output_node": bool(getattr(cls, "OUTPUT_NODE", False)),
+            }
+        return web.json_response(out)
+
     async def post_prompt(self, request):
         data = await request.json()
         prompt = data.get("prompt")
@@ -198,14 +218,21 @@ class DistributedServer:
         return web.json_response({"status": "ready"})
 
     async def post_clear_memory(self, request):
+        """Unload cached models + free the allocator (reference
+        job_routes.py:160-203 unloads ComfyUI models on demand)."""
         import gc
 
         import torch
 
+        from ..graph.builtin_nodes import _STACK_CACHE, _STACK_LOCK
+
+        with _STACK_LOCK:
+            n = len(_STACK_CACHE)
+            _STACK_CACHE.clear()
         gc.collect()
         if torch.cuda.is_available():
             torch.cuda.empty_cache()
-        return web.json_response({"status": "ok"})
+        return web.json_response({"status": "ok", "unloaded_models": n})
 
     # ---- USDU tile endpoints ---------------------------------------------
 

@@ -210,3 +210,20 @@ def test_prompt_execution_end_to_end(client):
         assert srv.prompt_queue.qsize() == 0
 
     run(loop, go())
+
+
+def test_object_info_and_clear_memory(client):
+    srv, cl, loop = client
+
+    async def go():
+        r = await cl.get("/object_info")
+        info = await r.json()
+        assert "DistributedCollector" in info
+        assert "UltimateSDUpscaleDistributed" in info
+        assert info["DistributedSeed"]["output"] == ["INT"]
+        assert "hidden" in info["DistributedCollector"]["input"]
+        r = await cl.post("/distributed/clear_memory", json={})
+        data = await r.json()
+        assert data["status"] == "ok" and "unloaded_models" in data
+
+    run(loop, go())
