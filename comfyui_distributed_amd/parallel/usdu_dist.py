@@ -16,7 +16,7 @@ import torch
 
 from ..engine.usdu import USDUParams, blend_results, plan_for_image, sample_tiles
 from ..utils.logging import debug_log
-from .dist import DistContext, gather_tensor_lists
+from .dist import DistContext, broadcast_tensor, gather_tensor_lists
 from .tile_queue import TileQueue
 
 
@@ -27,12 +27,17 @@ def run_distributed_usdu(
     cond: dict,
     uncond: dict | None,
     params: USDUParams,
-    image: torch.Tensor,
+    image: torch.Tensor | None,
     job_id: str = "usdu",
+    broadcast_input: bool = False,
 ) -> torch.Tensor | None:
-    """All ranks call this with the same input image (broadcast upstream or
-    constructed identically). Returns the blended canvas on rank 0, None on
-    workers."""
+    """All ranks call this with the same input image (constructed
+    identically per rank), or — with ``broadcast_input=True`` — only rank 0
+    provides it and it is broadcast over RCCL/xGMI to the others. Returns
+    the blended canvas on rank 0, None on workers."""
+    if broadcast_input:
+        image = broadcast_tensor(ctx, image if ctx.is_master else None)
+    assert image is not None, "non-master ranks need broadcast_input=True"
     # .clone(): .to() is a no-copy alias when dtype/device already match,
     # and the blend pass mutates the canvas in place.
     canvas = image.to(stack.device, torch.float32).clone().contiguous()

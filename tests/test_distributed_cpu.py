@@ -89,6 +89,41 @@ def _body_collector(ctx, port):
     return None
 
 
+def _body_broadcast(ctx, port):
+    from comfyui_distributed_amd.parallel.dist import broadcast_tensor
+
+    src = torch.arange(24, dtype=torch.float32).reshape(2, 3, 4) if ctx.is_master else None
+    out = broadcast_tensor(ctx, src)
+    if not ctx.is_master:
+        return {"ok": torch.equal(out, torch.arange(24, dtype=torch.float32).reshape(2, 3, 4))}
+    return None
+
+
+def _body_usdu_broadcast(ctx, port):
+    import torch.distributed as tdist
+
+    from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
+    from comfyui_distributed_amd.models import create_diffusion_stack
+    from comfyui_distributed_amd.parallel.usdu_dist import run_distributed_usdu
+
+    store = tdist.TCPStore("127.0.0.1", port + 1000, ctx.world_size,
+                           ctx.is_master)
+    stack = create_diffusion_stack("tiny", seed=7)
+    cond = stack.make_conditioning(0)
+    p = USDUParams(seed=3, steps=1, cfg=1.0, denoise=0.5, tile_width=16,
+                   tile_height=16, padding=16, mask_blur=2, tile_batch=2)
+    img = None
+    if ctx.is_master:
+        g = torch.Generator().manual_seed(77)
+        img = torch.rand(1, 32, 32, 3, generator=g)
+    out = run_distributed_usdu(ctx, store, stack, cond, None, p, img,
+                               broadcast_input=True)
+    if ctx.is_master:
+        ref = process_single_gpu(stack, cond, None, p, img)
+        return {"dist": out, "ref": ref}
+    return None
+
+
 def _body_usdu(ctx, port):
     import torch.distributed as tdist
 
@@ -142,4 +177,15 @@ def test_distributed_usdu_equals_single_gpu():
     out = _spawn("_body_usdu", port=PORT_BASE + 3)
     dist_c, ref_c = out[0]["dist"], out[0]["ref"]
     assert dist_c.shape == ref_c.shape
+    assert torch.allclose(dist_c, ref_c, atol=1e-5)
+
+
+def test_broadcast_tensor_gloo():
+    out = _spawn("_body_broadcast", port=PORT_BASE + 4)
+    assert out[1]["ok"] is True
+
+
+def test_usdu_with_input_broadcast():
+    out = _spawn("_body_usdu_broadcast", port=PORT_BASE + 5)
+    dist_c, ref_c = out[0]["dist"], out[0]["ref"]
     assert torch.allclose(dist_c, ref_c, atol=1e-5)
