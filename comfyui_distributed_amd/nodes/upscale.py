@@ -97,6 +97,9 @@ class UltimateSDUpscaleDistributed:
             force_uniform_tiles=force_uniform_tiles, tiled_decode=tiled_decode,
         )
         enabled = json.loads(enabled_worker_ids or "[]")
+        if getattr(model, "family", "") in ("wan", "flow"):
+            self.validate_4n1_batch(upscaled_image.shape[0],
+                                    getattr(model, "family", ""))
         mode = self.determine_processing_mode(
             upscaled_image.shape[0], len(enabled), dynamic_threshold
         )
