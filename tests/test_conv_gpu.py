@@ -58,6 +58,22 @@ def test_conv_fused_silu(extmod):
     assert (y - ref).abs().max().item() / (ref.abs().max().item() + 1e-6) < 0.05
 
 
+@pytest.mark.skipif("DISTGPU_CONV_V3" not in __import__("os").environ,
+                    reason="v3 is opt-in: run with DISTGPU_CONV_V3=1")
+def test_conv_v3_matches_torch(extmod):
+    """Validates the experimental 64-deep-k conv (the env flag must be set
+    before the first conv_nhwc call in the process)."""
+    from comfyui_distributed_amd.ops import dispatch
+
+    torch.manual_seed(7)
+    conv = torch.nn.Conv2d(128, 128, 3, padding=1).cuda().to(torch.bfloat16)
+    x = (torch.randn(2, 128, 40, 40) / 4).cuda().to(torch.bfloat16)
+    xcl = x.contiguous(memory_format=torch.channels_last)
+    y = dispatch.conv2d_mfma(xcl, conv).float()
+    ref = F.conv2d(x.float(), conv.weight.float(), conv.bias.float(), padding=1)
+    assert (y - ref).abs().max().item() / ref.abs().max().item() < 0.05
+
+
 def test_groupnorm_nhwc_matches_torch(extmod):
     torch.manual_seed(1)
     x = torch.randn(2, 128, 20, 20)
