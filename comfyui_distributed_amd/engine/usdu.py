@@ -142,7 +142,10 @@ def sample_tiles(
                     start_from_latent=latents.float(),
                 )
             with trace_range("usdu.vae_decode"):
-                out_img = stack.vae.decode(latent_out.to(stack.dtype))
+                if params.tiled_decode:
+                    out_img = stack.vae.decode_tiled(latent_out.to(stack.dtype))
+                else:
+                    out_img = stack.vae.decode(latent_out.to(stack.dtype))
         for j, (t, b) in enumerate(chunk):
             results[(t, b)] = out_img[j : j + 1].float()
     return results

@@ -184,6 +184,49 @@ class VAE(nn.Module):
         mean = moments[:, : self.cfg.latent_channels]
         return mean * self.cfg.scale_factor
 
+    def decode_tiled(self, latents: torch.Tensor, tile: int = 64,
+                     overlap: int = 8) -> torch.Tensor:
+        """Latent-tiled decode for very large canvases (the USDU node's
+        ``tiled_decode`` option; reference delegates to ComfyUI's tiled
+        VAE). Tiles overlap in latent space and are feathered together
+        with linear ramps."""
+        b, c, h, w = latents.shape
+        if h <= tile and w <= tile:
+            return self.decode(latents)
+        ds = self.downscale
+        step = tile - overlap
+        out = None
+        weight = None
+        for y0 in range(0, h, step):
+            for x0 in range(0, w, step):
+                y1, x1 = min(y0 + tile, h), min(x0 + tile, w)
+                ys, xs = max(0, y1 - tile), max(0, x1 - tile)
+                piece = self.decode(latents[:, :, ys:y1, xs:x1])  # [B,ph,pw,3]
+                ph, pw = piece.shape[1], piece.shape[2]
+                if out is None:
+                    out = torch.zeros(b, h * ds, w * ds, piece.shape[3],
+                                      dtype=piece.dtype, device=piece.device)
+                    weight = torch.zeros(1, h * ds, w * ds, 1,
+                                         dtype=piece.dtype, device=piece.device)
+                # linear feather over the overlap band (pixel space)
+                ramp_y = torch.ones(ph, device=piece.device)
+                ramp_x = torch.ones(pw, device=piece.device)
+                band = overlap * ds
+                if ys > 0:
+                    ramp_y[:band] = torch.linspace(0, 1, band, device=piece.device)
+                if y1 < h:
+                    ramp_y[-band:] = torch.linspace(1, 0, band, device=piece.device)
+                if xs > 0:
+                    ramp_x[:band] = torch.linspace(0, 1, band, device=piece.device)
+                if x1 < w:
+                    ramp_x[-band:] = torch.linspace(1, 0, band, device=piece.device)
+                m = (ramp_y[:, None] * ramp_x[None, :])[None, :, :, None]
+                out[:, ys * ds:y1 * ds, xs * ds:x1 * ds] += piece * m
+                weight[:, ys * ds:y1 * ds, xs * ds:x1 * ds] += m
+                if y1 >= h and x1 >= w:
+                    break
+        return (out / weight.clamp_min(1e-8)).clamp(0, 1)
+
     def decode(self, latents: torch.Tensor) -> torch.Tensor:
         p = next(self.parameters())
         z = latents.to(p.device, p.dtype) / self.cfg.scale_factor

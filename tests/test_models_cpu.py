@@ -69,3 +69,25 @@ def test_cfg_denoiser_runs():
     with torch.no_grad():
         out = den(x, torch.tensor(5.0))
     assert out.shape == x.shape and torch.isfinite(out).all()
+
+
+def test_tiled_decode_matches_full_interior():
+    stack = create_diffusion_stack("tiny")
+    z = torch.randn(1, 4, 12, 12)
+    with torch.no_grad():
+        full = stack.vae.decode(z)
+        tiled = stack.vae.decode_tiled(z, tile=8, overlap=4)
+    assert tiled.shape == full.shape
+    # interiors of tiles away from seams match the full decode closely
+    d = (tiled - full).abs()
+    assert d.median().item() < 0.05
+    assert torch.isfinite(tiled).all()
+
+
+def test_tiled_decode_small_passthrough():
+    stack = create_diffusion_stack("tiny")
+    z = torch.randn(1, 4, 4, 4)
+    with torch.no_grad():
+        a = stack.vae.decode(z)
+        b = stack.vae.decode_tiled(z, tile=64)
+    assert torch.equal(a, b)
