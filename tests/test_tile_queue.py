@@ -118,3 +118,66 @@ def test_scheduler_no_heartbeat_worker_never_started():
     # no heartbeat recorded at all -> worker kept (may still be starting;
     # reference gives workers a job-ready grace window)
     assert sched.check_and_requeue() == []
+
+
+def test_concurrent_pull_exactly_once():
+    """8 threads hammer one queue: every task is delivered exactly once."""
+    import threading
+
+    store = LocalStore()
+    master = TileQueue(store, "soak", 0)
+    master.init_job(500)
+    seen = []
+    lock = threading.Lock()
+
+    def puller(rank):
+        q = TileQueue(store, "soak", rank)
+        got = []
+        while True:
+            i = q.pop()
+            if i is None:
+                break
+            got.append(i)
+            q.mark_done(i)
+        with lock:
+            seen.extend(got)
+
+    threads = [threading.Thread(target=puller, args=(r,)) for r in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert sorted(seen) == list(range(500))
+    assert master.completed() == 500 and master.is_complete()
+
+
+def test_concurrent_requeue_no_duplication():
+    """Requeued tasks are also delivered exactly once under contention."""
+    import threading
+
+    store = LocalStore()
+    master = TileQueue(store, "rq", 0)
+    master.init_job(100)
+    # pre-assign and requeue 20 tasks as if a worker died
+    dead = [master.pop() for _ in range(20)]
+    master.requeue_tasks(dead)
+    seen = []
+    lock = threading.Lock()
+
+    def puller(rank):
+        q = TileQueue(store, "rq", rank)
+        got = []
+        while True:
+            i = q.pop()
+            if i is None:
+                break
+            got.append(i)
+        with lock:
+            seen.extend(got)
+
+    threads = [threading.Thread(target=puller, args=(r,)) for r in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert sorted(seen) == sorted(dead + [i for i in range(100) if i not in dead])
