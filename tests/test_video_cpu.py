@@ -43,3 +43,28 @@ def test_wan_frames_feed_batch_divider():
     frames = generate_video(stack, cond, None, p)
     parts = ImageBatchDivider().divide_batch(frames, 2)
     assert parts[0].shape[0] == 3 and parts[1].shape[0] == 2
+
+
+def test_per_frame_tile_upscale_pipeline():
+    """BASELINE config 5 composition: WAN frames -> per-frame distributed
+    tile upscale with an image diffusion stack (frames ride the batch dim
+    through the USDU engine)."""
+    from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
+    from comfyui_distributed_amd.ops import dispatch
+
+    wan = create_diffusion_stack("wan_tiny", seed=1)
+    cond_v = wan.make_conditioning(0)
+    frames = generate_video(wan, cond_v, None,
+                            VideoGenParams(seed=2, steps=1, cfg=1.0,
+                                           width=16, height=16, frames=5))
+    # 2x Lanczos pre-upscale (the USDU "No Upscale" contract)
+    up = dispatch.extract_resize(frames, (0, 0, 16, 16), 32, 32)
+    assert up.shape == (5, 32, 32, 3)
+
+    sd = create_diffusion_stack("tiny", seed=3)
+    cond_i = sd.make_conditioning(0)
+    p = USDUParams(seed=4, steps=1, cfg=1.0, denoise=0.4, tile_width=16,
+                   tile_height=16, padding=16, mask_blur=2, tile_batch=4)
+    out = process_single_gpu(sd, cond_i, None, p, up)
+    assert out.shape == (5, 32, 32, 3)
+    assert torch.isfinite(out).all()
