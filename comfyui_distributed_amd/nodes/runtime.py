@@ -44,6 +44,9 @@ class NodeRuntime:
     def interrupt(self):
         self._interrupted = True
 
+    def clear_interrupt(self):
+        self._interrupted = False
+
     def throw_if_interrupted(self):
         if self._interrupted:
             raise InterruptedError("processing interrupted by user")

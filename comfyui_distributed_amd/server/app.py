@@ -125,6 +125,9 @@ class DistributedServer:
         while True:
             prompt, client_id, prompt_id = await self.prompt_queue.get()
             self.executing = True
+            from ..nodes.runtime import get_runtime
+
+            get_runtime().clear_interrupt()  # interrupts are per-prompt
             try:
                 await loop.run_in_executor(None, self.executor.execute, prompt)
                 log(f"prompt {prompt_id} done (client {client_id})")

@@ -172,3 +172,25 @@ def test_pass_through_returns_unchanged():
     imgs = torch.rand(2, 4, 4, 3)
     out_imgs, out_audio = node.run(imgs, multi_job_id="j", pass_through=True)
     assert out_imgs is imgs
+
+
+def test_interrupt_aborts_collect_and_clears():
+    from comfyui_distributed_amd.utils import constants
+
+    rt = NodeRuntime()
+    set_runtime(rt)
+    node = DistributedCollectorNode()
+
+    async def go():
+        await rt.job_state.ensure_queue("ji")
+        rt.interrupt()
+        with pytest.raises(InterruptedError):
+            await node.collect_on_master(
+                torch.zeros(1, 4, 4, 3), None, "ji", json.dumps(["w1"]), False
+            )
+        # queue cleaned up on abort
+        assert "ji" not in rt.job_state.pending_jobs
+        rt.clear_interrupt()
+        rt.throw_if_interrupted()  # no raise after clear
+
+    asyncio.run(go())
