@@ -198,6 +198,14 @@ class DistributedServer:
         result = await orchestrate_distributed_execution(
             payload, self.job_state, self.enqueue_local
         )
+        # superset of the reference's response shape
+        # (docs/comfyui-distributed-api.md): prompt_id + worker_count +
+        # auto_prepare_supported alongside our participants/job_ids
+        result.setdefault("prompt_id", result.get("master_prompt_id", ""))
+        result["worker_count"] = sum(
+            1 for p in result.get("participants", []) if p != "master"
+        )
+        result["auto_prepare_supported"] = True
         return web.json_response(result)
 
     async def post_job_complete(self, request):

@@ -14,7 +14,8 @@ import asyncio
 import itertools
 
 from ..graph import PromptGraph, transform
-from ..graph.prompt import DISTRIBUTED_OUTPUT_CLASSES, NODE_CLASS_COLLECTOR
+from ..graph.prompt import (DISTRIBUTED_OUTPUT_CLASSES, NODE_CLASS_COLLECTOR,
+                            NODE_CLASS_UPSCALE)
 from ..utils import constants
 from ..utils.config import enabled_workers, is_master_delegate_only, load_config
 from ..utils.logging import log, trace_debug
@@ -123,6 +124,12 @@ async def orchestrate_distributed_execution(
     if delegate and not online:
         trace_debug(trace, "delegate-only with zero online workers — master fallback")
         delegate = False
+    if delegate and graph.nodes_of_class(NODE_CLASS_UPSCALE):
+        # reference limitation kept: delegate-only does not support USDU
+        # (the master must coordinate the tile job); fall back to full
+        # master participation (docs/comfyui-distributed-api.md)
+        trace_debug(trace, "delegate-only with USDU — master fallback")
+        delegate = False
     enabled_ids = [str(w["id"]) for w in online]
 
     # ---- load balancing ---------------------------------------------------
@@ -139,8 +146,9 @@ async def orchestrate_distributed_execution(
         chosen = select_least_busy(candidates)
         trace_debug(trace, f"load_balance chose {chosen}")
         if chosen == "master":
-            await enqueue_local(payload.prompt, payload.client_id)
-            return {"status": "queued", "participants": ["master"], "job_ids": {}}
+            pid = await enqueue_local(payload.prompt, payload.client_id)
+            return {"status": "queued", "participants": ["master"],
+                    "job_ids": {}, "master_prompt_id": pid}
         worker = next(w for w in online if str(w["id"]) == chosen)
         pruned = transform.prune_prompt_for_worker(graph)
         job_id_map = transform.generate_job_id_map(graph, prefix=None)
@@ -160,18 +168,21 @@ async def orchestrate_distributed_execution(
         )
         ok = await dispatch_worker_prompt(worker, wp.raw, payload.client_id)
         if not ok:
-            await enqueue_local(payload.prompt, payload.client_id)
-            return {"status": "queued", "participants": ["master"], "job_ids": {}}
-        await enqueue_local(master_prompt.raw, payload.client_id)
+            pid = await enqueue_local(payload.prompt, payload.client_id)
+            return {"status": "queued", "participants": ["master"],
+                    "job_ids": {}, "master_prompt_id": pid}
+        pid = await enqueue_local(master_prompt.raw, payload.client_id)
         return {"status": "queued", "participants": [chosen],
+                "master_prompt_id": pid,
                 "job_ids": job_id_map}
 
     # ---- regular fan-out --------------------------------------------------
     job_id_map = transform.generate_job_id_map(graph, prefix=None)
     if not graph.nodes_of_class(*DISTRIBUTED_OUTPUT_CLASSES) or not online:
         # nothing distributed (or nobody to distribute to): run locally
-        await enqueue_local(payload.prompt, payload.client_id)
-        return {"status": "queued", "participants": ["master"], "job_ids": {}}
+        pid = await enqueue_local(payload.prompt, payload.client_id)
+        return {"status": "queued", "participants": ["master"], "job_ids": {},
+                "master_prompt_id": pid}
 
     for jid in job_id_map.values():
         await job_state.ensure_queue(jid)
@@ -204,9 +215,9 @@ async def orchestrate_distributed_execution(
     failed = [wid for wid, ok in dispatched if not ok]
     if failed:
         log(f"orchestration: dispatch failed for {failed}")
-    await enqueue_local(master_prompt.raw, payload.client_id)
+    pid = await enqueue_local(master_prompt.raw, payload.client_id)
     participants = (["master"] if not delegate else []) + [
         wid for wid, ok in dispatched if ok
     ]
     return {"status": "queued", "participants": participants,
-            "job_ids": job_id_map}
+            "job_ids": job_id_map, "master_prompt_id": pid}

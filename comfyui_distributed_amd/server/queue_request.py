@@ -18,6 +18,9 @@ class QueueRequestPayload:
     delegate_master: bool = False
     trace_execution_id: str | None = None
     workflow: dict | None = None
+    #: wire-compat field; orchestration always runs with auto-prepare
+    #: semantics (reference api/queue_request.py:12-24)
+    auto_prepare: bool = True
 
 
 def parse_queue_request_payload(data: dict) -> QueueRequestPayload:
@@ -37,7 +40,11 @@ def parse_queue_request_payload(data: dict) -> QueueRequestPayload:
     client_id = data.get("client_id")
     if not client_id:
         raise QueueRequestError("missing 'client_id'")
+    auto_prepare = data.get("auto_prepare", True)
+    if not isinstance(auto_prepare, bool):
+        raise QueueRequestError("auto_prepare must be a boolean when provided")
     return QueueRequestPayload(
+        auto_prepare=auto_prepare,
         prompt=prompt,
         client_id=str(client_id),
         enabled_worker_ids=[str(w) for w in enabled],

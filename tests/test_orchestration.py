@@ -164,3 +164,30 @@ def test_select_least_busy_round_robin():
              ("b", {"exec_info": {"queue_remaining": 0}})]
     picks = {orchestration.select_least_busy(cands) for _ in range(10)}
     assert picks == {"a", "b"}  # round-robin among idle
+
+
+def test_auto_prepare_field_validation():
+    with pytest.raises(QueueRequestError):
+        parse_queue_request_payload({
+            "prompt": dist_prompt(), "client_id": "c",
+            "enabled_worker_ids": [], "auto_prepare": "yes",
+        })
+    p = parse_queue_request_payload({
+        "prompt": dist_prompt(), "client_id": "c",
+        "enabled_worker_ids": [], "auto_prepare": False,
+    })
+    assert p.auto_prepare is False
+
+
+def test_delegate_with_usdu_falls_back_to_master(two_worker_config, monkeypatch):
+    prompt = dist_prompt()
+    prompt["9"] = {"class_type": "UltimateSDUpscaleDistributed",
+                   "inputs": {"upscaled_image": ["2", 0], "seed": 1}}
+    payload = QueueRequestPayload(prompt=prompt, client_id="c",
+                                  enabled_worker_ids=["w1"],
+                                  delegate_master=True)
+    probes = {"w1": {"exec_info": {"queue_remaining": 0}}}
+    result, calls, _ = run_orch(payload, monkeypatch, probes)
+    # master participates despite delegate flag (reference limitation kept)
+    assert "master" in result["participants"]
+    assert "master_prompt_id" in result
