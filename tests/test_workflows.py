@@ -21,6 +21,7 @@ def load_wf(name):
     "seed_parallel_txt2img.json",
     "distributed_upscale.json",
     "distributed_wan_video.json",
+    "distributed_upscale_video.json",
 ])
 def test_workflows_validate(name):
     validate_prompt(load_wf(name), default_registry())
