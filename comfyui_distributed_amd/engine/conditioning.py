@@ -11,6 +11,8 @@ extras:
   control_hint       [B, C, H, W] pixel-space hint -> crop + resize
   mask               [B, H, W] or [B,1,H,W] pixel mask -> crop + resize
   area               (x, y, w, h) pixel rect -> intersect + rebase
+  gligen             list of (emb [L,D], (x, y, w, h) pixel box) -> clip +
+                     rebase each box to the tile; boxes fully outside drop
   reference_latents  [B, C, H/8, W/8] -> crop in latent space
 Non-spatial keys pass through by reference (no copies of shared tensors —
 the reference's clone_conditioning shares ControlNet models the same way).
@@ -23,7 +25,7 @@ import torch.nn.functional as F
 
 from ..utils.usdu_math import resize_region
 
-SPATIAL_KEYS = ("control_hint", "mask", "area", "reference_latents")
+SPATIAL_KEYS = ("control_hint", "mask", "area", "gligen", "reference_latents")
 
 
 def crop_tile_conditioning(
@@ -72,6 +74,25 @@ def crop_tile_conditioning(
                 (x2 - x1, y2 - y1), (pw, ph),
             )
             out["area"] = (rx1, ry1, rx2 - rx1, ry2 - ry1)
+
+    gligen = cond.get("gligen")
+    if gligen is not None:
+        # reference crop_gligen (usdu_utils.py:335-378): clip each position
+        # box to the tile, rebase to tile-local pixels, scale to processing
+        # size; boxes with no overlap are dropped entirely
+        kept = []
+        for emb, box in gligen:
+            bx, by, bw, bh = box
+            ix1, iy1 = max(bx, x1), max(by, y1)
+            ix2, iy2 = min(bx + bw, x2), min(by + bh, y2)
+            if ix2 <= ix1 or iy2 <= iy1:
+                continue
+            rx1, ry1, rx2, ry2 = resize_region(
+                (ix1 - x1, iy1 - y1, ix2 - x1, iy2 - y1),
+                (x2 - x1, y2 - y1), (pw, ph),
+            )
+            kept.append((emb, (rx1, ry1, rx2 - rx1, ry2 - ry1)))
+        out["gligen"] = kept or None
 
     ref = cond.get("reference_latents")
     if ref is not None:

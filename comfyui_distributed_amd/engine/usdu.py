@@ -88,13 +88,17 @@ def sample_tiles(
     base_denoiser = CFGDenoiser(stack.unet, schedule, cond, uncond, params.cfg)
 
     from .conditioning import SPATIAL_KEYS, crop_tile_conditioning
+    from .model_patch import crop_model_patches, stack_has_patches
 
-    has_spatial = any(k in (cond or {}) for k in SPATIAL_KEYS) or any(
-        k in (uncond or {}) for k in SPATIAL_KEYS
+    has_spatial = (
+        any(k in (cond or {}) for k in SPATIAL_KEYS)
+        or any(k in (uncond or {}) for k in SPATIAL_KEYS)
+        or stack_has_patches(stack)
     )
     work = [(t, b) for t in sorted(tile_indices) for b in range(B)]
-    # spatial conditioning differs per tile -> tiles can't share a sampler
-    # batch (reference processes per tile for the same reason)
+    # spatial conditioning / model patches differ per tile -> tiles can't
+    # share a sampler batch (reference processes per tile for the same
+    # reason: tile_ops.py crop context + crop_model_patch.py)
     step = 1 if has_spatial else max(params.tile_batch, 1)
     results: dict[tuple[int, int], torch.Tensor] = {}
     for i in range(0, len(work), step):
@@ -132,7 +136,10 @@ def sample_tiles(
                     for t, b in chunk
                 ]
             ).to(latents.device)
-            with trace_range("usdu.sample"):
+            plan0 = plans[chunk[0][0]]
+            with trace_range("usdu.sample"), crop_model_patches(
+                stack, plan0.crop_region, plan0.canvas_size, plan0.process_size
+            ):
                 latent_out = sample(
                     denoiser,
                     noise,
