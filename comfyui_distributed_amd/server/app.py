@@ -106,6 +106,9 @@ class DistributedServer:
         r.add_post("/distributed/launch_worker", self.post_launch_worker)
         r.add_post("/distributed/stop_worker", self.post_stop_worker)
         r.add_get("/distributed/managed_workers", self.get_managed_workers)
+        r.add_get("/distributed/worker_status", self.get_worker_status)
+        r.add_post("/distributed/auto_populate_workers",
+                   self.post_auto_populate_workers)
         r.add_get("/distributed/worker_log", self.get_worker_log)
         r.add_get("/distributed/local_log", self.get_local_log)
         r.add_post("/distributed/load_image", self.post_load_image)
@@ -521,6 +524,58 @@ class DistributedServer:
 
     async def get_tunnel_status(self, request):
         return web.json_response(self.tunnel.status())
+
+    async def get_worker_status(self, request):
+        """Master-side worker probe for the panel (browsers can't
+        cross-origin-probe workers; reference worker_routes.py:536-603
+        serves the same role)."""
+        wid = request.query.get("id")
+        worker = get_worker_by_id(load_config(), wid) if wid else None
+        if worker is None:
+            return _err("unknown worker", status=404)
+        from .workers import is_process_alive, load_managed_pid
+
+        info = await network.probe_worker(network.build_worker_url(worker))
+        pid = load_managed_pid(str(wid))
+        return web.json_response({
+            "id": str(wid),
+            "online": info is not None,
+            "queue_remaining": (info or {}).get("exec_info", {}).get(
+                "queue_remaining"),
+            "managed": pid is not None,
+            "pid_alive": bool(pid and is_process_alive(pid)),
+        })
+
+    async def post_auto_populate_workers(self, request):
+        """One local worker per GPU beyond the master's, first run only
+        (reference web/masterDetection.js:3-147 does this client-side on
+        first launch; here it is a server action the panel calls)."""
+        import torch
+
+        data = await request.json() if request.can_read_body else {}
+        force = bool(data.get("force"))
+        n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        created = []
+        async with config_transaction() as cfg:
+            if cfg["settings"].get("has_auto_populated_workers") and not force:
+                return web.json_response({"status": "already_populated",
+                                          "created": []})
+            master_dev = cfg["master"].get("cuda_device", 0)
+            base_port = int(cfg["master"].get("port", 8188))
+            existing_devs = {w.get("cuda_device") for w in cfg["workers"]}
+            for dev in range(n_gpus):
+                if dev == master_dev or dev in existing_devs:
+                    continue
+                wid = f"auto_gpu{dev}"
+                cfg["workers"].append({
+                    "id": wid, "name": f"GPU {dev}", "host": "",
+                    "port": base_port + 1 + dev, "cuda_device": dev,
+                    "enabled": True, "type": "local", "extra_args": "",
+                })
+                created.append(wid)
+            cfg["settings"]["has_auto_populated_workers"] = True
+        return web.json_response({"status": "ok", "created": created,
+                                  "gpu_count": n_gpus})
 
     # ---- worker process management ----------------------------------------
 

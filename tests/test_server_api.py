@@ -227,3 +227,42 @@ def test_object_info_and_clear_memory(client):
         assert data["status"] == "ok" and "unloaded_models" in data
 
     run(loop, go())
+
+
+def test_worker_status_and_auto_populate(client):
+    srv, cl, loop = client
+
+    async def go():
+        # unknown worker -> 404
+        r = await cl.get("/distributed/worker_status?id=ghost")
+        assert r.status == 404
+        # create an offline worker
+        r = await cl.post("/distributed/config/update_worker", json={
+            "id": "ws1", "name": "ws1", "port": 1, "host": "127.0.0.1",
+            "type": "remote", "enabled": True})
+        assert r.status == 200
+        r = await cl.get("/distributed/worker_status?id=ws1")
+        assert r.status == 200
+        st = await r.json()
+        assert st["online"] is False and st["managed"] is False
+        # a worker pointing at this very server probes online
+        r = await cl.post("/distributed/config/update_worker", json={
+            "id": "ws2", "name": "self", "port": cl.server.port,
+            "host": "127.0.0.1", "type": "remote", "enabled": True})
+        assert r.status == 200
+        r = await cl.get("/distributed/worker_status?id=ws2")
+        st = await r.json()
+        assert st["online"] is True and st["queue_remaining"] == 0
+
+        # auto-populate: 0 GPUs on CPU box -> nothing created, flag set
+        r = await cl.post("/distributed/auto_populate_workers", json={})
+        body = await r.json()
+        assert body["status"] == "ok" and body["created"] == []
+        r = await cl.post("/distributed/auto_populate_workers", json={})
+        body = await r.json()
+        assert body["status"] == "already_populated"
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+
+    run(loop, go())
