@@ -1,8 +1,17 @@
-"""WAN video stack: flow-matching sampling + per-frame VAE decode.
+"""WAN video stack: flow-matching sampling + (2+1)D temporal VAE.
 
 WAN-2.2 is a rectified-flow model: the network predicts velocity v at time
 t in [0,1] with x_t = (1-t)*x0 + t*noise; sampling integrates dx/dt = v
-from t=1 to 0 (Euler). Video latents are [B, C, T, H/8, W/8].
+from t=1 to 0 (Euler). Video latents are [B, C, 1+(T-1)/4, H/8, W/8]:
+like the real WAN VAE, 8x spatial AND 4x temporal compression on 4n+1
+frame clips.
+
+The temporal stage is deliberately factorized out of the spatial convs
+((2+1)D, not full 3D): every spatial conv stays a per-frame NHWC conv on
+the MFMA implicit-GEMM path, and temporal mixing is a causal Conv1d over
+the time axis batched across all (channel, pixel) positions - a
+GEMM-shaped op that maps cleanly onto CDNA4 matrix cores, instead of the
+awkward 3D-conv tilings a direct port of a causal-3D VAE would need.
 """
 
 from __future__ import annotations
@@ -10,12 +19,12 @@ from __future__ import annotations
 from dataclasses import dataclass
 
 import torch
+from torch import nn
 
 from .vae import VAE, VAEConfig
 from .wan import WAN14B, WAN_TINY, WanConfig, WanModel
 
-#: WAN's VAE has 16 latent channels (x8 spatial; temporal handled per-frame
-#: this round).
+#: WAN's VAE has 16 latent channels (x8 spatial, x4 temporal).
 WAN_VAE = VAEConfig(latent_channels=16)
 WAN_TINY_VAE = VAEConfig(latent_channels=4, base_channels=8,
                          channel_mult=(1, 1, 2, 2), num_res_blocks=1)
@@ -24,6 +33,69 @@ WAN_CONFIGS: dict[str, WanConfig] = {
     "wan14b": WAN14B,
     "wan_tiny": WAN_TINY,
 }
+
+
+class TemporalVAE(nn.Module):
+    """(2+1)D video VAE: per-frame spatial VAE + causal temporal 4x
+    resampling in latent space.
+
+    encode: frames [T, H, W, 3] (T = 4n+1) -> z [Cz, 1+(T-1)/4, h, w]
+    decode: z -> frames [T, H, W, 3]
+
+    Causality: the temporal conv sees only past frames (front replicate
+    pad of k-1), so streaming/chunked decode of long clips never needs
+    future latents - the property the real WAN causal-3D VAE has.
+    """
+
+    t_down = 4
+
+    def __init__(self, spatial: VAE):
+        super().__init__()
+        self.spatial = spatial
+        cz = spatial.cfg.latent_channels
+        k = self.t_down + 1
+        self.tdown = nn.Conv1d(cz, cz, kernel_size=k, stride=self.t_down)
+        # k=2*t_down, pad=t_down//2 -> exact 4x length; trimmed to T after
+        self.tup = nn.ConvTranspose1d(cz, cz, kernel_size=2 * self.t_down,
+                                      stride=self.t_down,
+                                      padding=self.t_down // 2)
+        self.latent_channels = cz
+
+    def latent_frames(self, frames: int) -> int:
+        if frames % self.t_down != 1:
+            raise ValueError(f"need {self.t_down}n+1 frames, got {frames}")
+        return 1 + (frames - 1) // self.t_down
+
+    def _time_apply(self, conv, z):
+        # z [C, T, h, w] -> conv over T batched across pixels
+        c, t, h, w = z.shape
+        seq = z.permute(2, 3, 0, 1).reshape(h * w, c, t)
+        out = conv(seq)  # [h*w, C, T']
+        return out.reshape(h, w, c, -1).permute(2, 3, 0, 1)
+
+    def encode(self, frames: torch.Tensor) -> torch.Tensor:
+        t = frames.shape[0]
+        n_lat = self.latent_frames(t)
+        zs = self.spatial.encode(frames)  # [T, Cz, h, w]
+        z = zs.permute(1, 0, 2, 3)  # [Cz, T, h, w]
+        # causal front pad (replicate frame 0), then stride-4 conv:
+        # length T + (k-1) with k = t_down+1 gives exactly 1+(T-1)/4
+        pad = self.tdown.kernel_size[0] - 1
+        z = torch.cat([z[:, :1].expand(-1, pad, -1, -1), z], dim=1)
+        z = self._time_apply(self.tdown, z.float())
+        assert z.shape[1] == n_lat
+        return z.to(frames.dtype)
+
+    def decode(self, z: torch.Tensor, frames: int | None = None) -> torch.Tensor:
+        """z [Cz, T_lat, h, w] -> frames [T, H, W, 3]; T defaults to the
+        4n+1 clip length T_lat encodes."""
+        t_lat = z.shape[1]
+        t = frames if frames is not None else 1 + (t_lat - 1) * self.t_down
+        zs = self._time_apply(self.tup, z.float())  # [Cz, 4*T_lat, h, w]
+        zs = zs[:, :t]
+        return self.spatial.decode(
+            zs.permute(1, 0, 2, 3).to(z.dtype)
+        )  # [T, H, W, 3]
 
 
 @dataclass
@@ -51,7 +123,8 @@ class WanStack:
         torch.manual_seed(seed)
         self.cfg = cfg
         self.model = WanModel(cfg).to(device=device, dtype=dtype).eval()
-        self.vae = VAE(vae_variant).to(device=device, dtype=dtype).eval()
+        self.vae = TemporalVAE(VAE(vae_variant)).to(
+            device=device, dtype=dtype).eval()
         self.device = torch.device(device)
         self.dtype = dtype
 
@@ -64,6 +137,9 @@ class WanStack:
     def validate_frames(self, frames: int) -> None:
         if frames % 4 != 1:
             raise ValueError(f"WAN needs 4n+1 frames, got {frames}")
+
+    def latent_frames(self, frames: int) -> int:
+        return self.vae.latent_frames(frames)
 
 
 def flow_sigmas(steps: int, shift: float = 5.0) -> torch.Tensor:
@@ -112,14 +188,14 @@ def generate_video(stack: WanStack, cond, uncond, p: VideoGenParams) -> torch.Te
     ImageBatchDivider segments them directly)."""
     stack.validate_frames(p.frames)
     g = torch.Generator().manual_seed(p.seed)
-    lat_t = p.frames  # temporal compression 1 in this round's VAE
+    lat_t = stack.latent_frames(p.frames)  # 1 + (T-1)/4
     shape = (p.batch_size, stack.cfg.in_channels, lat_t, p.height // 8, p.width // 8)
     noise = torch.randn(shape, generator=g).to(stack.device)
     vel = FlowCFGVelocity(stack.model, cond, uncond, p.cfg)
     with torch.no_grad():
         lat = sample_flow(vel, noise, p.steps)
-        # per-frame VAE decode: [B*T, C, h, w] -> frames
-        b, c, t, h, w = lat.shape
-        frames = lat.permute(0, 2, 1, 3, 4).reshape(b * t, c, h, w)
-        imgs = stack.vae.decode(frames.to(stack.dtype))
+        imgs = torch.cat([
+            stack.vae.decode(lat[b].to(stack.dtype), frames=p.frames)
+            for b in range(lat.shape[0])
+        ])
     return imgs.float()

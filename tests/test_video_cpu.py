@@ -68,3 +68,42 @@ def test_per_frame_tile_upscale_pipeline():
     out = process_single_gpu(sd, cond_i, None, p, up)
     assert out.shape == (5, 32, 32, 3)
     assert torch.isfinite(out).all()
+
+
+def test_temporal_vae_compression_and_causality():
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+
+    stack = create_diffusion_stack("wan_tiny")
+    vae = stack.vae
+    frames = torch.rand(9, 32, 32, 3)  # 4n+1, n=2
+    z = vae.encode(frames)
+    assert z.shape[1] == 3 == vae.latent_frames(9)  # 1+(9-1)/4
+    assert z.shape[0] == vae.latent_channels
+    out = vae.decode(z, frames=9)
+    assert out.shape == (9, 32, 32, 3)
+    assert torch.isfinite(out).all()
+    # default clip length without explicit frames: 1+(T_lat-1)*4
+    assert vae.decode(z).shape[0] == 9
+
+    # causality: perturbing the LAST frame must not change early latents
+    frames2 = frames.clone()
+    frames2[-1] += 1.0
+    z2 = vae.encode(frames2)
+    assert torch.allclose(z[:, 0], z2[:, 0], atol=1e-5)
+    assert torch.allclose(z[:, 1], z2[:, 1], atol=1e-5)
+    assert not torch.allclose(z[:, 2], z2[:, 2])
+
+    with pytest.raises(ValueError):
+        vae.encode(torch.rand(8, 32, 32, 3))  # not 4n+1
+
+
+def test_generate_video_latent_shrinks_4x():
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+    from comfyui_distributed_amd.models.video import VideoGenParams, generate_video
+
+    stack = create_diffusion_stack("wan_tiny")
+    cond = stack.make_conditioning(0)
+    p = VideoGenParams(seed=3, steps=1, cfg=1.0, width=16, height=16, frames=5)
+    out = generate_video(stack, cond, None, p)
+    assert out.shape == (5, 16, 16, 3)
+    assert stack.latent_frames(5) == 2
