@@ -68,3 +68,17 @@ def test_smoke_entry():
     import __graft_entry__
 
     __graft_entry__.smoke()
+
+
+def test_wan_tiny_temporal_vae_gpu():
+    from comfyui_distributed_amd.models import create_diffusion_stack
+    from comfyui_distributed_amd.models.video import VideoGenParams, generate_video
+
+    stack = create_diffusion_stack("wan_tiny", device="cuda:0",
+                                   dtype=torch.bfloat16)
+    cond = stack.make_conditioning(0)
+    p = VideoGenParams(seed=3, steps=1, cfg=1.0, width=32, height=32, frames=9)
+    assert stack.latent_frames(9) == 3  # causal 4x temporal compression
+    out = generate_video(stack, cond, None, p)
+    assert out.shape == (9, 32, 32, 3)
+    assert torch.isfinite(out).all()
