@@ -77,19 +77,62 @@ def validate_prompt(prompt: dict, registry: NodeRegistry) -> None:
 
 
 class Executor:
-    """Evaluates a prompt graph; caches per-node outputs within one run."""
+    """Evaluates a prompt graph; caches per-node outputs within one run AND
+    across consecutive runs (ComfyUI-style execution caching, which the
+    reference inherits from its host: a re-submitted workflow only
+    re-executes nodes whose input spec changed — see ComfyUI
+    execution.py's caching the reference's nodes rely on).
+
+    Cross-run reuse is keyed by a recursive spec fingerprint
+    (class_type + widget values + upstream fingerprints), never by tensor
+    contents, so it costs O(graph) per run. Output/side-effect nodes
+    (OUTPUT_NODE = True: SaveImage, PreviewImage, collectors, USDU) always
+    re-run — their upstream work is still reused."""
 
     def __init__(self, registry: NodeRegistry | None = None,
                  context: dict[str, Any] | None = None):
         self.registry = registry or default_registry()
         #: shared objects node bodies may need (device, model cache, ...)
         self.context = context or {}
+        #: previous run's {fingerprint: outputs}; replaced wholesale each
+        #: run so memory stays bounded by one workflow's outputs
+        self._run_cache: dict[str, tuple] = {}
+
+    def _fingerprints(self, graph: PromptGraph) -> dict[str, str]:
+        import hashlib
+        import json
+
+        fps: dict[str, str] = {}
+
+        def fp(nid: str) -> str:
+            if nid in fps:
+                return fps[nid]
+            node = graph.node(nid)
+            widgets = {}
+            links = {}
+            for name, value in node.get("inputs", {}).items():
+                if is_link(value):
+                    links[name] = (fp(str(value[0])), int(value[1]))
+                else:
+                    widgets[name] = value
+            spec = json.dumps(
+                [node["class_type"], widgets, links],
+                sort_keys=True, default=repr,
+            )
+            fps[nid] = hashlib.md5(spec.encode()).hexdigest()
+            return fps[nid]
+
+        for nid in graph.node_ids():
+            fp(nid)
+        return fps
 
     def execute(self, prompt: dict) -> dict[str, tuple]:
         """Run the graph; returns {node_id: outputs tuple} for OUTPUT_NODEs
         and every executed node."""
         validate_prompt(prompt, self.registry)
         graph = PromptGraph(prompt)
+        fps = self._fingerprints(graph)
+        next_run_cache: dict[str, tuple] = {}
         cache: dict[str, tuple] = {}
 
         def eval_node(nid: str) -> tuple:
@@ -97,6 +140,12 @@ class Executor:
                 return cache[nid]
             node = graph.node(nid)
             cls = self.registry.get(node["class_type"])
+            is_output = bool(getattr(cls, "OUTPUT_NODE", False))
+            if not is_output and fps[nid] in self._run_cache:
+                out = self._run_cache[fps[nid]]
+                cache[nid] = out
+                next_run_cache[fps[nid]] = out
+                return out
             kwargs = {}
             for name, value in node.get("inputs", {}).items():
                 if is_link(value):
@@ -112,6 +161,8 @@ class Executor:
             if not isinstance(out, tuple):
                 out = (out,)
             cache[nid] = out
+            if not is_output:
+                next_run_cache[fps[nid]] = out
             return out
 
         # evaluate every sink (output nodes and nodes nobody consumes)
@@ -124,4 +175,5 @@ class Executor:
             is_output = bool(getattr(cls, "OUTPUT_NODE", False))
             if is_output or nid not in consumed:
                 eval_node(nid)
+        self._run_cache = next_run_cache
         return cache

@@ -79,3 +79,48 @@ def test_batch_divider_in_graph():
     outs = cache["2"]
     assert len(outs) == 10
     assert outs[0].shape[0] == 1 and outs[1].shape[0] == 0
+
+
+def test_cross_run_output_caching():
+    """ComfyUI-style execution caching: a re-submitted workflow reuses
+    unchanged nodes' outputs; changing a widget re-executes the node and
+    its downstream; output nodes always re-run."""
+    from comfyui_distributed_amd.graph.executor import Executor, NodeRegistry
+
+    calls = {"a": 0, "sink": 0}
+
+    class Producer:
+        RETURN_TYPES = ("INT",)
+        FUNCTION = "run"
+
+        def run(self, value):
+            calls["a"] += 1
+            return (value * 2,)
+
+    class Sink:
+        RETURN_TYPES = ()
+        OUTPUT_NODE = True
+        FUNCTION = "run"
+
+        def run(self, x):
+            calls["sink"] += 1
+            return ()
+
+    reg = NodeRegistry()
+    reg.register("Producer", Producer)
+    reg.register("Sink", Sink)
+    ex = Executor(registry=reg)
+    prompt = {"1": {"class_type": "Producer", "inputs": {"value": 3}},
+              "2": {"class_type": "Sink", "inputs": {"x": ["1", 0]}}}
+    out1 = ex.execute(prompt)
+    assert out1["1"] == (6,) and calls == {"a": 1, "sink": 1}
+    out2 = ex.execute(prompt)  # producer cached, sink re-runs
+    assert out2["1"] == (6,) and calls == {"a": 1, "sink": 2}
+    prompt["1"]["inputs"]["value"] = 5  # widget change invalidates
+    out3 = ex.execute(prompt)
+    assert out3["1"] == (10,) and calls["a"] == 2
+    # fingerprint is positional spec, not node id: renumbering still hits
+    renamed = {"9": {"class_type": "Producer", "inputs": {"value": 5}},
+               "8": {"class_type": "Sink", "inputs": {"x": ["9", 0]}}}
+    ex.execute(renamed)
+    assert calls["a"] == 2
