@@ -107,6 +107,12 @@ class DistributedServer:
         r.add_post("/distributed/stop_worker", self.post_stop_worker)
         r.add_get("/distributed/managed_workers", self.get_managed_workers)
         r.add_get("/distributed/worker_status", self.get_worker_status)
+        r.add_get("/distributed/local-worker-status",
+                  self.get_local_worker_status)
+        r.add_get("/distributed/remote_worker_log/{worker_id}",
+                  self.get_remote_worker_log)
+        r.add_post("/distributed/worker/clear_launching",
+                   self.post_clear_launching)
         r.add_post("/distributed/auto_populate_workers",
                    self.post_auto_populate_workers)
         r.add_get("/distributed/worker_log", self.get_worker_log)
@@ -545,6 +551,77 @@ class DistributedServer:
             "managed": pid is not None,
             "pid_alive": bool(pid and is_process_alive(pid)),
         })
+
+    async def get_local_worker_status(self, request):
+        """Status of every LOCAL worker in one call (reference
+        worker_routes.py:536-603: the panel polls this instead of N probes).
+        Disabled workers are reported without being probed."""
+        cfg = load_config()
+        local = [w for w in cfg.get("workers", [])
+                 if network.normalize_host(w.get("host", "")) in
+                 ("", "localhost", "127.0.0.1")]
+
+        async def one(w):
+            wid = str(w.get("id"))
+            if not w.get("enabled", False):
+                return wid, {"online": False, "enabled": False,
+                             "processing": False, "queue_count": 0}
+            info = await network.probe_worker(network.build_worker_url(w),
+                                              timeout=2.0)
+            if info is None:
+                return wid, {"online": False, "enabled": True,
+                             "processing": False, "queue_count": 0,
+                             "error": "Unavailable"}
+            qr = info.get("exec_info", {}).get("queue_remaining", 0)
+            return wid, {"online": True, "enabled": True,
+                         "processing": qr > 0, "queue_count": qr}
+
+        results = await asyncio.gather(*(one(w) for w in local))
+        return web.json_response({"status": "success",
+                                  "worker_statuses": dict(results)})
+
+    async def get_remote_worker_log(self, request):
+        """Proxy a remote worker's in-memory log (reference
+        worker_routes.py:649-695); local workers use the file-tail
+        endpoint instead."""
+        wid = str(request.match_info["worker_id"]).strip()
+        worker = get_worker_by_id(load_config(), wid)
+        if worker is None:
+            return _err("unknown worker", status=404)
+        if not network.normalize_host(worker.get("host", "")):
+            return _err(f"worker {wid} is local; use /distributed/worker_log",
+                        status=400)
+        lines = min(max(int(request.query.get("lines", "300")), 1), 3000)
+        import aiohttp
+
+        session = await network.get_client_session()
+        url = network.build_worker_url(worker) + "/distributed/local_log"
+        try:
+            async with session.get(
+                url, params={"lines": str(lines)},
+                timeout=aiohttp.ClientTimeout(total=5),
+            ) as resp:
+                if resp.status >= 400:
+                    return _err(f"worker returned HTTP {resp.status}",
+                                status=resp.status)
+                return web.json_response(await resp.json())
+        except Exception as exc:
+            return _err(f"worker unreachable: {exc}", status=502)
+
+    async def post_clear_launching(self, request):
+        """Clear a worker's 'launching' marker once it is confirmed up
+        (reference worker_routes.py:115-135)."""
+        data = await request.json()
+        wid = data.get("worker_id")
+        if wid is None:
+            return _err("missing worker_id")
+        if get_worker_by_id(load_config(), wid) is None:
+            return _err("unknown worker", status=404)
+        async with config_transaction() as cfg:
+            entry = cfg.get("managed_processes", {}).get(str(wid))
+            if entry:
+                entry.pop("launching", None)
+        return web.json_response({"status": "ok"})
 
     async def post_auto_populate_workers(self, request):
         """One local worker per GPU beyond the master's, first run only

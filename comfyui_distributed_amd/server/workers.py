@@ -115,7 +115,7 @@ def persist_managed_pid(worker_id: str, pid: int) -> None:
     async def txn():
         async with config_transaction() as cfg:
             cfg.setdefault("managed_processes", {})[str(worker_id)] = {
-                "pid": pid, "started": time.time(),
+                "pid": pid, "started": time.time(), "launching": True,
             }
 
     _run(txn())

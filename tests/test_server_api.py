@@ -266,3 +266,54 @@ def test_worker_status_and_auto_populate(client):
         await close_client_session()
 
     run(loop, go())
+
+
+def test_local_worker_status_and_clear_launching(client):
+    srv, cl, loop = client
+
+    async def go():
+        # disabled local + self-pointing enabled local + remote (excluded)
+        for body in (
+            {"id": "la", "name": "a", "port": 1, "host": "", "enabled": False},
+            {"id": "lb", "name": "b", "port": cl.server.port,
+             "host": "127.0.0.1", "enabled": True},
+            {"id": "rc", "name": "c", "port": 9, "host": "10.9.9.9",
+             "type": "remote", "enabled": True},
+        ):
+            r = await cl.post("/distributed/config/update_worker", json=body)
+            assert r.status == 200
+        r = await cl.get("/distributed/local-worker-status")
+        assert r.status == 200
+        st = (await r.json())["worker_statuses"]
+        assert set(st) == {"la", "lb"}  # remote excluded
+        assert st["la"] == {"online": False, "enabled": False,
+                            "processing": False, "queue_count": 0}
+        assert st["lb"]["online"] is True and st["lb"]["queue_count"] == 0
+
+        # remote log proxy rejects local workers; unknown -> 404
+        r = await cl.get("/distributed/remote_worker_log/la")
+        assert r.status == 400
+        r = await cl.get("/distributed/remote_worker_log/ghost")
+        assert r.status == 404
+        r = await cl.get("/distributed/remote_worker_log/rc")
+        assert r.status == 502  # unreachable remote
+
+        # clear_launching drops the marker set by launch_worker persistence
+        from comfyui_distributed_amd.utils.config import (
+            config_transaction, load_config)
+
+        async with config_transaction() as cfg:
+            cfg.setdefault("managed_processes", {})["lb"] = {
+                "pid": 1, "launching": True}
+        r = await cl.post("/distributed/worker/clear_launching",
+                          json={"worker_id": "lb"})
+        assert r.status == 200
+        assert "launching" not in load_config()["managed_processes"]["lb"]
+        r = await cl.post("/distributed/worker/clear_launching",
+                          json={"worker_id": "ghost"})
+        assert r.status == 404
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+
+    run(loop, go())
