@@ -104,3 +104,124 @@ def test_distributed_usdu_static_over_http(tmp_config, monkeypatch):
     assert torch.allclose(canvas, ref, atol=0.02), (
         (canvas - ref).abs().max().item()
     )
+
+
+@pytest.mark.timeout(180)
+def test_zombie_worker_tiles_requeued_over_http(tmp_config, monkeypatch):
+    """Protocol-level fault injection: a 'worker' that exists only on the
+    wire (answers probes with an idle queue) claims tiles via
+    /distributed/request_image, heartbeats once, and never submits. The
+    timeout monitor must probe it, requeue its tiles, and the master must
+    finish the job with the exact single-GPU result."""
+    from aiohttp import web
+
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+    monkeypatch.setattr(constants, "HEARTBEAT_INTERVAL", 0.3)
+
+    async def go():
+        # fake worker: passes preflight + accepts dispatch + probes idle
+        fake = web.Application()
+        fake.router.add_get(
+            "/prompt", lambda r: web.json_response(
+                {"exec_info": {"queue_remaining": 0}}))
+        fake.router.add_post(
+            "/prompt", lambda r: web.json_response({"prompt_id": "fake"}))
+        fake_client = TestClient(TestServer(fake))
+        await fake_client.start_server()
+
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w2", "name": "zombie", "host": "127.0.0.1",
+            "port": fake_client.server.port, "cuda_device": 0,
+            "enabled": True, "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        cfg["settings"]["worker_timeout_seconds"] = 1
+        save_config(cfg)
+
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "tiny"}},
+            "2": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "detail", "clip": ["1", 1]}},
+            "3": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "", "clip": ["1", 1]}},
+            "4": {"class_type": "LoadImage", "inputs": {"image": "synthetic:48x48"}},
+            "5": {"class_type": "UltimateSDUpscaleDistributed", "inputs": {
+                "upscaled_image": ["4", 0], "model": ["1", 0],
+                "positive": ["2", 0], "negative": ["3", 0], "vae": ["1", 2],
+                "seed": 3, "steps": 1, "cfg": 1.0, "sampler_name": "euler",
+                "scheduler": "karras", "denoise": 0.5, "tile_width": 16,
+                "tile_height": 16, "padding": 16, "mask_blur": 2,
+                "force_uniform_tiles": True, "tiled_decode": False}},
+            "6": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["5", 0], "load_balance": False}},
+            "7": {"class_type": "PreviewImage", "inputs": {"images": ["6", 0]}},
+        }
+        resp = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "zt", "enabled_worker_ids": ["w2"]})
+        assert resp.status == 200
+        body = await resp.json()
+        assert "w2" in body["participants"]
+        job_id = body["job_ids"]["5"]
+
+        # zombie protocol: wait for the tile job, claim up to 2 tiles, die
+        claimed = []
+        for _ in range(200):
+            r = await mc.post("/distributed/job_status", json={"job_id": job_id})
+            if r.status == 200 and (await r.json()).get("ready"):
+                break
+            await asyncio.sleep(0.05)
+        for _ in range(2):
+            r = await mc.post("/distributed/request_image", json={
+                "job_id": job_id, "worker_id": "w2"})
+            if r.status == 200:
+                idx = (await r.json()).get("tile_idx")
+                if idx is not None:
+                    claimed.append(idx)
+        await mc.post("/distributed/heartbeat", json={
+            "job_id": job_id, "worker_id": "w2"})
+        # ...and never submit anything again
+
+        for _ in range(600):
+            if previews:
+                break
+            await asyncio.sleep(0.25)
+
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await fake_client.close()
+        await mc.close()
+        return previews, claimed
+
+    previews, claimed = asyncio.run(go())
+    assert previews, "job never completed after zombie worker death"
+    canvas = previews[0]
+    assert canvas.shape == (1, 48, 48, 3)
+
+    # exactness: master-recovered canvas == single-GPU reference
+    from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
+    from comfyui_distributed_amd.graph.builtin_nodes import (
+        _STACK_CACHE, stable_text_seed)
+
+    stack = _STACK_CACHE[("tiny", "cpu")]
+    cond = stack.make_conditioning(stable_text_seed("detail"))
+    uncond = stack.make_conditioning(stable_text_seed(""))
+    p = USDUParams(seed=3, steps=1, cfg=1.0, sampler_name="euler",
+                   scheduler="karras", denoise=0.5, tile_width=16,
+                   tile_height=16, padding=16, mask_blur=2)
+    g = torch.Generator().manual_seed(0)
+    img = torch.rand(1, 48, 48, 3, generator=g)
+    ref = process_single_gpu(stack, cond, uncond, p, img)
+    assert torch.allclose(canvas, ref, atol=1e-5), (
+        (canvas - ref).abs().max().item()
+    )
