@@ -20,7 +20,6 @@ the reference's clone_conditioning shares ControlNet models the same way).
 
 from __future__ import annotations
 
-import torch
 import torch.nn.functional as F
 
 from ..utils.usdu_math import resize_region

@@ -17,7 +17,6 @@ import re
 import shutil
 import subprocess
 import threading
-import time
 from collections import deque
 
 from ..utils.config import config_transaction
