@@ -63,11 +63,11 @@ def launch_worker(worker: dict, monitor: bool = True) -> subprocess.Popen:
         cmd = [sys.executable, "-m",
                "comfyui_distributed_amd.server.worker_monitor",
                "--master-pid", str(os.getpid()), "--", *cmd]
-    logf = open(worker_log_path(wid), "ab")
-    proc = subprocess.Popen(
-        cmd, env=env, stdout=logf, stderr=subprocess.STDOUT,
-        start_new_session=True,
-    )
+    with open(worker_log_path(wid), "ab") as logf:
+        proc = subprocess.Popen(
+            cmd, env=env, stdout=logf, stderr=subprocess.STDOUT,
+            start_new_session=True,
+        )  # the child holds its own fd; the parent's copy closes here
     log(f"launched worker {wid} pid={proc.pid} on GPU "
         f"{worker.get('cuda_device', 0)}")
     persist_managed_pid(wid, proc.pid)
@@ -93,6 +93,11 @@ def stop_worker(handle: subprocess.Popen | None, worker_id: str) -> None:
     except ProcessLookupError:
         pass
     finally:
+        if handle is not None:
+            try:
+                handle.wait(timeout=2.0)  # reap so no zombie lingers
+            except Exception:
+                pass
         forget_managed_pid(worker_id)
 
 

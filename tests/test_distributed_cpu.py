@@ -2,11 +2,9 @@
 collector gather ordering, distributed USDU vs single-GPU equivalence.
 These exercise the same code paths RCCL runs on the GPU node."""
 
-import json
 import os
 import tempfile
 
-import pytest
 import torch
 import torch.multiprocessing as mp
 
@@ -129,7 +127,6 @@ def _body_usdu(ctx, port):
 
     from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
     from comfyui_distributed_amd.models import create_diffusion_stack
-    from comfyui_distributed_amd.parallel.tile_queue import TileQueue
     from comfyui_distributed_amd.parallel.usdu_dist import run_distributed_usdu
 
     store = tdist.TCPStore("127.0.0.1", port + 1000, ctx.world_size,

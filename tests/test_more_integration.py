@@ -4,7 +4,6 @@ seed-parallel video with frame gathering."""
 import asyncio
 
 import pytest
-import torch
 from aiohttp.test_utils import TestClient, TestServer
 
 from comfyui_distributed_amd.server.app import DistributedServer

@@ -1,9 +1,7 @@
 """CPU-path op tests (the same math the GPU kernels implement; the GPU
 numerics tests in test_ops_gpu.py compare the HIP kernels against these)."""
 
-import math
 
-import pytest
 import torch
 import torch.nn.functional as F
 

@@ -4,7 +4,6 @@ All tests are gpu-marked; they fail loudly (KernelUnavailableError) if the
 extension is missing on a GPU box — the HIP path must be the one that runs.
 """
 
-import math
 
 import pytest
 import torch

@@ -2,10 +2,8 @@
 tunnel reader (fake cloudflared)."""
 
 import os
-import signal
 import subprocess
 import sys
-import time
 
 import pytest
 

@@ -7,8 +7,6 @@ seeds, per-process model init) that single-process tests cannot."""
 import asyncio
 import os
 import socket
-import sys
-import time
 
 import pytest
 import torch

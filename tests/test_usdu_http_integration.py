@@ -123,11 +123,15 @@ def test_zombie_worker_tiles_requeued_over_http(tmp_config, monkeypatch):
     async def go():
         # fake worker: passes preflight + accepts dispatch + probes idle
         fake = web.Application()
-        fake.router.add_get(
-            "/prompt", lambda r: web.json_response(
-                {"exec_info": {"queue_remaining": 0}}))
-        fake.router.add_post(
-            "/prompt", lambda r: web.json_response({"prompt_id": "fake"}))
+
+        async def fake_get(_r):
+            return web.json_response({"exec_info": {"queue_remaining": 0}})
+
+        async def fake_post(_r):
+            return web.json_response({"prompt_id": "fake"})
+
+        fake.router.add_get("/prompt", fake_get)
+        fake.router.add_post("/prompt", fake_post)
         fake_client = TestClient(TestServer(fake))
         await fake_client.start_server()
 

@@ -3,7 +3,6 @@ tests/test_distributed_value.py coverage, reimplemented for this node)."""
 
 import json
 
-import pytest
 
 from comfyui_distributed_amd.nodes.utilities import DistributedValue
 

@@ -4,7 +4,6 @@ import json
 from pathlib import Path
 
 import pytest
-import torch
 
 from comfyui_distributed_amd.graph.executor import Executor, default_registry, validate_prompt
 

@@ -3,7 +3,6 @@ dispatch.py:62-95 semantics) against the real server app."""
 
 import asyncio
 
-import pytest
 from aiohttp.test_utils import TestClient, TestServer
 
 from comfyui_distributed_amd.server.app import DistributedServer
