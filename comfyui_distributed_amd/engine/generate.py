@@ -33,8 +33,38 @@ def generate_latents(stack, cond, uncond, p: GenParams) -> torch.Tensor:
         return sample(denoiser, noise, sigmas, sampler=p.sampler_name, seed=p.seed)
 
 
+def generate_latents_flux(stack, cond, uncond, p: GenParams) -> torch.Tensor:
+    """Rectified-flow sampling for the Flux family (velocity prediction,
+    Euler integration over the shifted time schedule)."""
+    from ..models.video import sample_flow
+
+    b = p.batch_size
+    shape = (b, stack.cfg.in_channels, p.height // 8, p.width // 8)
+    g = torch.Generator(device="cpu").manual_seed(p.seed)
+    noise = torch.randn(shape, generator=g).to(stack.device)
+    ctx = cond["context"].expand(b, -1, -1).to(stack.dtype)
+    vec = cond["vec"].expand(b, -1).to(stack.dtype)
+
+    def velocity(x, t):
+        tt = (t * 1000.0).reshape(-1).to(x.device).expand(b)
+        v = stack.model(x.to(stack.dtype), tt.to(stack.dtype), ctx, vec)
+        if p.cfg != 1.0 and uncond is not None:
+            vu = stack.model(
+                x.to(stack.dtype), tt.to(stack.dtype),
+                uncond["context"].expand(b, -1, -1).to(stack.dtype),
+                uncond["vec"].expand(b, -1).to(stack.dtype))
+            v = vu + p.cfg * (v - vu)
+        return v
+
+    return sample_flow(velocity, noise, p.steps, shift=stack.flow_shift)
+
+
 def generate_images(stack, cond, uncond, p: GenParams) -> torch.Tensor:
     """Returns [B, H, W, 3] float32 in [0,1] on the stack device."""
+    if getattr(stack, "family", "sd") == "flux":
+        with torch.no_grad():
+            latents = generate_latents_flux(stack, cond, uncond, p)
+            return stack.vae.decode(latents.to(stack.dtype)).float()
     latents = generate_latents(stack, cond, uncond, p)
     with torch.no_grad():
         return stack.vae.decode(latents.to(stack.dtype)).float()

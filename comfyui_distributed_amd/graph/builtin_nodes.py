@@ -220,6 +220,41 @@ class WanVideoGenerate(_ContextNode):
         return (generate_video(model, positive, negative, p).cpu(),)
 
 
+class FluxGenerate(_ContextNode):
+    """txt2img with a Flux-family MMDiT stack (rectified flow); the SD
+    KSampler node does not apply to flow models, so flux workflows use
+    this node the way video ones use WanVideoGenerate."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {
+            "required": {
+                "model": ("MODEL",),
+                "positive": ("CONDITIONING",),
+                "seed": ("INT", {"default": 0}),
+                "steps": ("INT", {"default": 20}),
+                "cfg": ("FLOAT", {"default": 1.0}),
+                "width": ("INT", {"default": 1024}),
+                "height": ("INT", {"default": 1024}),
+                "batch_size": ("INT", {"default": 1}),
+            },
+            "optional": {"negative": ("CONDITIONING",)},
+        }
+
+    RETURN_TYPES = ("IMAGE",)
+    FUNCTION = "generate"
+    CATEGORY = "sampling"
+
+    def generate(self, model, positive, seed, steps, cfg, width, height,
+                 batch_size=1, negative=None):
+        from ..engine.generate import GenParams, generate_images
+
+        p = GenParams(seed=int(seed), steps=int(steps), cfg=float(cfg),
+                      width=int(width), height=int(height),
+                      batch_size=int(batch_size))
+        return (generate_images(model, positive, negative, p).cpu(),)
+
+
 class LoadImage(_ContextNode):
     """Loads an image from the input directory (or a synthetic one when the
     name is "synthetic:<W>x<H>")."""
@@ -298,4 +333,5 @@ BUILTIN_CLASS_MAPPINGS = {
     "SaveImage": SaveImage,
     "PreviewImage": PreviewImage,
     "WanVideoGenerate": WanVideoGenerate,
+    "FluxGenerate": FluxGenerate,
 }

@@ -94,6 +94,13 @@ def create_diffusion_stack(name: str, device="cpu", dtype=None, seed: int = 0):
         if dtype is None:
             dtype = torch.bfloat16 if str(device).startswith("cuda") else torch.float32
         return WanStack(WAN_CONFIGS[name], device=device, dtype=dtype, seed=seed)
+    if name in ("flux12b", "flux_tiny"):
+        from .flux import FLUX12B, FLUX_TINY, FluxStack
+
+        if dtype is None:
+            dtype = torch.bfloat16 if str(device).startswith("cuda") else torch.float32
+        cfg = FLUX12B if name == "flux12b" else FLUX_TINY
+        return FluxStack(cfg, device=device, dtype=dtype, seed=seed)
     cfg = MODEL_CONFIGS[name]
     if dtype is None:
         dtype = torch.bfloat16 if (isinstance(device, str) and device.startswith("cuda")) or (

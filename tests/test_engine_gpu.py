@@ -82,3 +82,16 @@ def test_wan_tiny_temporal_vae_gpu():
     out = generate_video(stack, cond, None, p)
     assert out.shape == (9, 32, 32, 3)
     assert torch.isfinite(out).all()
+
+
+def test_flux_tiny_gpu_generation():
+    from comfyui_distributed_amd.engine.generate import GenParams, generate_images
+    from comfyui_distributed_amd.models import create_diffusion_stack
+
+    stack = create_diffusion_stack("flux_tiny", device="cuda:0",
+                                   dtype=torch.bfloat16)
+    cond = stack.make_conditioning(0)
+    p = GenParams(seed=4, steps=2, cfg=1.0, width=64, height=64, batch_size=2)
+    imgs = generate_images(stack, cond, None, p)
+    assert imgs.shape == (2, 64, 64, 3)
+    assert torch.isfinite(imgs).all()

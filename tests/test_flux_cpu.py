@@ -1,0 +1,71 @@
+"""Flux-family MMDiT: param counts, shapes, determinism, seed-parallel."""
+
+import pytest
+import torch
+
+from comfyui_distributed_amd.models.registry import create_diffusion_stack
+
+
+def test_flux_tiny_forward_shapes():
+    stack = create_diffusion_stack("flux_tiny")
+    cond = stack.make_conditioning(0)
+    x = torch.randn(2, 16, 8, 8)
+    t = torch.full((2,), 500.0)
+    out = stack.model(x, t, cond["context"].expand(2, -1, -1),
+                      cond["vec"].expand(2, -1))
+    assert out.shape == x.shape
+    assert torch.isfinite(out).all()
+
+
+def test_flux12b_param_count():
+    """Flux-dev scale: ~11-12B parameters at dim 3072, 19+38 blocks."""
+    from comfyui_distributed_amd.models.flux import FLUX12B, FluxModel
+
+    # count without materializing 12B floats: sum shapes via meta device
+    with torch.device("meta"):
+        m = FluxModel(FLUX12B)
+    n = sum(p.numel() for p in m.parameters())
+    assert 10.5e9 < n < 13.5e9, n / 1e9
+
+
+def test_flux_generation_deterministic():
+    from comfyui_distributed_amd.engine.generate import GenParams, generate_images
+
+    stack = create_diffusion_stack("flux_tiny")
+    cond = stack.make_conditioning(1)
+    p = GenParams(seed=7, steps=2, cfg=1.0, width=32, height=32, batch_size=1)
+    a = generate_images(stack, cond, None, p)
+    b = generate_images(stack, cond, None, p)
+    assert a.shape == (1, 32, 32, 3)
+    assert torch.equal(a, b)
+    c = generate_images(stack, cond, None, GenParams(
+        seed=8, steps=2, cfg=1.0, width=32, height=32))
+    assert not torch.allclose(a, c)
+
+
+def test_flux_cfg_path():
+    from comfyui_distributed_amd.engine.generate import GenParams, generate_images
+
+    stack = create_diffusion_stack("flux_tiny")
+    cond, uncond = stack.make_conditioning(1), stack.make_conditioning(2)
+    p = GenParams(seed=7, steps=1, cfg=3.0, width=16, height=16)
+    out = generate_images(stack, cond, uncond, p)
+    assert torch.isfinite(out).all()
+
+
+def test_flux_workflow_graph():
+    from comfyui_distributed_amd.graph.executor import Executor
+
+    prompt = {
+        "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "flux_tiny"}},
+        "2": {"class_type": "CLIPTextEncode",
+              "inputs": {"text": "a fox", "clip": ["1", 1]}},
+        "3": {"class_type": "FluxGenerate", "inputs": {
+            "model": ["1", 0], "positive": ["2", 0], "seed": 4, "steps": 1,
+            "cfg": 1.0, "width": 16, "height": 16, "batch_size": 2}},
+        "4": {"class_type": "PreviewImage", "inputs": {"images": ["3", 0]}},
+    }
+    previews = []
+    ex = Executor(context={"preview_images": previews, "device": "cpu"})
+    ex.execute(prompt)
+    assert previews and previews[0].shape == (2, 16, 16, 3)

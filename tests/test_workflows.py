@@ -21,6 +21,7 @@ def load_wf(name):
     "distributed_upscale.json",
     "distributed_wan_video.json",
     "distributed_upscale_video.json",
+    "distributed_flux_txt2img.json",
 ])
 def test_workflows_validate(name):
     validate_prompt(load_wf(name), default_registry())
