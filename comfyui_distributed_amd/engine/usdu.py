@@ -58,6 +58,27 @@ def _tile_noise(seed: int, tile_idx: int, batch_idx: int, shape) -> torch.Tensor
     return torch.randn(shape, generator=g)
 
 
+def _flow_velocity(stack, cond, uncond, cfg_scale: float):
+    """CFG velocity closure for flow-family stacks (Flux) inside USDU —
+    the flow-matching counterpart of CFGDenoiser."""
+
+    def velocity(x, t):
+        b = x.shape[0]
+        tt = (t * 1000.0).reshape(-1).to(x.device).expand(b).to(stack.dtype)
+        ctx = cond["context"].expand(b, -1, -1).to(stack.dtype)
+        vec = cond["vec"].expand(b, -1).to(stack.dtype)
+        v = stack.model(x.to(stack.dtype), tt, ctx, vec)
+        if cfg_scale != 1.0 and uncond is not None:
+            vu = stack.model(
+                x.to(stack.dtype), tt,
+                uncond["context"].expand(b, -1, -1).to(stack.dtype),
+                uncond["vec"].expand(b, -1).to(stack.dtype))
+            v = vu + cfg_scale * (v - vu)
+        return v.float()
+
+    return velocity
+
+
 def sample_tiles(
     stack,
     cond: dict,
@@ -81,11 +102,16 @@ def sample_tiles(
     if not tile_indices:
         return {}
     B = canvas.shape[0]
-    schedule: NoiseSchedule = stack.schedule
-    sigmas = schedule.sigmas(params.steps, params.scheduler, params.denoise).to(
-        canvas.device
-    )
-    base_denoiser = CFGDenoiser(stack.unet, schedule, cond, uncond, params.cfg)
+    is_flow = getattr(stack, "family", "sd") == "flux"
+    if is_flow:
+        sigmas = base_denoiser = None  # flow models integrate velocity
+    else:
+        schedule: NoiseSchedule = stack.schedule
+        sigmas = schedule.sigmas(
+            params.steps, params.scheduler, params.denoise
+        ).to(canvas.device)
+        base_denoiser = CFGDenoiser(stack.unet, schedule, cond, uncond,
+                                    params.cfg)
 
     from .conditioning import SPATIAL_KEYS, crop_tile_conditioning
     from .model_patch import crop_model_patches, stack_has_patches
@@ -112,12 +138,14 @@ def sample_tiles(
             chunk_uncond = crop_tile_conditioning(
                 uncond, plan.crop_region, plan.canvas_size, plan.process_size
             )
-        denoiser = (
-            base_denoiser
-            if not has_spatial
-            else CFGDenoiser(stack.unet, schedule, chunk_cond, chunk_uncond,
-                             params.cfg)
-        )
+        denoiser = None
+        if not is_flow:
+            denoiser = (
+                base_denoiser
+                if not has_spatial
+                else CFGDenoiser(stack.unet, schedule, chunk_cond,
+                                 chunk_uncond, params.cfg)
+            )
         # ---- extract + resample each (tile, batch) crop to process size ----
         crops = []
         for t, b in chunk:
@@ -140,14 +168,27 @@ def sample_tiles(
             with trace_range("usdu.sample"), crop_model_patches(
                 stack, plan0.crop_region, plan0.canvas_size, plan0.process_size
             ):
-                latent_out = sample(
-                    denoiser,
-                    noise,
-                    sigmas,
-                    sampler=params.sampler_name,
-                    seed=params.seed,
-                    start_from_latent=latents.float(),
-                )
+                if is_flow:
+                    from ..models.video import sample_flow
+
+                    latent_out = sample_flow(
+                        _flow_velocity(stack, chunk_cond, chunk_uncond,
+                                       params.cfg),
+                        noise,
+                        params.steps,
+                        shift=getattr(stack, "flow_shift", 3.0),
+                        start_from_latent=latents.float(),
+                        denoise=params.denoise,
+                    )
+                else:
+                    latent_out = sample(
+                        denoiser,
+                        noise,
+                        sigmas,
+                        sampler=params.sampler_name,
+                        seed=params.seed,
+                        start_from_latent=latents.float(),
+                    )
             with trace_range("usdu.vae_decode"):
                 if params.tiled_decode:
                     out_img = stack.vae.decode_tiled(latent_out.to(stack.dtype))

@@ -171,10 +171,23 @@ class FlowCFGVelocity:
 
 
 def sample_flow(velocity_fn, noise: torch.Tensor, steps: int,
-                shift: float = 5.0) -> torch.Tensor:
-    """Euler integration of the rectified flow from noise (t=1) to data."""
-    sig = flow_sigmas(steps, shift)
-    x = noise.float()
+                shift: float = 5.0, start_from_latent: torch.Tensor | None = None,
+                denoise: float = 1.0) -> torch.Tensor:
+    """Euler integration of the rectified flow from noise (t=1) to data.
+
+    img2img: with ``start_from_latent`` and ``denoise`` < 1 the schedule is
+    truncated to [denoise, 0] and integration starts from the flow
+    interpolant x_t0 = (1-t0)*x0 + t0*noise at the (shifted) strength t0 —
+    the flow-matching analogue of the SD img2img start used by USDU."""
+    t_lin = torch.linspace(1.0, 0.0, steps + 1)
+    if start_from_latent is not None and denoise < 1.0:
+        t_lin = t_lin * denoise
+    sig = shift * t_lin / (1 + (shift - 1) * t_lin)
+    if start_from_latent is not None:
+        t0 = sig[0].to(noise.device)
+        x = (1 - t0) * start_from_latent.float() + t0 * noise.float()
+    else:
+        x = noise.float()
     for i in range(steps):
         t, t_next = sig[i], sig[i + 1]
         v = velocity_fn(x, t)
