@@ -37,6 +37,8 @@ CONFIG_PRESETS = {
     "usdu-sdxl-8k": {"model": "sdxl", "src_size": 2048, "scale": 4,
                      "tile": 1024},
     "wan-t2v": {"model": "wan14b"},
+    # beyond-BASELINE: Flux MMDiT seed-parallel generation
+    "gen-flux": {"model": "flux12b", "cfg": 1.0},
 }
 
 
@@ -106,7 +108,9 @@ def bench_generation(args, ctx, stack, cond, uncond):
             seed_parallel_generate(ctx, stack, cond, uncond, p)
 
         unit, per_step = "images/s", world
-        model_name = "sdxl-unet-2.6B-random-init"
+        model_name = ("flux-mmdit-11.9B-random-init"
+                      if args.config == "gen-flux"
+                      else "sdxl-unet-2.6B-random-init")
         cfg_extra = {"size": "1024x1024",
                      "sampler": f"euler/{args.sampler_steps}steps/cfg{args.cfg}"}
 
@@ -153,7 +157,7 @@ def main():
     cond = stack.make_conditioning(0)
     uncond = stack.make_conditioning(1)
 
-    if args.config in ("gen-sdxl", "wan-t2v"):
+    if args.config in ("gen-sdxl", "wan-t2v", "gen-flux"):
         bench_generation(args, ctx, stack, cond, uncond)
         return
 

@@ -95,3 +95,18 @@ def test_flux_tiny_gpu_generation():
     imgs = generate_images(stack, cond, None, p)
     assert imgs.shape == (2, 64, 64, 3)
     assert torch.isfinite(imgs).all()
+
+
+def test_flux_usdu_tile_gpu():
+    from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
+    from comfyui_distributed_amd.models import create_diffusion_stack
+
+    stack = create_diffusion_stack("flux_tiny", device="cuda:0",
+                                   dtype=torch.bfloat16)
+    cond = stack.make_conditioning(1)
+    p = USDUParams(seed=5, steps=1, cfg=1.0, denoise=0.5, tile_width=64,
+                   tile_height=64, padding=32, mask_blur=4, tile_batch=4)
+    img = torch.rand(1, 128, 128, 3, generator=torch.Generator().manual_seed(2))
+    out = process_single_gpu(stack, cond, None, p, img)
+    assert out.shape == (1, 128, 128, 3)
+    assert torch.isfinite(out).all()
