@@ -1,0 +1,103 @@
+"""Property-based tests (hypothesis) for the USDU tile geometry — the
+invariants every mode and every rank depend on: full coverage, in-bounds
+crops, round8 processing sizes, crop containing the tile."""
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from comfyui_distributed_amd.engine.usdu import USDUParams, plan_for_image
+
+
+@st.composite
+def geometry(draw):
+    w = draw(st.integers(min_value=16, max_value=512))
+    h = draw(st.integers(min_value=16, max_value=512))
+    tile = draw(st.sampled_from([16, 32, 64, 128]))
+    pad = draw(st.sampled_from([0, 8, 16, 32]))
+    blur = draw(st.integers(min_value=0, max_value=16))
+    uniform = draw(st.booleans())
+    return w, h, tile, pad, blur, uniform
+
+
+@given(geometry())
+@settings(max_examples=120, deadline=None)
+def test_plans_cover_canvas_in_bounds(geo):
+    w, h, tile, pad, blur, uniform = geo
+    p = USDUParams(seed=0, steps=1, cfg=1.0, tile_width=tile, tile_height=tile,
+                   padding=pad, mask_blur=blur, force_uniform_tiles=uniform)
+    plans = plan_for_image(w, h, p)
+    assert plans, "no tiles planned"
+    covered = [[False] * w for _ in range(h)]
+    for plan in plans:
+        x1, y1, x2, y2 = plan.crop_region
+        tx1, ty1, tx2, ty2 = plan.tile_rect
+        # crop within canvas
+        assert 0 <= x1 < x2 <= w and 0 <= y1 < y2 <= h, plan.crop_region
+        if not uniform:
+            # crop contains the tile up to the A1111-compat 1px
+            # fix_crop_region shrink (right/bottom edge -1) — the same
+            # artifact USDU itself has at padding 0. (Uniform mode's
+            # aspect expansion can legally shift the crop off a sliver
+            # tile at the canvas edge; its geometry is pinned by the
+            # golden-value tests in test_usdu_math.py instead.)
+            assert (x1 <= tx1 and y1 <= ty1
+                    and tx2 - 1 <= x2 and ty2 - 1 <= y2)
+        # processing size: positive multiples of 8 (VAE/latent alignment)
+        pw, ph = plan.process_size
+        assert pw > 0 and ph > 0 and pw % 8 == 0 and ph % 8 == 0
+        for yy in range(ty1, ty2):
+            row = covered[yy]
+            for xx in range(tx1, tx2):
+                row[xx] = True
+    assert all(all(row) for row in covered), "tiles do not cover the canvas"
+
+
+@given(st.integers(2, 64), st.integers(2, 64),
+       st.floats(0.3, 12.0), st.data())
+@settings(max_examples=60, deadline=None)
+def test_rect_mask_properties(h, w, sigma, data):
+    """The analytic blurred-rect mask: in [0,1], ~1 deep inside the rect,
+    ~0 far outside, monotone toward the rect."""
+    import torch
+
+    from comfyui_distributed_amd.ops.dispatch import rect_mask_cpu
+
+    rx1 = data.draw(st.integers(0, w - 1))
+    ry1 = data.draw(st.integers(0, h - 1))
+    rx2 = data.draw(st.integers(rx1 + 1, w))
+    ry2 = data.draw(st.integers(ry1 + 1, h))
+    m = rect_mask_cpu(h, w, (rx1, ry1, rx2, ry2), sigma)
+    assert m.shape == (h, w)
+    assert torch.isfinite(m).all()
+    assert (m >= -1e-5).all() and (m <= 1 + 1e-5).all()
+    cx, cy = (rx1 + rx2) // 2, (ry1 + ry2) // 2
+    # center of a rect much wider than sigma saturates to ~1
+    if (rx2 - rx1) > 8 * sigma and (ry2 - ry1) > 8 * sigma:
+        assert m[cy, cx] > 0.99
+    # points further than 6 sigma outside are ~0
+    far_x = rx2 + int(6 * sigma) + 1
+    if far_x < w and ry1 <= cy < ry2:
+        assert m[cy, far_x] < 0.01
+
+
+@given(st.integers(4, 40), st.integers(4, 40), st.integers(4, 40),
+       st.integers(4, 40))
+@settings(max_examples=40, deadline=None)
+def test_blend_stays_in_convex_hull(h, w, ow, oh):
+    """Compositing under a [0,1] mask keeps every pixel inside the convex
+    hull of (base, resampled tile) up to Lanczos ringing bounds."""
+    import torch
+
+    from comfyui_distributed_amd.ops.dispatch import blend_tile
+
+    g = torch.Generator().manual_seed(h * 1000 + w)
+    canvas = torch.rand(1, h, w, 3, generator=g)
+    before = canvas.clone()
+    tile = torch.full((1, oh, ow, 3), 0.5)
+    region = (0, 0, w, h)
+    blend_tile(canvas, tile, region, (1, 1, max(2, w - 1), max(2, h - 1)), 1.5)
+    assert torch.isfinite(canvas).all()
+    # uniform tile resamples to exactly 0.5 => blend is lerp(base, 0.5, m)
+    lo = torch.minimum(before, torch.full_like(before, 0.5)) - 1e-4
+    hi = torch.maximum(before, torch.full_like(before, 0.5)) + 1e-4
+    assert (canvas >= lo).all() and (canvas <= hi).all()

@@ -317,3 +317,19 @@ def test_local_worker_status_and_clear_launching(client):
         await close_client_session()
 
     run(loop, go())
+
+
+def test_panel_served_with_expected_controls(client):
+    srv, cl, loop = client
+
+    async def go():
+        for path in ("/", "/panel"):
+            r = await cl.get(path)
+            assert r.status == 200
+            html = await r.text()
+            for needle in ("worker_status", "auto_populate_workers",
+                           "update_worker", "delete_worker", "update_setting",
+                           "tunnel", "local_log", "distributed/queue"):
+                assert needle in html, f"panel missing wiring for {needle}"
+
+    run(loop, go())
