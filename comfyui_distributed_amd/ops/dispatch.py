@@ -301,7 +301,10 @@ def _lanczos_weight_matrix(
     tap enumeration (inclusive [floor(c-s+.5), floor(c+s+.5)], max 16 taps,
     edge clamp, renormalized)."""
     centers = src_lo + (torch.arange(n_out, dtype=torch.float64) + 0.5) * scale - 0.5
-    fscale = max(scale, 1.0)
+    # clamp like the HIP kernel: the full window must fit in 16 taps, or
+    # truncation can leave a near-zero weight sum and the renormalization
+    # explodes (downscales > 2.5x)
+    fscale = min(max(scale, 1.0), (16 - 1) / (2.0 * LANCZOS_A))
     support = LANCZOS_A * fscale
     x0 = torch.floor(centers - support + 0.5)
     x1 = torch.floor(centers + support + 0.5)

@@ -41,8 +41,12 @@ __device__ void lanczos_sample(const float* __restrict__ src, int src_h,
                                int src_w, int channels, long long src_batch_off,
                                float sx1, float sy1, float scale_x, float scale_y,
                                int ox, int oy, float* out_px) {
-  const float fscale_x = fmaxf(scale_x, 1.0f);
-  const float fscale_y = fmaxf(scale_y, 1.0f);
+  // Clamp the widened filter so the full window fits in MAX_TAPS: a
+  // truncated Lanczos window can have a near-zero weight sum, and the
+  // renormalization would explode (observed at downscales > 2.5x).
+  const float max_fscale = (MAX_TAPS - 1) / (2.0f * LANCZOS_A);
+  const float fscale_x = fminf(fmaxf(scale_x, 1.0f), max_fscale);
+  const float fscale_y = fminf(fmaxf(scale_y, 1.0f), max_fscale);
   const float cx = sx1 + (ox + 0.5f) * scale_x - 0.5f;
   const float cy = sy1 + (oy + 0.5f) * scale_y - 0.5f;
   const float support_x = LANCZOS_A * fscale_x;
