@@ -229,3 +229,83 @@ def test_zombie_worker_tiles_requeued_over_http(tmp_config, monkeypatch):
     assert torch.allclose(canvas, ref, atol=1e-5), (
         (canvas - ref).abs().max().item()
     )
+
+
+@pytest.mark.timeout(180)
+def test_distributed_usdu_dynamic_over_http(tmp_config, monkeypatch):
+    """Image-parallel (dynamic) mode end-to-end over real HTTP: batch >=
+    dynamic_threshold routes whole images through the pull queue; the
+    worker returns full frames via /distributed/submit_image."""
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+    monkeypatch.setattr(constants, "JOB_READY_POLL_INTERVAL", 0.2)
+
+    async def go():
+        worker_srv = DistributedServer(is_worker=True)
+        wc = TestClient(TestServer(worker_srv.build_app()))
+        await wc.start_server()
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "worker1", "host": "127.0.0.1",
+            "port": wc.server.port, "cuda_device": 0,
+            "enabled": True, "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "tiny"}},
+            "2": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "detail", "clip": ["1", 1]}},
+            "3": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "", "clip": ["1", 1]}},
+            "4": {"class_type": "EmptyLatentImage",
+                  "inputs": {"width": 32, "height": 32, "batch_size": 2}},
+            "5": {"class_type": "KSampler", "inputs": {
+                "model": ["1", 0], "seed": 2, "steps": 1, "cfg": 1.0,
+                "sampler_name": "euler", "scheduler": "karras",
+                "positive": ["2", 0], "negative": ["3", 0],
+                "latent_image": ["4", 0], "denoise": 1.0}},
+            "6": {"class_type": "VAEDecode",
+                  "inputs": {"samples": ["5", 0], "vae": ["1", 2]}},
+            "7": {"class_type": "UltimateSDUpscaleDistributed", "inputs": {
+                "upscaled_image": ["6", 0], "model": ["1", 0],
+                "positive": ["2", 0], "negative": ["3", 0], "vae": ["1", 2],
+                "seed": 3, "steps": 1, "cfg": 1.0, "sampler_name": "euler",
+                "scheduler": "karras", "denoise": 0.5, "tile_width": 16,
+                "tile_height": 16, "padding": 16, "mask_blur": 2,
+                "force_uniform_tiles": True, "tiled_decode": False,
+                "dynamic_threshold": 2}},
+            "8": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["7", 0], "load_balance": False}},
+            "9": {"class_type": "PreviewImage", "inputs": {"images": ["8", 0]}},
+        }
+        resp = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "dyn", "enabled_worker_ids": ["w1"]})
+        assert resp.status == 200
+
+        for _ in range(600):
+            if previews:
+                break
+            await asyncio.sleep(0.25)
+
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await wc.close()
+        await mc.close()
+        return previews
+
+    previews = asyncio.run(go())
+    assert previews, "dynamic USDU never completed"
+    canvas = previews[0]
+    assert canvas.shape == (2, 32, 32, 3)
+    assert torch.isfinite(canvas).all()
