@@ -255,6 +255,35 @@ class FluxGenerate(_ContextNode):
         return (generate_images(model, positive, negative, p).cpu(),)
 
 
+class SyntheticAudio(_ContextNode):
+    """Deterministic seeded audio source (sine + noise) so audio pipelines
+    (collector audio gather, AudioBatchDivider) are wireable end-to-end —
+    the reference delegates audio generation to host-app nodes; this is
+    the random-init stand-in (same stance as the image models)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {
+            "seed": ("INT", {"default": 0}),
+            "seconds": ("FLOAT", {"default": 1.0}),
+            "sample_rate": ("INT", {"default": 44100}),
+            "freq": ("FLOAT", {"default": 440.0}),
+        }}
+
+    RETURN_TYPES = ("AUDIO",)
+    FUNCTION = "generate"
+    CATEGORY = "audio"
+
+    def generate(self, seed=0, seconds=1.0, sample_rate=44100, freq=440.0):
+        n = max(1, int(float(seconds) * int(sample_rate)))
+        t = torch.arange(n, dtype=torch.float32) / float(sample_rate)
+        g = torch.Generator().manual_seed(int(seed))
+        wave = (0.8 * torch.sin(2 * torch.pi * float(freq) * t)
+                + 0.05 * torch.randn(n, generator=g))
+        return ({"waveform": wave.expand(2, -1).unsqueeze(0).contiguous(),
+                 "sample_rate": int(sample_rate)},)
+
+
 class LoadImage(_ContextNode):
     """Loads an image from the input directory (or a synthetic one when the
     name is "synthetic:<W>x<H>")."""
@@ -334,4 +363,5 @@ BUILTIN_CLASS_MAPPINGS = {
     "PreviewImage": PreviewImage,
     "WanVideoGenerate": WanVideoGenerate,
     "FluxGenerate": FluxGenerate,
+    "SyntheticAudio": SyntheticAudio,
 }
