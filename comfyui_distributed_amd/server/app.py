@@ -98,6 +98,7 @@ class DistributedServer:
         r.add_post("/distributed/config/update_master", self.post_update_master)
         r.add_post("/distributed/check_file", self.post_check_file)
         r.add_post("/upload/image", self.post_upload_image)
+        r.add_get("/view", self.get_view)
         r.add_get("/distributed/network_info", self.get_network_info)
         r.add_get("/distributed/system_info", self.get_system_info)
         r.add_post("/distributed/tunnel/start", self.post_tunnel_start)
@@ -252,7 +253,10 @@ class DistributedServer:
         gc.collect()
         if torch.cuda.is_available():
             torch.cuda.empty_cache()
-        return web.json_response({"status": "ok", "unloaded_models": n})
+        relayed = await self._fanout_to_workers(request,
+                                                "/distributed/clear_memory")
+        return web.json_response({"status": "ok", "unloaded_models": n,
+                                  **relayed})
 
     # ---- USDU tile endpoints ---------------------------------------------
 
@@ -476,6 +480,24 @@ class DistributedServer:
                         fh.write(chunk)
                 saved.append(fname)
         return web.json_response({"saved": saved})
+
+    async def get_view(self, request):
+        """Serve a saved output (ComfyUI GET /view parity: the panel and
+        external clients fetch results by filename). ``type=input`` reads
+        the input dir instead."""
+        from pathlib import Path
+
+        name = os.path.basename(str(request.query.get("filename", "")))
+        if not name:
+            return _err("missing filename")
+        kind = request.query.get("type", "output")
+        key = "input_dir" if kind == "input" else "output_dir"
+        path = Path(self.executor.context.get(key, kind)) / name
+        if not path.is_file():
+            return _err("not found", status=404)
+        ctype = "image/png" if name.lower().endswith(".png") else \
+            "application/octet-stream"
+        return web.Response(body=path.read_bytes(), content_type=ctype)
 
     async def get_network_info(self, request):
         import torch
@@ -726,11 +748,43 @@ class DistributedServer:
 
     async def post_interrupt(self, request):
         """User interrupt: flags the node runtime so every wait loop raises
-        (reference checks comfy.model_management interrupts in each loop)."""
+        (reference checks comfy.model_management interrupts in each loop).
+        With {"fanout": true} the master also relays the interrupt to every
+        enabled worker (the reference's interrupt-all lives client-side in
+        web/workerUtils.js:4-125; here the master fans out server-side)."""
         from ..nodes.runtime import get_runtime
 
         get_runtime().interrupt()
-        return web.json_response({"status": "interrupted"})
+        relayed = await self._fanout_to_workers(request, "/interrupt")
+        return web.json_response({"status": "interrupted", **relayed})
+
+    async def _fanout_to_workers(self, request, path: str) -> dict:
+        """POST ``path`` to every enabled worker when the request body asks
+        for fanout; returns {} otherwise."""
+        try:
+            data = await request.json() if request.can_read_body else {}
+        except Exception:
+            data = {}
+        if not data.get("fanout"):
+            return {}
+        import aiohttp
+
+        session = await network.get_client_session()
+
+        async def one(w):
+            try:
+                url = network.build_worker_url(w) + path
+                async with session.post(
+                    url, json={}, timeout=aiohttp.ClientTimeout(total=5)
+                ) as resp:
+                    return str(w.get("id")), resp.status == 200
+            except Exception:
+                return str(w.get("id")), False
+
+        workers = [w for w in load_config().get("workers", [])
+                   if w.get("enabled")]
+        results = await asyncio.gather(*(one(w) for w in workers))
+        return {"fanout": dict(results)}
 
     async def ws_handler(self, request):
         """WebSocket orchestration endpoint: accepts dispatch_prompt

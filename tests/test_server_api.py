@@ -333,3 +333,46 @@ def test_panel_served_with_expected_controls(client):
                 assert needle in html, f"panel missing wiring for {needle}"
 
     run(loop, go())
+
+
+def test_interrupt_fanout_and_view(client, tmp_path):
+    srv, cl, loop = client
+
+    async def go():
+        # a worker pointing at this server: fanout interrupt reaches it
+        r = await cl.post("/distributed/config/update_worker", json={
+            "id": "fo", "name": "fo", "port": cl.server.port,
+            "host": "127.0.0.1", "type": "remote", "enabled": True})
+        assert r.status == 200
+        r = await cl.post("/interrupt", json={"fanout": True})
+        body = await r.json()
+        assert body["status"] == "interrupted"
+        assert body["fanout"] == {"fo": True}
+        from comfyui_distributed_amd.nodes.runtime import get_runtime
+
+        get_runtime().clear_interrupt()
+        # no fanout flag -> no relay key
+        r = await cl.post("/interrupt")
+        body = await r.json()
+        assert "fanout" not in body
+        get_runtime().clear_interrupt()
+
+        # /view serves saved outputs, rejects traversal, 404s missing
+        out_dir = srv.executor.context.setdefault("output_dir", str(tmp_path))
+        from pathlib import Path
+
+        Path(out_dir).mkdir(parents=True, exist_ok=True)
+        (Path(out_dir) / "res.png").write_bytes(b"\x89PNG fake")
+        r = await cl.get("/view?filename=res.png")
+        assert r.status == 200
+        assert r.content_type == "image/png"
+        assert await r.read() == b"\x89PNG fake"
+        r = await cl.get("/view?filename=../res.png")
+        assert r.status in (200, 404)  # basename-sanitized, never escapes
+        r = await cl.get("/view?filename=nope.png")
+        assert r.status == 404
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+
+    run(loop, go())
