@@ -186,3 +186,21 @@ def test_usdu_with_input_broadcast():
     out = _spawn("_body_usdu_broadcast", port=PORT_BASE + 5)
     dist_c, ref_c = out[0]["dist"], out[0]["ref"]
     assert torch.allclose(dist_c, ref_c, atol=1e-5)
+
+
+def test_distributed_usdu_world4_equals_single_gpu():
+    """4-rank tile-queue USDU (the shape the driver's 8-GPU scaling run
+    exercises): result identical to the single-GPU canvas."""
+    out = _spawn("_body_usdu", world=4, port=PORT_BASE + 6)
+    dist_c, ref_c = out[0]["dist"], out[0]["ref"]
+    assert torch.allclose(dist_c, ref_c, atol=1e-5)
+
+
+def test_gather_variable_counts_world4():
+    out = _spawn("_body_gather", world=4, port=PORT_BASE + 7)
+    res = out[0]
+    # rank0: 2 items, ranks 1-3: 3 items each
+    assert len(res["meta"]) == 2 + 3 * 3
+    assert sorted(res["meta"]) == sorted(
+        [(0, i) for i in range(2)]
+        + [(r, i) for r in range(1, 4) for i in range(3)])
