@@ -31,6 +31,13 @@ from .runtime import get_runtime
 EMPTY_AUDIO = {"waveform": torch.zeros(1, 2, 1), "sample_rate": 44100}
 
 
+def _is_real_audio(audio) -> bool:
+    """False for None and for the 1-sample EMPTY_AUDIO placeholder (its
+    numel is 2, so a numel check would ship empty envelopes on image-only
+    workflows)."""
+    return audio is not None and audio["waveform"].shape[-1] > 1
+
+
 class DistributedCollectorNode:
     @classmethod
     def INPUT_TYPES(cls):
@@ -85,7 +92,7 @@ class DistributedCollectorNode:
         rt = get_runtime()
         url = f"{master_url}/distributed/job_complete"
         n = images.shape[0]
-        has_audio = audio is not None and audio["waveform"].numel() > 1
+        has_audio = _is_real_audio(audio)
         if n == 0:
             payload = {
                 "job_id": str(job_id), "worker_id": str(worker_id),
@@ -194,7 +201,7 @@ class DistributedCollectorNode:
     @staticmethod
     def _combine_audio(local_audio, worker_audio, enabled_order):
         audios = []
-        if local_audio is not None and local_audio["waveform"].numel() > 1:
+        if _is_real_audio(local_audio):
             audios.append(local_audio)
         seen = set()
         for wid in enabled_order:

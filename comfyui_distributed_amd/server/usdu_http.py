@@ -255,12 +255,8 @@ def _worker_dynamic(params: USDUParams, stack, cond, uncond, image, job_id,
             break
         time.sleep(constants.JOB_READY_POLL_INTERVAL)
     while True:
-        resp = run_async_in_server_loop(
-            rt.post_json(f"{master_url}/distributed/request_image",
-                         {"job_id": job_id, "worker_id": worker_id}),
-            timeout=60.0,
-        )
-        idx = resp.get("image_idx")
+        resp = _request_work(rt, master_url, job_id, worker_id)
+        idx = resp.get("image_idx") if resp is not None else None
         if idx is None:
             break
         idx = int(idx)
@@ -317,12 +313,8 @@ def _worker_static(params: USDUParams, stack, cond, uncond, image, job_id,
 
     processed = 0
     while True:
-        resp = run_async_in_server_loop(
-            rt.post_json(f"{master_url}/distributed/request_image",
-                         {"job_id": job_id, "worker_id": worker_id}),
-            timeout=60.0,
-        )
-        idx = resp.get("tile_idx")
+        resp = _request_work(rt, master_url, job_id, worker_id)
+        idx = resp.get("tile_idx") if resp is not None else None
         if idx is None:
             break
         res = sample_tiles(stack, cond, uncond, params, canvas, plans, [int(idx)])
@@ -341,6 +333,29 @@ def _worker_static(params: USDUParams, stack, cond, uncond, image, job_id,
             flush(False)
     flush(True)
     debug_log(f"usdu worker {worker_id}: processed {processed} tiles")
+    return None
+
+
+def _request_work(rt, master_url: str, job_id, worker_id) -> dict | None:
+    """Pull the next work item. A 404 means the job is gone — either not
+    yet initialized (transient, retried briefly) or already completed and
+    cleaned up by the master; both end with a graceful None instead of
+    crashing the worker's prompt (reference worker_comms.py:124-188 does a
+    404-retry loop for the same race)."""
+    for attempt in range(3):
+        try:
+            return run_async_in_server_loop(
+                rt.post_json(f"{master_url}/distributed/request_image",
+                             {"job_id": job_id, "worker_id": worker_id}),
+                timeout=60.0,
+            )
+        except Exception as exc:  # noqa: BLE001
+            if getattr(exc, "status", None) == 404:
+                if attempt < 2:
+                    time.sleep(0.2)
+                    continue
+                return None
+            raise
     return None
 
 
