@@ -187,16 +187,6 @@ async def orchestrate_distributed_execution(
     for jid in job_id_map.values():
         await job_state.ensure_queue(jid)
 
-    master_graph = graph
-    if delegate:
-        master_graph = transform.prepare_delegate_master_prompt(
-            graph, graph.nodes_of_class(NODE_CLASS_COLLECTOR)
-        )
-    master_prompt = transform.apply_participant_overrides(
-        master_graph, is_master=True, participant_id="master",
-        enabled_worker_ids=enabled_ids, job_id_map=job_id_map,
-    )
-
     sem = asyncio.Semaphore(constants.WORKER_PREP_CONCURRENCY)
 
     async def prep_and_dispatch(worker):
@@ -215,9 +205,27 @@ async def orchestrate_distributed_execution(
     failed = [wid for wid, ok in dispatched if not ok]
     if failed:
         log(f"orchestration: dispatch failed for {failed}")
+    ok_ids = [wid for wid, ok in dispatched if ok]
+
+    # Build the master prompt AFTER dispatch so its collectors only expect
+    # workers that actually received the job — a mid-dispatch failure must
+    # not cost the collector its full straggler timeout.
+    if delegate and not ok_ids:
+        trace_debug(trace, "delegate-only but every dispatch failed — "
+                           "master runs the original prompt")
+        pid = await enqueue_local(payload.prompt, payload.client_id)
+        return {"status": "queued", "participants": ["master"], "job_ids": {},
+                "master_prompt_id": pid}
+    master_graph = graph
+    if delegate:
+        master_graph = transform.prepare_delegate_master_prompt(
+            graph, graph.nodes_of_class(NODE_CLASS_COLLECTOR)
+        )
+    master_prompt = transform.apply_participant_overrides(
+        master_graph, is_master=True, participant_id="master",
+        enabled_worker_ids=ok_ids, job_id_map=job_id_map,
+    )
     pid = await enqueue_local(master_prompt.raw, payload.client_id)
-    participants = (["master"] if not delegate else []) + [
-        wid for wid, ok in dispatched if ok
-    ]
+    participants = (["master"] if not delegate else []) + ok_ids
     return {"status": "queued", "participants": participants,
             "job_ids": job_id_map, "master_prompt_id": pid}

@@ -55,7 +55,9 @@ def run_orch(payload, monkeypatch, probes, dispatch_ok=True):
     async def fake_dispatch(worker, prompt, client_id, timeout=30.0,
                             use_websocket=None):
         calls["dispatched"].append((str(worker["id"]), prompt))
-        return dispatch_ok
+        if isinstance(dispatch_ok, bool):
+            return dispatch_ok
+        return str(worker["id"]) in dispatch_ok
 
     async def enqueue_local(prompt, client_id):
         calls["local"].append(prompt)
@@ -191,3 +193,24 @@ def test_delegate_with_usdu_falls_back_to_master(two_worker_config, monkeypatch)
     # master participates despite delegate flag (reference limitation kept)
     assert "master" in result["participants"]
     assert "master_prompt_id" in result
+
+
+def test_failed_dispatch_excluded_from_master_collector(two_worker_config,
+                                                        monkeypatch):
+    """If a worker accepts the probe but the dispatch fails, the master's
+    collector must not wait for it: the master prompt is built after
+    dispatch with only the successful ids."""
+    import json
+
+    payload = QueueRequestPayload(prompt=dist_prompt(), client_id="c",
+                                  enabled_worker_ids=["w1", "w2"])
+    probes = {"w1": {"exec_info": {"queue_remaining": 0}},
+              "w2": {"exec_info": {"queue_remaining": 0}}}
+    result, calls, _ = run_orch(payload, monkeypatch, probes,
+                                dispatch_ok={"w1"})
+    assert result["participants"] == ["master", "w1"]
+    assert len(calls["local"]) == 1
+    master_prompt = calls["local"][0]
+    collector = next(v for v in master_prompt.values()
+                     if v["class_type"] == "DistributedCollector")
+    assert json.loads(collector["inputs"]["enabled_worker_ids"]) == ["w1"]
