@@ -446,9 +446,20 @@ class DistributedServer:
     # ---- introspection ----------------------------------------------------
 
     async def post_check_file(self, request):
+        """md5-check a synced media file. ``filename`` resolves against THIS
+        server's input dir (the cross-machine case — the sender's absolute
+        path means nothing here); ``path`` is the same-filesystem fallback."""
         data = await request.json()
-        path = data.get("path")
-        if not path or not os.path.isfile(path):
+        candidates = []
+        name = data.get("filename")
+        if name:
+            candidates.append(os.path.join(
+                str(self.executor.context.get("input_dir", "input")),
+                os.path.basename(str(name))))
+        if data.get("path"):
+            candidates.append(str(data["path"]))
+        path = next((p for p in candidates if os.path.isfile(p)), None)
+        if path is None:
             return web.json_response({"exists": False})
         h = hashlib.md5()
         with open(path, "rb") as fh:

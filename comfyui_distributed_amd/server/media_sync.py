@@ -114,7 +114,8 @@ async def sync_worker_media(prompt: dict, worker: dict,
             if remote.get("exists") and remote.get("md5") == local_md5:
                 return
             data = aiohttp.FormData()
-            data.add_field("image", open(local, "rb"), filename=filename)
+            # bytes, not an open handle: FormData would leak the fd
+            data.add_field("image", local.read_bytes(), filename=filename)
             try:
                 async with session.post(
                     f"{url}/upload/image", data=data,

@@ -31,3 +31,35 @@ def test_sync_local_worker_is_noop():
     prompt = {"1": {"class_type": "LoadImage", "inputs": {"image": "cat.png"}}}
     out = asyncio.run(sync_worker_media(prompt, {"id": "w", "type": "local"}))
     assert out == prompt
+
+
+def test_check_file_resolves_filename_against_input_dir(tmp_config, tmp_path):
+    """The cross-machine md5 skip: the worker resolves `filename` against
+    its OWN input dir instead of trusting the sender's absolute path."""
+    import asyncio
+    import hashlib
+
+    from aiohttp.test_utils import TestClient, TestServer
+
+    from comfyui_distributed_amd.server.app import DistributedServer
+
+    async def go():
+        srv = DistributedServer()
+        srv.executor.context["input_dir"] = str(tmp_path)
+        (tmp_path / "cat.png").write_bytes(b"pngdata")
+        cl = TestClient(TestServer(srv.build_app()))
+        await cl.start_server()
+        try:
+            r = await cl.post("/distributed/check_file",
+                              json={"filename": "cat.png",
+                                    "path": "/nonexistent/master/cat.png"})
+            body = await r.json()
+            assert body["exists"] is True
+            assert body["md5"] == hashlib.md5(b"pngdata").hexdigest()
+            r = await cl.post("/distributed/check_file",
+                              json={"filename": "missing.png"})
+            assert (await r.json())["exists"] is False
+        finally:
+            await cl.close()
+
+    asyncio.run(go())
