@@ -197,10 +197,14 @@ class VAE(nn.Module):
         step = tile - overlap
         out = None
         weight = None
+        seen: set = set()
         for y0 in range(0, h, step):
             for x0 in range(0, w, step):
                 y1, x1 = min(y0 + tile, h), min(x0 + tile, w)
                 ys, xs = max(0, y1 - tile), max(0, x1 - tile)
+                if (ys, xs) in seen:  # edge re-anchoring can repeat a tile
+                    continue
+                seen.add((ys, xs))
                 piece = self.decode(latents[:, :, ys:y1, xs:x1])  # [B,ph,pw,3]
                 ph, pw = piece.shape[1], piece.shape[2]
                 if out is None:
