@@ -124,6 +124,12 @@ class PromptGraph:
         closure.discard(str(node_id))
         return any(self.class_of(nid) == target_class for nid in closure)
 
+    def add_node(self, node_id: str, node: dict) -> None:
+        """Insert a node and invalidate the class index (direct ``raw``
+        writes after a ``nodes_of_class`` call would leave it stale)."""
+        self.raw[str(node_id)] = node
+        self._by_class = None
+
     def next_free_id(self) -> str:
         """Smallest unused positive numeric id (prompt ids are numeric
         strings in the ComfyUI wire format)."""

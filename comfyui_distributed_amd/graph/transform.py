@@ -59,10 +59,10 @@ def prune_prompt_for_worker(graph: PromptGraph) -> PromptGraph:
     for nid in dist_nodes:
         if nid not in consumed:
             sink_id = pruned.next_free_id()
-            pruned.raw[sink_id] = {
+            pruned.add_node(sink_id, {
                 "class_type": NODE_CLASS_PREVIEW,
                 "inputs": {"images": [nid, 0]},
-            }
+            })
     return pruned
 
 
@@ -98,10 +98,10 @@ def prepare_delegate_master_prompt(
                 del inputs[name]
         inputs["delegate_only"] = True
     if placeholder_needed:
-        out.raw[placeholder_id] = {
+        out.add_node(placeholder_id, {
             "class_type": NODE_CLASS_EMPTY_IMAGE,
             "inputs": {"width": 64, "height": 64},
-        }
+        })
     # Drop dangling links of downstream nodes whose sources were pruned.
     for nid in out.node_ids():
         inputs = out.inputs(nid)
