@@ -22,6 +22,8 @@ def load_wf(name):
     "distributed_wan_video.json",
     "distributed_upscale_video.json",
     "distributed_flux_txt2img.json",
+    "parameter_sweep.json",
+    "distributed_audio_collect.json",
 ])
 def test_workflows_validate(name):
     validate_prompt(load_wf(name), default_registry())
