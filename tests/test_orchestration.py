@@ -214,3 +214,25 @@ def test_failed_dispatch_excluded_from_master_collector(two_worker_config,
     collector = next(v for v in master_prompt.values()
                      if v["class_type"] == "DistributedCollector")
     assert json.loads(collector["inputs"]["enabled_worker_ids"]) == ["w1"]
+
+
+def test_probe_concurrency_is_bounded(monkeypatch):
+    """probe_workers must respect the concurrency semaphore (reference
+    tests/test_dispatch_selection.py:167)."""
+    import asyncio
+
+    peak = {"now": 0, "max": 0}
+
+    async def slow_probe(url, timeout=5.0):
+        peak["now"] += 1
+        peak["max"] = max(peak["max"], peak["now"])
+        await asyncio.sleep(0.02)
+        peak["now"] -= 1
+        return {"exec_info": {"queue_remaining": 0}}
+
+    monkeypatch.setattr(orchestration.network, "probe_worker", slow_probe)
+    workers = [{"id": f"w{i}", "host": "", "port": 8000 + i}
+               for i in range(12)]
+    res = asyncio.run(orchestration.probe_workers(workers, concurrency=3))
+    assert len(res) == 12 and all(v is not None for v in res.values())
+    assert peak["max"] <= 3
