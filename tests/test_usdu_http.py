@@ -12,6 +12,7 @@ from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
 from comfyui_distributed_amd.models import create_diffusion_stack
 from comfyui_distributed_amd.nodes.runtime import NodeRuntime, set_runtime
 from comfyui_distributed_amd.server import usdu_http
+from comfyui_distributed_amd.utils import constants
 from comfyui_distributed_amd.server.job_state import TileJobState
 
 
@@ -135,3 +136,45 @@ def test_timeout_grace_for_busy_worker():
         assert job.pending_tasks.empty()
 
     asyncio.run(go())
+
+
+class GoneJobRuntime(NodeRuntime):
+    """Master whose job is already cleaned up: request_image 404s."""
+
+    def __init__(self):
+        super().__init__()
+        self.completions = []
+
+    async def post_json(self, url, payload, timeout=60.0):
+        if url.endswith("/distributed/job_status"):
+            return {"ready": True}
+        if url.endswith("/distributed/request_image"):
+            err = RuntimeError("404, message='unknown job'")
+            err.status = 404
+            raise err
+        if url.endswith("/distributed/submit_tiles"):
+            self.completions.append(payload)
+            return {"status": "ok"}
+        if url.endswith("/distributed/heartbeat"):
+            return {"status": "ok"}
+        raise AssertionError(f"unexpected url {url}")
+
+
+def test_worker_static_graceful_when_job_gone(monkeypatch):
+    """A worker whose master finished and cleaned the job must exit
+    gracefully AND still send the empty-batch completion signal (reference
+    test_static_mode.py: flush_empty_final_still_sends_completion_signal +
+    the 404-retry in worker_comms.py:124-188)."""
+    monkeypatch.setattr(constants, "JOB_READY_POLL_ATTEMPTS", 1)
+    stack, cond, p, img = tiny_setup()
+    rt = GoneJobRuntime()
+    set_runtime(rt)
+    usdu_http.run_usdu_role(
+        mode="static", params=p, stack=stack, cond=cond, uncond=None,
+        image=img, job_id="gone", is_worker=True,
+        master_url="http://master:8188", enabled_workers=["w1"],
+        worker_id="w1",
+    )
+    assert len(rt.completions) == 1
+    final = rt.completions[0]
+    assert final["is_last"] is True and final["tiles"] == []
