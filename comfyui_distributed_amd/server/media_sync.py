@@ -77,14 +77,32 @@ async def fetch_worker_system_info(worker_url: str) -> dict:
     return {"path_separator": "/"}
 
 
+async def is_local_worker(worker: dict) -> bool:
+    """Same-machine check (reference workers/detection.py:11-47): loopback
+    host or type "local" is local by definition; a "cloud" worker may still
+    be co-located (e.g. one container per GPU on one box) — compare its
+    machine id from /distributed/system_info against ours."""
+    host = network.normalize_host(worker.get("host", "")) or ""
+    if host in ("", "localhost", "127.0.0.1", "0.0.0.0") or \
+            worker.get("type", "local") == "local":
+        return True
+    if worker.get("type") == "cloud":
+        import uuid
+
+        info = await fetch_worker_system_info(network.build_worker_url(worker))
+        return info.get("machine_id") == hex(uuid.getnode())
+    return False
+
+
 async def sync_worker_media(prompt: dict, worker: dict,
                             input_dir: str | Path = "input",
                             concurrency: int | None = None) -> dict:
     """Upload referenced media files the worker is missing; returns the
-    prompt with rewritten worker-local paths. Local workers (shared FS)
-    are a no-op (reference behavior: sync only for remote/cloud)."""
+    prompt with rewritten worker-local paths. Workers on this machine
+    (shared FS) are a no-op (reference behavior: sync only for
+    genuinely-remote workers)."""
     refs = find_media_references(prompt)
-    if not refs or worker.get("type", "local") == "local":
+    if not refs or await is_local_worker(worker):
         return prompt
     url = network.build_worker_url(worker)
     info = await fetch_worker_system_info(url)

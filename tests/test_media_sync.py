@@ -63,3 +63,36 @@ def test_check_file_resolves_filename_against_input_dir(tmp_config, tmp_path):
             await cl.close()
 
     asyncio.run(go())
+
+
+def test_is_local_worker_heuristics(tmp_config):
+    import asyncio
+    import uuid
+
+    from comfyui_distributed_amd.server import media_sync
+
+    run = asyncio.run
+    assert run(media_sync.is_local_worker({"host": "", "type": "remote"}))
+    assert run(media_sync.is_local_worker({"host": "127.0.0.1",
+                                           "type": "remote"}))
+    assert run(media_sync.is_local_worker({"host": "10.0.0.9",
+                                           "type": "local"}))
+    assert not run(media_sync.is_local_worker({"host": "10.0.0.9",
+                                               "type": "remote"}))
+
+    # cloud worker: machine-id comparison decides (faked system_info)
+    async def same_info(url):
+        return {"machine_id": hex(uuid.getnode())}
+
+    async def other_info(url):
+        return {"machine_id": "0xdeadbeef"}
+
+    w = {"host": "198.51.100.4", "port": 8189, "type": "cloud"}
+    orig = media_sync.fetch_worker_system_info
+    try:
+        media_sync.fetch_worker_system_info = same_info
+        assert run(media_sync.is_local_worker(w)) is True
+        media_sync.fetch_worker_system_info = other_info
+        assert run(media_sync.is_local_worker(w)) is False
+    finally:
+        media_sync.fetch_worker_system_info = orig
