@@ -23,3 +23,20 @@ with torch.distributed over RCCL.
 """
 
 __version__ = "0.1.0"
+
+# Node maps at package root (reference __init__.py:1-29 exposes the same
+# names so a host app can discover the node classes; kept lazy so plain
+# `import comfyui_distributed_amd` stays light).
+
+
+def __getattr__(name):
+    if name in ("NODE_CLASS_MAPPINGS", "NODE_DISPLAY_NAME_MAPPINGS"):
+        from . import nodes as _nodes
+        from .graph import builtin_nodes as _builtin
+
+        if name == "NODE_CLASS_MAPPINGS":
+            merged = dict(_builtin.BUILTIN_CLASS_MAPPINGS)
+            merged.update(_nodes.NODE_CLASS_MAPPINGS)
+            return merged
+        return dict(_nodes.NODE_DISPLAY_NAME_MAPPINGS)
+    raise AttributeError(name)
