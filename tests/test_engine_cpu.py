@@ -84,3 +84,24 @@ def test_per_tile_noise_deterministic():
     n3 = _tile_noise(7, 4, 0, (4, 4, 4))
     assert torch.equal(n1, n2)
     assert not torch.equal(n1, n3)
+
+
+def test_usdu_with_adm_conditioned_model():
+    """SDXL-style vector conditioning (cond["y"]) flows through the USDU
+    per-tile sampler (CFGDenoiser y-batching incl. the CFG concat path)."""
+    from comfyui_distributed_amd.models.registry import (
+        TINY_VAE, DiffusionStack, StackConfig, TINY_UNET)
+    from dataclasses import replace
+
+    cfg = StackConfig("tinyxl", replace(TINY_UNET, adm_in_channels=16),
+                      TINY_VAE, native_size=64)
+    stack = DiffusionStack(cfg, seed=3)
+    cond = stack.make_conditioning(0)
+    uncond = stack.make_conditioning(1)
+    assert cond.get("y") is not None and cond["y"].shape == (1, 16)
+    p = USDUParams(seed=1, steps=1, cfg=3.0, denoise=0.5, tile_width=16,
+                   tile_height=16, padding=16, mask_blur=2, tile_batch=4)
+    img = torch.rand(1, 32, 32, 3, generator=torch.Generator().manual_seed(4))
+    out = process_single_gpu(stack, cond, uncond, p, img)
+    assert out.shape == (1, 32, 32, 3)
+    assert torch.isfinite(out).all()
