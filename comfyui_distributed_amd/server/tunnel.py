@@ -109,10 +109,16 @@ class TunnelManager:
         async with self._lock:
             if self.proc is not None:
                 self.proc.terminate()
-                try:
-                    self.proc.wait(timeout=5)
-                except subprocess.TimeoutExpired:
-                    self.proc.kill()
+
+                def _reap(p=self.proc):
+                    try:
+                        p.wait(timeout=5)
+                    except subprocess.TimeoutExpired:
+                        p.kill()
+                        p.wait(timeout=5)
+
+                # off-loop: waiting on the child must not stall the server
+                await asyncio.get_running_loop().run_in_executor(None, _reap)
                 self.proc = None
             self.url = None
             async with config_transaction() as cfg:
