@@ -53,6 +53,7 @@ class DistributedServer:
         self.device = device
         self.prompt_queue: asyncio.Queue = asyncio.Queue()
         self.executing = False
+        self.history: dict[str, dict] = {}
         self.executor = Executor(context={"device": device} if device else {})
         self.registry = self.executor.registry
         self._exec_task: asyncio.Task | None = None
@@ -99,6 +100,8 @@ class DistributedServer:
         r.add_post("/distributed/check_file", self.post_check_file)
         r.add_post("/upload/image", self.post_upload_image)
         r.add_get("/view", self.get_view)
+        r.add_get("/history", self.get_history)
+        r.add_get("/history/{prompt_id}", self.get_history)
         r.add_get("/distributed/network_info", self.get_network_info)
         r.add_get("/distributed/system_info", self.get_system_info)
         r.add_post("/distributed/tunnel/start", self.post_tunnel_start)
@@ -138,13 +141,37 @@ class DistributedServer:
             from ..nodes.runtime import get_runtime
 
             get_runtime().clear_interrupt()  # interrupts are per-prompt
+            saved = self.executor.context.setdefault("saved_images", [])
+            n_before = len(saved)
             try:
                 await loop.run_in_executor(None, self.executor.execute, prompt)
                 log(f"prompt {prompt_id} done (client {client_id})")
+                self._record_history(prompt_id, "success",
+                                     saved[n_before:], None)
             except Exception as exc:  # noqa: BLE001
                 log(f"prompt {prompt_id} FAILED: {exc!r}")
+                self._record_history(prompt_id, "error",
+                                     saved[n_before:], repr(exc))
             finally:
                 self.executing = False
+
+    HISTORY_LIMIT = 200
+
+    def _record_history(self, prompt_id, status, new_files, error):
+        """ComfyUI /history parity: clients poll it to learn a prompt's
+        outcome and fetch outputs via /view."""
+        entry = {
+            "status": {"status_str": status,
+                       "completed": status == "success",
+                       "messages": [error] if error else []},
+            "outputs": {"images": [
+                {"filename": os.path.basename(str(f)), "type": "output"}
+                for f in new_files
+            ]},
+        }
+        self.history[str(prompt_id)] = entry
+        while len(self.history) > self.HISTORY_LIMIT:
+            self.history.pop(next(iter(self.history)))
 
     async def enqueue_local(self, prompt: dict, client_id: str) -> str:
         validate_prompt(prompt, self.registry)
@@ -491,6 +518,13 @@ class DistributedServer:
                         fh.write(chunk)
                 saved.append(fname)
         return web.json_response({"saved": saved})
+
+    async def get_history(self, request):
+        pid = request.match_info.get("prompt_id")
+        if pid is None:
+            return web.json_response(self.history)
+        entry = self.history.get(str(pid))
+        return web.json_response({str(pid): entry} if entry else {})
 
     async def get_view(self, request):
         """Serve a saved output (ComfyUI GET /view parity: the panel and

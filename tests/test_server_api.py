@@ -376,3 +376,51 @@ def test_interrupt_fanout_and_view(client, tmp_path):
         await close_client_session()
 
     run(loop, go())
+
+
+def test_history_records_prompt_outcomes(client, tmp_path):
+    srv, cl, loop = client
+
+    async def go():
+        srv.executor.context["output_dir"] = str(tmp_path)
+        srv.executor.context["device"] = "cpu"
+        good = {
+            "1": {"class_type": "LoadImage",
+                  "inputs": {"image": "synthetic:8x8"}},
+            "2": {"class_type": "SaveImage",
+                  "inputs": {"images": ["1", 0], "filename_prefix": "hist"}},
+        }
+        r = await cl.post("/prompt", json={"prompt": good, "client_id": "h"})
+        assert r.status == 200
+        pid = (await r.json())["prompt_id"]
+        for _ in range(100):
+            r = await cl.get(f"/history/{pid}")
+            body = await r.json()
+            if body:
+                break
+            await asyncio.sleep(0.05)
+        entry = body[pid]
+        assert entry["status"]["completed"] is True
+        images = entry["outputs"]["images"]
+        assert images and images[0]["filename"].startswith("hist")
+        # the recorded output is fetchable via /view
+        r = await cl.get(f"/view?filename={images[0]['filename']}")
+        assert r.status == 200
+
+        bad = {"1": {"class_type": "LoadImage",
+                     "inputs": {"image": "does-not-exist.png"}},
+               "2": {"class_type": "SaveImage",
+                     "inputs": {"images": ["1", 0]}}}
+        r = await cl.post("/prompt", json={"prompt": bad, "client_id": "h"})
+        pid2 = (await r.json())["prompt_id"]
+        for _ in range(100):
+            body = await (await cl.get(f"/history/{pid2}")).json()
+            if body:
+                break
+            await asyncio.sleep(0.05)
+        assert body[pid2]["status"]["completed"] is False
+        # full listing contains both
+        full = await (await cl.get("/history")).json()
+        assert pid in full and pid2 in full
+
+    run(loop, go())
