@@ -329,7 +329,8 @@ def test_panel_served_with_expected_controls(client):
             html = await r.text()
             for needle in ("worker_status", "auto_populate_workers",
                            "update_worker", "delete_worker", "update_setting",
-                           "tunnel", "local_log", "distributed/queue"):
+                           "tunnel", "local_log", "distributed/queue",
+                           "/history", "/view?filename="):
                 assert needle in html, f"panel missing wiring for {needle}"
 
     run(loop, go())
