@@ -100,3 +100,70 @@ def test_seed_parallel_roundtrip_over_http(tmp_config, monkeypatch):
     # master image + worker image, combined on the master
     assert combined.shape == (2, 16, 16, 3)
     assert torch.isfinite(combined).all()
+
+
+@pytest.mark.timeout(180)
+def test_three_concurrent_collector_jobs(tmp_config, monkeypatch):
+    """Three distributed jobs queued back-to-back: per-job collector queues
+    must stay isolated (no cross-job image leakage), and the sequential
+    prompt loops on both servers must drain all of them."""
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.2)
+
+    async def go():
+        worker_srv = DistributedServer(is_worker=True)
+        wc = TestClient(TestServer(worker_srv.build_app()))
+        await wc.start_server()
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "w", "host": "127.0.0.1",
+            "port": wc.server.port, "cuda_device": 0, "enabled": True,
+            "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+
+        def prompt_for(size):
+            return {
+                "1": {"class_type": "DistributedSeed", "inputs": {"seed": size}},
+                "2": {"class_type": "LoadImage",
+                      "inputs": {"image": f"synthetic:{size}x{size}"}},
+                "3": {"class_type": "DistributedCollector",
+                      "inputs": {"images": ["2", 0], "load_balance": False}},
+                "4": {"class_type": "PreviewImage", "inputs": {"images": ["3", 0]}},
+            }
+
+        sizes = [8, 12, 16]
+        for s in sizes:
+            r = await mc.post("/distributed/queue", json={
+                "prompt": prompt_for(s), "client_id": f"c{s}",
+                "enabled_worker_ids": ["w1"]})
+            assert r.status == 200
+
+        for _ in range(600):
+            if len(previews) >= 3:
+                break
+            await asyncio.sleep(0.2)
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await wc.close()
+        await mc.close()
+        return previews
+
+    previews = asyncio.run(go())
+    assert len(previews) == 3
+    # jobs completed in submission order; each combined batch = 2 ranks of
+    # the RIGHT size (no cross-job mixing)
+    got = sorted(p.shape[1] for p in previews)
+    assert got == [8, 12, 16]
+    for p in previews:
+        assert p.shape[0] == 2 and torch.isfinite(p).all()
