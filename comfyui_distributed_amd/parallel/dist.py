@@ -115,24 +115,40 @@ def gather_tensor_lists(
     ndim = int(shape_t[0].item())
     ref_shape = tuple(int(x) for x in shape_t[1 : 1 + ndim])
 
+    # One batched p2p group: every source's isend pairs with a
+    # pre-posted irecv on rank 0, so all sources stream CONCURRENTLY over
+    # their own xGMI links instead of rank 0 draining them one at a time
+    # (xGMI is 7 point-to-point links per GPU — serial recv leaves 6 idle).
     if ctx.is_master:
-        all_tensors: list[torch.Tensor] = list(tensors)
-        all_meta: list[tuple] = list(meta)
+        bufs: dict[int, tuple[torch.Tensor, torch.Tensor]] = {}
+        ops = []
         for src in range(1, ctx.world_size):
             n = counts[src]
             if n == 0:
                 continue
             buf = torch.empty((n, *ref_shape), dtype=torch.float32, device=device)
             mbuf = torch.empty((n, meta_width), dtype=torch.int64, device=device)
-            dist.recv(buf, src=src)
-            dist.recv(mbuf, src=src)
-            for i in range(n):
+            bufs[src] = (buf, mbuf)
+            ops.append(dist.P2POp(dist.irecv, buf, src))
+            ops.append(dist.P2POp(dist.irecv, mbuf, src))
+        if ops:
+            for work in dist.batch_isend_irecv(ops):
+                work.wait()
+        all_tensors: list[torch.Tensor] = list(tensors)
+        all_meta: list[tuple] = list(meta)
+        for src in range(1, ctx.world_size):
+            if src not in bufs:
+                continue
+            buf, mbuf = bufs[src]
+            for i in range(counts[src]):
                 all_tensors.append(buf[i])
                 all_meta.append(tuple(int(x) for x in mbuf[i]))
         return all_tensors, all_meta
     if counts[ctx.rank] > 0:
-        dist.send(payload.to(torch.float32), dst=0)
-        dist.send(meta_t, dst=0)
+        ops = [dist.P2POp(dist.isend, payload.to(torch.float32), 0),
+               dist.P2POp(dist.isend, meta_t, 0)]
+        for work in dist.batch_isend_irecv(ops):
+            work.wait()
     return None
 
 
