@@ -102,7 +102,15 @@ def sample_tiles(
     if not tile_indices:
         return {}
     B = canvas.shape[0]
-    is_flow = getattr(stack, "family", "sd") == "flux"
+    family = getattr(stack, "family", "sd")
+    is_flow = family == "flux"
+    if not is_flow and not hasattr(stack, "schedule"):
+        raise ValueError(
+            f"model family {family!r} cannot drive the tile upscaler: "
+            "it has no image img2img path (video models upscale per-frame "
+            "through an IMAGE model — see workflows/"
+            "distributed_upscale_video.json)"
+        )
     if is_flow:
         sigmas = base_denoiser = None  # flow models integrate velocity
     else:

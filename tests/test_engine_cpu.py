@@ -105,3 +105,15 @@ def test_usdu_with_adm_conditioned_model():
     out = process_single_gpu(stack, cond, uncond, p, img)
     assert out.shape == (1, 32, 32, 3)
     assert torch.isfinite(out).all()
+
+
+def test_usdu_rejects_video_stack_with_clear_error():
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+
+    stack = create_diffusion_stack("wan_tiny")
+    cond = stack.make_conditioning(0)
+    p = USDUParams(seed=1, steps=1, cfg=1.0, denoise=0.5, tile_width=16,
+                   tile_height=16, padding=16, mask_blur=2)
+    img = torch.rand(1, 32, 32, 3)
+    with pytest.raises(ValueError, match="per-frame"):
+        process_single_gpu(stack, cond, None, p, img)
