@@ -1,5 +1,6 @@
 """End-to-end engine tests on CPU with the tiny stack."""
 
+import pytest
 import torch
 
 from comfyui_distributed_amd.engine import (
