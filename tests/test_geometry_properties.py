@@ -101,3 +101,40 @@ def test_blend_stays_in_convex_hull(h, w, ow, oh):
     lo = torch.minimum(before, torch.full_like(before, 0.5)) - 1e-4
     hi = torch.maximum(before, torch.full_like(before, 0.5)) + 1e-4
     assert (canvas >= lo).all() and (canvas <= hi).all()
+
+
+@given(st.integers(1, 4), st.integers(1, 32), st.integers(1, 32))
+@settings(max_examples=40, deadline=None)
+def test_png_codec_roundtrip_exact_uint8(b, h, w):
+    """PNG is lossless: any uint8-quantized image survives the wire
+    bit-exactly (the collector determinism contract depends on it)."""
+    import torch
+
+    from comfyui_distributed_amd.utils.image import (
+        decode_png_base64, encode_png_base64)
+
+    g = torch.Generator().manual_seed(b * 7919 + h * 31 + w)
+    raw = torch.randint(0, 256, (b, h, w, 3), generator=g, dtype=torch.uint8)
+    img = raw.float() / 255.0
+    for i in range(b):  # the wire sends one image per envelope
+        out = decode_png_base64(encode_png_base64(img[i : i + 1]))
+        assert out.shape == (1, h, w, 3)
+        assert torch.equal((out * 255.0).round().to(torch.uint8),
+                           raw[i : i + 1])
+
+
+@given(st.integers(1, 2), st.integers(1, 2), st.integers(1, 5000),
+       st.sampled_from([8000, 22050, 44100]))
+@settings(max_examples=40, deadline=None)
+def test_audio_envelope_roundtrip(b, c, n, sr):
+    import torch
+
+    from comfyui_distributed_amd.utils.audio import (
+        decode_audio_payload, encode_audio_payload)
+
+    g = torch.Generator().manual_seed(n)
+    audio = {"waveform": torch.randn(b, c, n, generator=g),
+             "sample_rate": sr}
+    out = decode_audio_payload(encode_audio_payload(audio))
+    assert out["sample_rate"] == sr
+    assert torch.equal(out["waveform"], audio["waveform"])
