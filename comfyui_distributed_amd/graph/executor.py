@@ -47,13 +47,31 @@ def validate_prompt(prompt: dict, registry: NodeRegistry) -> None:
     graph = PromptGraph(prompt)
     for nid, node in graph.items():
         cls_name = node.get("class_type", "")
-        if registry.get(cls_name) is None:
+        cls = registry.get(cls_name)
+        if cls is None:
             errors.setdefault(nid, []).append(f"unknown node class {cls_name!r}")
             continue
-        for name, value in node.get("inputs", {}).items():
+        inputs = node.get("inputs", {})
+        for name, value in inputs.items():
             if is_link(value) and str(value[0]) not in prompt:
                 errors.setdefault(nid, []).append(
                     f"input {name!r} links to missing node {value[0]!r}"
+                )
+        # required inputs without a declared default must be present
+        # (ComfyUI validates this at queue time; reference workflows rely
+        # on the structured node_errors response)
+        try:
+            spec = cls.INPUT_TYPES().get("required", {})
+        except Exception:  # noqa: BLE001 - INPUT_TYPES needing context
+            spec = {}
+        for name, decl in spec.items():
+            has_default = (
+                isinstance(decl, (tuple, list)) and len(decl) > 1
+                and isinstance(decl[1], dict) and "default" in decl[1]
+            )
+            if name not in inputs and not has_default:
+                errors.setdefault(nid, []).append(
+                    f"required input {name!r} missing"
                 )
     # cycle check via DFS
     state: dict[str, int] = {}

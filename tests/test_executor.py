@@ -124,3 +124,20 @@ def test_cross_run_output_caching():
                "8": {"class_type": "Sink", "inputs": {"x": ["9", 0]}}}
     ex.execute(renamed)
     assert calls["a"] == 2
+
+
+def test_validation_rejects_missing_required_input():
+    from comfyui_distributed_amd.graph.executor import (
+        default_registry, validate_prompt)
+    from comfyui_distributed_amd.utils.errors import PromptValidationError
+
+    # KSampler without its model input: caught at validation, not at
+    # execution (ComfyUI-parity structured node_errors)
+    bad = {"1": {"class_type": "KSampler", "inputs": {"seed": 1}}}
+    with pytest.raises(PromptValidationError) as exc:
+        validate_prompt(bad, default_registry())
+    msgs = exc.value.node_errors["1"]
+    assert any("model" in m for m in msgs)
+    # defaults still allowed to be absent
+    ok = {"1": {"class_type": "DistributedSeed", "inputs": {}}}
+    validate_prompt(ok, default_registry())
