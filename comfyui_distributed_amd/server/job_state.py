@@ -50,6 +50,10 @@ class ImageJobState:
 
 
 class JobState:
+    #: abandoned jobs (a prompt that died before its collector/USDU node
+    #: ran) are swept after this many seconds so queues can't leak forever
+    STALE_JOB_TTL = 3600.0
+
     def __init__(self):
         # collector jobs
         self.pending_jobs: dict[str, asyncio.Queue] = {}
@@ -57,15 +61,24 @@ class JobState:
         # tile/image jobs
         self.tile_jobs: dict[str, TileJobState | ImageJobState] = {}
         self.tile_jobs_lock = asyncio.Lock()
+        self._created: dict[str, float] = {}
+
+    def _sweep_locked(self, store: dict) -> None:
+        cutoff = time.monotonic() - self.STALE_JOB_TTL
+        for jid in [j for j in store if self._created.get(j, cutoff) < cutoff]:
+            store.pop(jid, None)
+            self._created.pop(jid, None)
 
     # ---- collector jobs ---------------------------------------------------
 
     async def ensure_queue(self, job_id: str) -> asyncio.Queue:
         async with self.jobs_lock:
+            self._sweep_locked(self.pending_jobs)
             q = self.pending_jobs.get(job_id)
             if q is None:
                 q = asyncio.Queue()
                 self.pending_jobs[job_id] = q
+                self._created[job_id] = time.monotonic()
             return q
 
     async def get_queue_waiting(
@@ -87,6 +100,7 @@ class JobState:
     async def drop_queue(self, job_id: str) -> None:
         async with self.jobs_lock:
             self.pending_jobs.pop(job_id, None)
+            self._created.pop(job_id, None)
 
     # ---- tile jobs --------------------------------------------------------
 
@@ -97,7 +111,9 @@ class JobState:
         for t in range(n_tiles):
             job.pending_tasks.put_nowait(t)
         async with self.tile_jobs_lock:
+            self._sweep_locked(self.tile_jobs)
             self.tile_jobs[job_id] = job
+            self._created[job_id] = time.monotonic()
         return job
 
     async def init_dynamic_job(self, job_id: str, n_images: int) -> ImageJobState:
@@ -105,7 +121,9 @@ class JobState:
         for i in range(n_images):
             job.pending_images.put_nowait(i)
         async with self.tile_jobs_lock:
+            self._sweep_locked(self.tile_jobs)
             self.tile_jobs[job_id] = job
+            self._created[job_id] = time.monotonic()
         return job
 
     async def get_tile_job(self, job_id: str):
@@ -115,3 +133,4 @@ class JobState:
     async def cleanup_job(self, job_id: str) -> None:
         async with self.tile_jobs_lock:
             self.tile_jobs.pop(job_id, None)
+            self._created.pop(job_id, None)

@@ -61,7 +61,17 @@ async def _master_static(mode, params: USDUParams, stack, cond, uncond, image,
     plans = plan_for_image(W, H, params)
     job = await state.init_static_job(job_id, len(plans), batch_size=B)
     loop = asyncio.get_running_loop()
+    try:
+        return await _master_static_body(mode, params, stack, cond, uncond,
+                                         canvas, B, plans, job, job_id, loop)
+    finally:
+        # interrupt/exception included: never leak the job entry
+        await state.cleanup_job(job_id)
 
+
+async def _master_static_body(mode, params, stack, cond, uncond, canvas, B,
+                              plans, job, job_id, loop):
+    rt = get_runtime()
     results: dict[tuple[int, int], torch.Tensor] = {}
 
     # -- master participates: pull tiles from its own queue ---------------
@@ -123,7 +133,6 @@ async def _master_static(mode, params: USDUParams, stack, cond, uncond, image,
                     break
 
     blend_results(canvas, results, plans, params)
-    await state.cleanup_job(job_id)
     return canvas
 
 
@@ -175,6 +184,16 @@ async def _master_dynamic(params: USDUParams, stack, cond, uncond, image,
     plans = plan_for_image(W, H, params)
     job = await state.init_dynamic_job(job_id, B)
     loop = asyncio.get_running_loop()
+    try:
+        return await _master_dynamic_body(params, stack, cond, uncond,
+                                          canvas, B, plans, job, loop)
+    finally:
+        await state.cleanup_job(job_id)
+
+
+async def _master_dynamic_body(params, stack, cond, uncond, canvas, B, plans,
+                               job, loop):
+    rt = get_runtime()
     done_images: dict[int, torch.Tensor] = {}
     timeout = get_worker_timeout_seconds()
     last_check = time.monotonic()
@@ -219,7 +238,6 @@ async def _master_dynamic(params: USDUParams, stack, cond, uncond, image,
                     break
     for i, img in done_images.items():
         canvas[i : i + 1] = img
-    await state.cleanup_job(job_id)
     return canvas
 
 

@@ -178,3 +178,23 @@ def test_worker_static_graceful_when_job_gone(monkeypatch):
     assert len(rt.completions) == 1
     final = rt.completions[0]
     assert final["is_last"] is True and final["tiles"] == []
+
+
+def test_master_static_cleans_job_on_failure(monkeypatch):
+    """A master whose sampling raises must still remove the job entry
+    (no stale tile jobs after interrupt/error)."""
+    stack, cond, p, img = tiny_setup()
+    rt = NodeRuntime()
+    set_runtime(rt)
+
+    def boom(*a, **k):
+        raise RuntimeError("sampler exploded")
+
+    monkeypatch.setattr(usdu_http, "sample_tiles", boom)
+    with pytest.raises(RuntimeError, match="exploded"):
+        usdu_http.run_usdu_role(
+            mode="static", params=p, stack=stack, cond=cond, uncond=None,
+            image=img, job_id="doomed", is_worker=False, master_url="",
+            enabled_workers=[], worker_id="",
+        )
+    assert asyncio.run(rt.job_state.get_tile_job("doomed")) is None
