@@ -53,9 +53,25 @@ def validate_prompt(prompt: dict, registry: NodeRegistry) -> None:
             continue
         inputs = node.get("inputs", {})
         for name, value in inputs.items():
-            if is_link(value) and str(value[0]) not in prompt:
+            if not is_link(value):
+                continue
+            src = str(value[0])
+            if src not in prompt:
                 errors.setdefault(nid, []).append(
                     f"input {name!r} links to missing node {value[0]!r}"
+                )
+                continue
+            src_cls = registry.get(prompt[src].get("class_type", ""))
+            rts = getattr(src_cls, "RETURN_TYPES", None)
+            # ByPassTypeTuple marks variable-output nodes (batch dividers):
+            # their socket count follows a widget, so no bounds check
+            # (same convention as the reference nodes/utilities.py:252)
+            if type(rts).__name__ == "ByPassTypeTuple":
+                continue
+            if rts is not None and int(value[1]) >= len(rts):
+                errors.setdefault(nid, []).append(
+                    f"input {name!r} links to output {value[1]} of node "
+                    f"{src!r} which has only {len(rts)} output(s)"
                 )
         # required inputs without a declared default must be present
         # (ComfyUI validates this at queue time; reference workflows rely

@@ -141,3 +141,17 @@ def test_validation_rejects_missing_required_input():
     # defaults still allowed to be absent
     ok = {"1": {"class_type": "DistributedSeed", "inputs": {}}}
     validate_prompt(ok, default_registry())
+
+
+def test_validation_rejects_out_of_range_output_index():
+    from comfyui_distributed_amd.graph.executor import (
+        default_registry, validate_prompt)
+    from comfyui_distributed_amd.utils.errors import PromptValidationError
+
+    bad = {
+        "1": {"class_type": "DistributedSeed", "inputs": {"seed": 1}},
+        "2": {"class_type": "PreviewImage", "inputs": {"images": ["1", 5]}},
+    }
+    with pytest.raises(PromptValidationError) as exc:
+        validate_prompt(bad, default_registry())
+    assert any("output 5" in m for m in exc.value.node_errors["2"])
