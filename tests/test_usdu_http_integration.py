@@ -309,3 +309,108 @@ def test_distributed_usdu_dynamic_over_http(tmp_config, monkeypatch):
     canvas = previews[0]
     assert canvas.shape == (2, 32, 32, 3)
     assert torch.isfinite(canvas).all()
+
+
+@pytest.mark.timeout(180)
+def test_zombie_worker_image_requeued_dynamic_mode(tmp_config, monkeypatch):
+    """Dynamic-mode fault injection: the zombie claims a whole-image index
+    via the pull queue and never submits; the monitor must requeue it and
+    the master must finish every image."""
+    from aiohttp import web
+
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+    monkeypatch.setattr(constants, "HEARTBEAT_INTERVAL", 0.3)
+
+    async def go():
+        fake = web.Application()
+
+        async def fake_get(_r):
+            return web.json_response({"exec_info": {"queue_remaining": 0}})
+
+        async def fake_post(_r):
+            return web.json_response({"prompt_id": "fake"})
+
+        fake.router.add_get("/prompt", fake_get)
+        fake.router.add_post("/prompt", fake_post)
+        fc = TestClient(TestServer(fake))
+        await fc.start_server()
+
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "wz", "name": "zombie", "host": "127.0.0.1",
+            "port": fc.server.port, "cuda_device": 0, "enabled": True,
+            "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        cfg["settings"]["worker_timeout_seconds"] = 1
+        save_config(cfg)
+
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "tiny"}},
+            "2": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "x", "clip": ["1", 1]}},
+            "3": {"class_type": "EmptyLatentImage",
+                  "inputs": {"width": 32, "height": 32, "batch_size": 2}},
+            "4": {"class_type": "KSampler", "inputs": {
+                "model": ["1", 0], "seed": 2, "steps": 1, "cfg": 1.0,
+                "sampler_name": "euler", "scheduler": "karras",
+                "positive": ["2", 0], "negative": ["2", 0],
+                "latent_image": ["3", 0], "denoise": 1.0}},
+            "5": {"class_type": "VAEDecode",
+                  "inputs": {"samples": ["4", 0], "vae": ["1", 2]}},
+            "6": {"class_type": "UltimateSDUpscaleDistributed", "inputs": {
+                "upscaled_image": ["5", 0], "model": ["1", 0],
+                "positive": ["2", 0], "negative": ["2", 0], "vae": ["1", 2],
+                "seed": 3, "steps": 1, "cfg": 1.0, "sampler_name": "euler",
+                "scheduler": "karras", "denoise": 0.5, "tile_width": 16,
+                "tile_height": 16, "padding": 16, "mask_blur": 2,
+                "force_uniform_tiles": True, "tiled_decode": False,
+                "dynamic_threshold": 2}},
+            "7": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["6", 0], "load_balance": False}},
+            "8": {"class_type": "PreviewImage", "inputs": {"images": ["7", 0]}},
+        }
+        resp = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "dz", "enabled_worker_ids": ["wz"]})
+        assert resp.status == 200
+        job_id = (await resp.json())["job_ids"]["6"]
+
+        # zombie: wait for readiness, claim one image index, vanish
+        claimed = None
+        for _ in range(200):
+            r = await mc.post("/distributed/job_status", json={"job_id": job_id})
+            if r.status == 200 and (await r.json()).get("ready"):
+                break
+            await asyncio.sleep(0.05)
+        r = await mc.post("/distributed/request_image", json={
+            "job_id": job_id, "worker_id": "wz"})
+        if r.status == 200:
+            claimed = (await r.json()).get("image_idx")
+        await mc.post("/distributed/heartbeat", json={
+            "job_id": job_id, "worker_id": "wz"})
+
+        for _ in range(600):
+            if previews:
+                break
+            await asyncio.sleep(0.25)
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await fc.close()
+        await mc.close()
+        return previews, claimed
+
+    previews, claimed = asyncio.run(go())
+    assert previews, "dynamic job never completed after zombie claim"
+    canvas = previews[0]
+    assert canvas.shape == (2, 32, 32, 3)
+    assert torch.isfinite(canvas).all()
