@@ -174,7 +174,11 @@ class Executor:
                 return cache[nid]
             node = graph.node(nid)
             cls = self.registry.get(node["class_type"])
-            is_output = bool(getattr(cls, "OUTPUT_NODE", False))
+            # OUTPUT_NODEs and nodes declaring IS_CHANGED (ComfyUI's
+            # always-re-execute convention, e.g. the USDU node's nan)
+            # never reuse a cross-run cached result
+            is_output = bool(getattr(cls, "OUTPUT_NODE", False)) or \
+                hasattr(cls, "IS_CHANGED")
             if not is_output and fps[nid] in self._run_cache:
                 out = self._run_cache[fps[nid]]
                 cache[nid] = out

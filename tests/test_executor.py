@@ -155,3 +155,73 @@ def test_validation_rejects_out_of_range_output_index():
     with pytest.raises(PromptValidationError) as exc:
         validate_prompt(bad, default_registry())
     assert any("output 5" in m for m in exc.value.node_errors["2"])
+
+
+def test_is_changed_nodes_never_cached_across_runs():
+    from comfyui_distributed_amd.graph.executor import Executor, NodeRegistry
+
+    calls = {"n": 0}
+
+    class Volatile:
+        RETURN_TYPES = ("INT",)
+        FUNCTION = "run"
+
+        @classmethod
+        def IS_CHANGED(cls, **kw):
+            return float("nan")
+
+        def run(self):
+            calls["n"] += 1
+            return (calls["n"],)
+
+    class Sink:
+        RETURN_TYPES = ()
+        OUTPUT_NODE = True
+        FUNCTION = "run"
+
+        def run(self, x):
+            return ()
+
+    reg = NodeRegistry()
+    reg.register("Volatile", Volatile)
+    reg.register("Sink", Sink)
+    ex = Executor(registry=reg)
+    prompt = {"1": {"class_type": "Volatile", "inputs": {}},
+              "2": {"class_type": "Sink", "inputs": {"x": ["1", 0]}}}
+    ex.execute(prompt)
+    ex.execute(prompt)
+    assert calls["n"] == 2  # re-executed despite identical fingerprint
+
+
+def test_auto_populate_creates_workers_for_fake_gpus(tmp_config, monkeypatch):
+    import asyncio
+
+    import torch
+    from aiohttp.test_utils import TestClient, TestServer
+
+    from comfyui_distributed_amd.server.app import DistributedServer
+
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(torch.cuda, "device_count", lambda: 4)
+
+    async def go():
+        srv = DistributedServer()
+        cl = TestClient(TestServer(srv.build_app()))
+        await cl.start_server()
+        try:
+            r = await cl.post("/distributed/auto_populate_workers", json={})
+            body = await r.json()
+            assert body["gpu_count"] == 4
+            # master sits on GPU 0 -> workers for GPUs 1..3
+            assert body["created"] == ["auto_gpu1", "auto_gpu2", "auto_gpu3"]
+            cfg = await (await cl.get("/distributed/config")).json()
+            devs = sorted(w["cuda_device"] for w in cfg["workers"])
+            assert devs == [1, 2, 3]
+            ports = sorted(w["port"] for w in cfg["workers"])
+            assert ports == [cfg["master"]["port"] + 2,
+                             cfg["master"]["port"] + 3,
+                             cfg["master"]["port"] + 4]
+        finally:
+            await cl.close()
+
+    asyncio.run(go())
