@@ -13,8 +13,10 @@ import math
 
 import torch
 
-SAMPLERS = ("euler", "euler_ancestral", "dpmpp_2m")
-SCHEDULERS = ("normal", "karras")
+SAMPLERS = ("euler", "euler_ancestral", "heun", "dpm_2", "dpmpp_2m",
+            "dpmpp_2m_sde")
+SCHEDULERS = ("normal", "karras", "exponential", "sgm_uniform", "simple",
+              "beta")
 
 
 def make_alphas_cumprod(n: int = 1000, beta_start: float = 0.00085,
@@ -60,14 +62,42 @@ class NoiseSchedule:
         if denoise <= 0:
             return torch.zeros(1)
         total = steps if denoise >= 1.0 else max(int(math.ceil(steps / denoise)), steps)
+        n_train = len(self.sigmas_all)
         if scheduler == "karras":
             rho = 7.0
             ramp = torch.linspace(0, 1, total)
             mn, mx = self.sigma_min, self.sigma_max
             s = (mx ** (1 / rho) + ramp * (mn ** (1 / rho) - mx ** (1 / rho))) ** rho
         elif scheduler == "normal":
-            t = torch.linspace(len(self.sigmas_all) - 1, 0, total)
-            s = self.sigmas_all[t.long().clamp(0, len(self.sigmas_all) - 1)]
+            t = torch.linspace(n_train - 1, 0, total)
+            s = self.sigmas_all[t.long().clamp(0, n_train - 1)]
+        elif scheduler == "exponential":
+            s = torch.exp(torch.linspace(
+                math.log(self.sigma_max), math.log(self.sigma_min), total))
+        elif scheduler == "sgm_uniform":
+            # uniform in timestep, excluding the final training step (the
+            # SGM/EDM convention ComfyUI exposes under this name)
+            t = torch.linspace(n_train - 1, 0, total + 1)[:-1]
+            s = self.sigmas_all[t.long().clamp(0, n_train - 1)]
+        elif scheduler == "simple":
+            # fixed stride over the training schedule
+            stride = n_train / total
+            idx = [n_train - 1 - int(i * stride) for i in range(total)]
+            s = self.sigmas_all[torch.tensor(idx).clamp(0, n_train - 1)]
+        elif scheduler == "beta":
+            # Beta(0.6, 0.6)-spaced timesteps (ComfyUI's "beta" scheduler):
+            # clusters steps at both schedule ends
+            try:
+                from scipy.stats import beta as _beta
+
+                ts = [round(_beta.ppf(1 - (i / total), 0.6, 0.6)
+                            * (n_train - 1)) for i in range(total)]
+            except Exception:  # scipy-less fallback: symmetric cosine
+                u = torch.linspace(1, 0, total)
+                ts = ((0.5 - 0.5 * torch.cos(u * math.pi))
+                      * (n_train - 1)).round().tolist()
+            s = self.sigmas_all[torch.tensor(
+                [int(t) for t in ts]).clamp(0, n_train - 1)]
         else:
             raise ValueError(f"unknown scheduler {scheduler!r}")
         s = torch.cat([s, torch.zeros(1)])
@@ -208,8 +238,14 @@ def sample(denoiser, noise_or_latent: torch.Tensor, sigmas: torch.Tensor,
         return _sample_euler(denoiser, x, sigmas)
     if sampler == "euler_ancestral":
         return _sample_euler_ancestral(denoiser, x, sigmas, gen)
+    if sampler == "heun":
+        return _sample_heun(denoiser, x, sigmas)
+    if sampler == "dpm_2":
+        return _sample_dpm_2(denoiser, x, sigmas)
     if sampler == "dpmpp_2m":
         return _sample_dpmpp_2m(denoiser, x, sigmas)
+    if sampler == "dpmpp_2m_sde":
+        return _sample_dpmpp_2m_sde(denoiser, x, sigmas, gen)
     raise ValueError(f"unknown sampler {sampler!r}")
 
 
@@ -235,6 +271,70 @@ def _sample_euler_ancestral(denoiser, x, sigmas, gen):
         x = x + d * (sigma_down - sigma)
         noise = torch.randn(x.shape, generator=gen, dtype=torch.float32).to(x.device)
         x = x + noise * sigma_up
+    return x
+
+
+def _sample_heun(denoiser, x, sigmas):
+    """Heun's 2nd-order method: Euler predictor + trapezoidal corrector
+    (2 model calls per step except the final sigma-0 step)."""
+    for i in range(len(sigmas) - 1):
+        sigma, sigma_next = sigmas[i], sigmas[i + 1]
+        denoised = denoiser(x, sigma)
+        d = (x - denoised) / sigma
+        dt = sigma_next - sigma
+        if sigma_next == 0:
+            x = x + d * dt
+            continue
+        x2 = x + d * dt
+        denoised2 = denoiser(x2, sigma_next)
+        d2 = (x2 - denoised2) / sigma_next
+        x = x + (d + d2) / 2 * dt
+    return x
+
+
+def _sample_dpm_2(denoiser, x, sigmas):
+    """DPM-Solver-2: midpoint (log-sigma geometric mean) 2nd-order step."""
+    for i in range(len(sigmas) - 1):
+        sigma, sigma_next = sigmas[i], sigmas[i + 1]
+        denoised = denoiser(x, sigma)
+        d = (x - denoised) / sigma
+        if sigma_next == 0:
+            x = x + d * (sigma_next - sigma)
+            continue
+        sigma_mid = sigma.log().lerp(sigma_next.log(), 0.5).exp()
+        x2 = x + d * (sigma_mid - sigma)
+        denoised2 = denoiser(x2, sigma_mid)
+        d2 = (x2 - denoised2) / sigma_mid
+        x = x + d2 * (sigma_next - sigma)
+    return x
+
+
+def _sample_dpmpp_2m_sde(denoiser, x, sigmas, gen, eta: float = 1.0):
+    """DPM++ 2M SDE (midpoint noise addition), seeded like the ancestral
+    sampler so distributed results stay deterministic."""
+    old_denoised = None
+    h_last = None
+    for i in range(len(sigmas) - 1):
+        sigma, sigma_next = sigmas[i], sigmas[i + 1]
+        denoised = denoiser(x, sigma)
+        if sigma_next == 0:
+            x = denoised
+        else:
+            t, s = -sigma.log(), -sigma_next.log()
+            h = s - t
+            eta_h = eta * h
+            x = (sigma_next / sigma) * (-eta_h).exp() * x + \
+                (-h - eta_h).expm1().neg() * denoised
+            if old_denoised is not None:
+                r = h_last / h
+                x = x + ((-h - eta_h).expm1().neg() / (-h - eta_h) + 1) * \
+                    (1 / r) * (denoised - old_denoised)
+            if eta > 0:
+                noise = torch.randn(x.shape, generator=gen,
+                                    dtype=torch.float32).to(x.device)
+                x = x + noise * sigma_next * (-2 * eta_h).expm1().neg().sqrt()
+            h_last = h
+        old_denoised = denoised
     return x
 
 
