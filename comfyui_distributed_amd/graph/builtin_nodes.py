@@ -147,12 +147,17 @@ class KSampler(_ContextNode):
         )
         denoiser = CFGDenoiser(stack.unet, stack.schedule, positive, negative,
                                float(cfg))
+        mask = latent_image.get("noise_mask")
         start = None
-        if denoise < 1.0 and latent_image["samples"].abs().sum() > 0:
+        if mask is not None or (
+            denoise < 1.0 and latent_image["samples"].abs().sum() > 0
+        ):
             start = latent_image["samples"].to(stack.device).float()
         with torch.no_grad():
             out = sample(denoiser, noise, sigmas, sampler=sampler_name,
-                         seed=int(seed), start_from_latent=start)
+                         seed=int(seed), start_from_latent=start,
+                         denoise_mask=None if mask is None
+                         else mask.to(stack.device))
         return ({"samples": out.cpu()},)
 
 
@@ -253,6 +258,50 @@ class FluxGenerate(_ContextNode):
                       width=int(width), height=int(height),
                       batch_size=int(batch_size))
         return (generate_images(model, positive, negative, p).cpu(),)
+
+
+class ImageToMask(_ContextNode):
+    """One channel of an IMAGE as a MASK [B,H,W] (ComfyUI parity node)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {
+            "image": ("IMAGE",),
+            "channel": (["red", "green", "blue"],),
+        }}
+
+    RETURN_TYPES = ("MASK",)
+    FUNCTION = "convert"
+    CATEGORY = "mask"
+
+    def convert(self, image, channel="red"):
+        idx = {"red": 0, "green": 1, "blue": 2}.get(str(channel), 0)
+        return (image[..., idx].clone(),)
+
+
+class SetLatentNoiseMask(_ContextNode):
+    """Attach an inpainting mask to a LATENT: 1 = re-generate, 0 = keep.
+    The pixel-space MASK is resampled to latent resolution here (ComfyUI
+    SetLatentNoiseMask parity); KSampler pins the unmasked region."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"samples": ("LATENT",), "mask": ("MASK",)}}
+
+    RETURN_TYPES = ("LATENT",)
+    FUNCTION = "set_mask"
+    CATEGORY = "latent"
+
+    def set_mask(self, samples, mask):
+        import torch.nn.functional as F
+
+        z = samples["samples"]
+        m = mask[:, None].float()  # [B,1,H,W]
+        m = F.interpolate(m, size=z.shape[-2:], mode="bilinear",
+                          align_corners=False)
+        out = dict(samples)
+        out["noise_mask"] = m.clamp(0, 1)
+        return (out,)
 
 
 class SyntheticAudio(_ContextNode):
@@ -364,4 +413,6 @@ BUILTIN_CLASS_MAPPINGS = {
     "WanVideoGenerate": WanVideoGenerate,
     "FluxGenerate": FluxGenerate,
     "SyntheticAudio": SyntheticAudio,
+    "ImageToMask": ImageToMask,
+    "SetLatentNoiseMask": SetLatentNoiseMask,
 }

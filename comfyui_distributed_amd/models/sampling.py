@@ -222,31 +222,62 @@ class CFGDenoiser:
         return x - eps.float() * sig.reshape(-1, 1, 1, 1)
 
 
+class _MaskedDenoiser:
+    """Inpainting wrapper: outside the mask the latent is pinned to the
+    original re-noised at the current sigma before every model call
+    (ComfyUI's latent-preservation behavior for LATENT noise_mask)."""
+
+    def __init__(self, inner, orig, noise, mask):
+        self.inner = inner
+        self.orig = orig
+        self.noise = noise
+        self.mask = mask
+
+    def __call__(self, x, sigma):
+        keep = self.orig + self.noise * sigma
+        x = x * self.mask + keep * (1 - self.mask)
+        return self.inner(x, sigma)
+
+
 def sample(denoiser, noise_or_latent: torch.Tensor, sigmas: torch.Tensor,
            sampler: str = "euler", seed: int | None = None,
-           start_from_latent: torch.Tensor | None = None) -> torch.Tensor:
+           start_from_latent: torch.Tensor | None = None,
+           denoise_mask: torch.Tensor | None = None) -> torch.Tensor:
     """Run the sampler loop. ``noise_or_latent`` is pure noise for txt2img;
     for img2img pass ``start_from_latent`` and the noised start is formed
-    here as latent + noise * sigmas[0]."""
+    here as latent + noise * sigmas[0]. ``denoise_mask`` ([B or 1, 1, h, w],
+    1 = denoise, 0 = keep) enables inpainting: the unmasked region of
+    ``start_from_latent`` is preserved exactly."""
     x = noise_or_latent.float() * sigmas[0]
     if start_from_latent is not None:
         x = start_from_latent.float() + noise_or_latent.float() * sigmas[0]
+    if denoise_mask is not None:
+        if start_from_latent is None:
+            raise ValueError("denoise_mask needs start_from_latent")
+        m = denoise_mask.to(x.device, torch.float32)
+        denoiser = _MaskedDenoiser(denoiser, start_from_latent.float(),
+                                   noise_or_latent.float(), m)
     gen = None
     if seed is not None:
         gen = torch.Generator(device="cpu").manual_seed(seed)
     if sampler == "euler":
-        return _sample_euler(denoiser, x, sigmas)
-    if sampler == "euler_ancestral":
-        return _sample_euler_ancestral(denoiser, x, sigmas, gen)
-    if sampler == "heun":
-        return _sample_heun(denoiser, x, sigmas)
-    if sampler == "dpm_2":
-        return _sample_dpm_2(denoiser, x, sigmas)
-    if sampler == "dpmpp_2m":
-        return _sample_dpmpp_2m(denoiser, x, sigmas)
-    if sampler == "dpmpp_2m_sde":
-        return _sample_dpmpp_2m_sde(denoiser, x, sigmas, gen)
-    raise ValueError(f"unknown sampler {sampler!r}")
+        out = _sample_euler(denoiser, x, sigmas)
+    elif sampler == "euler_ancestral":
+        out = _sample_euler_ancestral(denoiser, x, sigmas, gen)
+    elif sampler == "heun":
+        out = _sample_heun(denoiser, x, sigmas)
+    elif sampler == "dpm_2":
+        out = _sample_dpm_2(denoiser, x, sigmas)
+    elif sampler == "dpmpp_2m":
+        out = _sample_dpmpp_2m(denoiser, x, sigmas)
+    elif sampler == "dpmpp_2m_sde":
+        out = _sample_dpmpp_2m_sde(denoiser, x, sigmas, gen)
+    else:
+        raise ValueError(f"unknown sampler {sampler!r}")
+    if denoise_mask is not None:
+        m = denoise_mask.to(out.device, torch.float32)
+        out = out * m + start_from_latent.float() * (1 - m)
+    return out
 
 
 def _sample_euler(denoiser, x, sigmas):
