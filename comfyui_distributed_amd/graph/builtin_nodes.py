@@ -260,6 +260,65 @@ class FluxGenerate(_ContextNode):
         return (generate_images(model, positive, negative, p).cpu(),)
 
 
+class LatentUpscale(_ContextNode):
+    """Resize a LATENT (hires-fix style). Widths/heights are pixel-space
+    and divided by the VAE downscale (8), matching ComfyUI semantics."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {
+            "samples": ("LATENT",),
+            "upscale_method": (["nearest-exact", "bilinear", "bicubic"],),
+            "width": ("INT", {"default": 1024}),
+            "height": ("INT", {"default": 1024}),
+        }}
+
+    RETURN_TYPES = ("LATENT",)
+    FUNCTION = "upscale"
+    CATEGORY = "latent"
+
+    def upscale(self, samples, upscale_method="bilinear", width=1024,
+                height=1024):
+        import torch.nn.functional as F
+
+        mode = {"nearest-exact": "nearest-exact", "bilinear": "bilinear",
+                "bicubic": "bicubic"}[str(upscale_method)]
+        z = samples["samples"].float()
+        out = F.interpolate(
+            z, size=(int(height) // 8, int(width) // 8), mode=mode,
+            align_corners=False if mode in ("bilinear", "bicubic") else None,
+        )
+        res = dict(samples)
+        res["samples"] = out
+        res.pop("noise_mask", None)  # stale resolution
+        return (res,)
+
+
+class ImageScale(_ContextNode):
+    """Pixel-space resize on the fused Lanczos-3 resample kernel (the same
+    HIP kernel the tile engine uses for extract/resize)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {
+            "image": ("IMAGE",),
+            "width": ("INT", {"default": 1024}),
+            "height": ("INT", {"default": 1024}),
+        }}
+
+    RETURN_TYPES = ("IMAGE",)
+    FUNCTION = "scale"
+    CATEGORY = "image"
+
+    def scale(self, image, width=1024, height=1024):
+        from ..ops import dispatch as ops
+
+        b, h, w, _ = image.shape
+        out = ops.extract_resize(image.float(), (0, 0, w, h),
+                                 int(width), int(height))
+        return (out.clamp(0, 1),)
+
+
 class ImageToMask(_ContextNode):
     """One channel of an IMAGE as a MASK [B,H,W] (ComfyUI parity node)."""
 
@@ -414,5 +473,7 @@ BUILTIN_CLASS_MAPPINGS = {
     "FluxGenerate": FluxGenerate,
     "SyntheticAudio": SyntheticAudio,
     "ImageToMask": ImageToMask,
+    "LatentUpscale": LatentUpscale,
+    "ImageScale": ImageScale,
     "SetLatentNoiseMask": SetLatentNoiseMask,
 }

@@ -219,3 +219,22 @@ def test_inpainting_preserves_unmasked_region():
     # masked (right) half actually changed
     assert not torch.allclose(zl[:, :, :, zl.shape[-1] // 2:],
                               z[:, :, :, z.shape[-1] // 2:])
+
+
+def test_latent_upscale_and_image_scale_nodes():
+    from comfyui_distributed_amd.graph.builtin_nodes import (
+        ImageScale, LatentUpscale)
+
+    lat = {"samples": torch.randn(1, 4, 8, 8)}
+    out = LatentUpscale().upscale(lat, "bilinear", width=128, height=96)[0]
+    assert out["samples"].shape == (1, 4, 12, 16)
+    out2 = LatentUpscale().upscale(lat, "nearest-exact", 128, 128)[0]
+    assert out2["samples"].shape == (1, 4, 16, 16)
+
+    img = torch.rand(2, 16, 24, 3)
+    up = ImageScale().scale(img, width=48, height=32)[0]
+    assert up.shape == (2, 32, 48, 3)
+    assert torch.isfinite(up).all() and (up >= 0).all() and (up <= 1).all()
+    # identity resize returns (numerically) the same image
+    same = ImageScale().scale(img, width=24, height=16)[0]
+    assert torch.allclose(same, img, atol=1e-4)
