@@ -24,6 +24,8 @@ def load_wf(name):
     "distributed_flux_txt2img.json",
     "parameter_sweep.json",
     "distributed_audio_collect.json",
+    "inpaint.json",
+    "hires_fix.json",
 ])
 def test_workflows_validate(name):
     validate_prompt(load_wf(name), default_registry())
