@@ -46,8 +46,11 @@ class FluxConfig:
 
 
 FLUX12B = FluxConfig()
-FLUX_TINY = FluxConfig(dim=64, depth_double=1, depth_single=1, heads=2,
-                       context_dim=32, vec_dim=16, axes_dim=(8, 12, 12),
+# head_dim 64 (dim/heads), matching the attention kernel's validated
+# D range (40..160) — a smaller head dim would leave the GPU path
+# outside its tested contract
+FLUX_TINY = FluxConfig(dim=128, depth_double=1, depth_single=1, heads=2,
+                       context_dim=32, vec_dim=16, axes_dim=(16, 24, 24),
                        guidance_embed=False)
 
 
