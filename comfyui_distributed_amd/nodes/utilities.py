@@ -136,8 +136,10 @@ class DistributedValue:
             return (default,)
         try:
             idx = parse_worker_index(worker_id)
-            raw = values.get(str(idx + 1), "")  # 1-indexed worker keys
-            if raw:
+            raw = values.get(str(idx + 1))  # 1-indexed worker keys
+            # falsy values (0, 0.0) are legitimate overrides — only a
+            # missing key or empty string falls back to the default
+            if raw is not None and raw != "":
                 return (self._coerce(raw, value_type),)
         except (ValueError, IndexError):
             pass

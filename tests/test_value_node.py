@@ -55,3 +55,20 @@ def test_coercion_failure_keeps_default_value():
 
 def test_combo_stays_string():
     assert run("euler", {"_type": "COMBO", "1": "dpmpp_2m"}, True, "worker_0") == "dpmpp_2m"
+
+
+def test_zero_is_a_legitimate_worker_override():
+    import json
+
+    from comfyui_distributed_amd.nodes.utilities import DistributedValue
+
+    node = DistributedValue()
+    vals = json.dumps({"_type": "FLOAT", "1": 0.0, "2": 5.0})
+    out = node.distribute("3.5", vals, is_worker=True, worker_id="worker_0")
+    assert out == (0.0,)  # not the 3.5 default
+    out = node.distribute("3.5", vals, is_worker=True, worker_id="worker_1")
+    assert out == (5.0,)
+    # empty string still falls back
+    vals2 = json.dumps({"_type": "FLOAT", "1": ""})
+    out = node.distribute("3.5", vals2, is_worker=True, worker_id="worker_0")
+    assert out == (3.5,)
