@@ -101,6 +101,7 @@ class DistributedServer:
         r.add_post("/upload/image", self.post_upload_image)
         r.add_get("/view", self.get_view)
         r.add_get("/history", self.get_history)
+        r.add_get("/distributed/workflow_examples", self.get_workflow_examples)
         r.add_get("/history/{prompt_id}", self.get_history)
         r.add_get("/distributed/network_info", self.get_network_info)
         r.add_get("/distributed/system_info", self.get_system_info)
@@ -518,6 +519,28 @@ class DistributedServer:
                         fh.write(chunk)
                 saved.append(fname)
         return web.json_response({"saved": saved})
+
+    async def get_workflow_examples(self, request):
+        """List the shipped example workflows (repo ``workflows/`` dir or
+        ``DISTGPU_WORKFLOWS_DIR``); ``?name=`` returns one prompt JSON."""
+        import json as _json
+        from pathlib import Path
+
+        wf_dir = Path(os.environ.get("DISTGPU_WORKFLOWS_DIR", "workflows"))
+        name = request.query.get("name")
+        if name:
+            path = wf_dir / os.path.basename(str(name))
+            if not path.is_file() or path.suffix != ".json":
+                return _err("not found", status=404)
+            try:
+                data = _json.loads(path.read_text())
+            except ValueError:
+                return _err("invalid workflow JSON", status=500)
+            data.pop("_comment", None)
+            return web.json_response({"name": path.name, "prompt": data})
+        names = sorted(p.name for p in wf_dir.glob("*.json")) \
+            if wf_dir.is_dir() else []
+        return web.json_response({"workflows": names})
 
     async def get_history(self, request):
         pid = request.match_info.get("prompt_id")

@@ -425,3 +425,23 @@ def test_history_records_prompt_outcomes(client, tmp_path):
         assert pid in full and pid2 in full
 
     run(loop, go())
+
+
+def test_workflow_examples_endpoint(client, tmp_path, monkeypatch):
+    srv, cl, loop = client
+    (tmp_path / "demo.json").write_text(
+        '{"_comment": "x", "1": {"class_type": "DistributedSeed",'
+        ' "inputs": {"seed": 1}}}')
+    monkeypatch.setenv("DISTGPU_WORKFLOWS_DIR", str(tmp_path))
+
+    async def go():
+        r = await cl.get("/distributed/workflow_examples")
+        assert (await r.json())["workflows"] == ["demo.json"]
+        r = await cl.get("/distributed/workflow_examples?name=demo.json")
+        body = await r.json()
+        assert "_comment" not in body["prompt"]
+        assert body["prompt"]["1"]["class_type"] == "DistributedSeed"
+        r = await cl.get("/distributed/workflow_examples?name=../etc/passwd")
+        assert r.status == 404
+
+    run(loop, go())
