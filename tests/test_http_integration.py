@@ -167,3 +167,72 @@ def test_three_concurrent_collector_jobs(tmp_config, monkeypatch):
     assert got == [8, 12, 16]
     for p in previews:
         assert p.shape[0] == 2 and torch.isfinite(p).all()
+
+
+@pytest.mark.timeout(120)
+def test_simultaneous_queue_requests(tmp_config, monkeypatch):
+    """Two /distributed/queue POSTs in flight at once: orchestration
+    (probe, job-id minting, queue pre-create) must interleave safely."""
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.2)
+
+    async def go():
+        worker_srv = DistributedServer(is_worker=True)
+        wc = TestClient(TestServer(worker_srv.build_app()))
+        await wc.start_server()
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "w", "host": "127.0.0.1",
+            "port": wc.server.port, "cuda_device": 0, "enabled": True,
+            "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+
+        def prompt_for(seed):
+            return {
+                "1": {"class_type": "DistributedSeed", "inputs": {"seed": seed}},
+                "2": {"class_type": "LoadImage",
+                      "inputs": {"image": "synthetic:8x8"}},
+                "3": {"class_type": "DistributedCollector",
+                      "inputs": {"images": ["2", 0], "load_balance": False}},
+                "4": {"class_type": "PreviewImage", "inputs": {"images": ["3", 0]}},
+            }
+
+        r1, r2 = await asyncio.gather(
+            mc.post("/distributed/queue", json={
+                "prompt": prompt_for(1), "client_id": "a",
+                "enabled_worker_ids": ["w1"]}),
+            mc.post("/distributed/queue", json={
+                "prompt": prompt_for(2), "client_id": "b",
+                "enabled_worker_ids": ["w1"]}),
+        )
+        assert r1.status == 200 and r2.status == 200
+        j1, j2 = await r1.json(), await r2.json()
+        # job ids unique across concurrent requests
+        ids1, ids2 = set(j1["job_ids"].values()), set(j2["job_ids"].values())
+        assert ids1 and ids2 and not (ids1 & ids2)
+
+        for _ in range(400):
+            if len(previews) >= 2:
+                break
+            await asyncio.sleep(0.2)
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await wc.close()
+        await mc.close()
+        return previews
+
+    previews = asyncio.run(go())
+    assert len(previews) == 2
+    for p in previews:
+        assert p.shape == (2, 8, 8, 3)
