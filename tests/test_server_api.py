@@ -445,3 +445,25 @@ def test_workflow_examples_endpoint(client, tmp_path, monkeypatch):
         assert r.status == 404
 
     run(loop, go())
+
+
+def test_panel_javascript_syntax():
+    """Extract the panel's inline script and syntax-check it with node
+    (the reference ships a vitest suite for its JS; this is the
+    equivalent guard for the single-page panel)."""
+    import shutil
+    import subprocess
+    import tempfile
+    from pathlib import Path
+
+    if shutil.which("node") is None:
+        pytest.skip("node not available")
+    html = (Path(__file__).resolve().parent.parent /
+            "comfyui_distributed_amd/server/static/panel.html").read_text()
+    js = html.split("<script>")[1].split("</script>")[0]
+    with tempfile.NamedTemporaryFile("w", suffix=".js", delete=False) as fh:
+        fh.write(js)
+        path = fh.name
+    proc = subprocess.run(["node", "--check", path],
+                          capture_output=True, text=True, timeout=30)
+    assert proc.returncode == 0, proc.stderr
