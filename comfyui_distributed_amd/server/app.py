@@ -74,7 +74,19 @@ class DistributedServer:
     # ------------------------------------------------------------------ app
 
     def build_app(self) -> web.Application:
-        app = web.Application(client_max_size=constants.MAX_PAYLOAD_SIZE + 2**20)
+        import json as _json
+
+        @web.middleware
+        async def bad_json_middleware(request, handler):
+            """Malformed JSON bodies are a client error, not a 500
+            (reference utils/network.py handle_api_error envelope)."""
+            try:
+                return await handler(request)
+            except (_json.JSONDecodeError, UnicodeDecodeError):
+                return _err("invalid JSON body")
+
+        app = web.Application(client_max_size=constants.MAX_PAYLOAD_SIZE + 2**20,
+                              middlewares=[bad_json_middleware])
         r = app.router
         r.add_get("/", self.get_panel)
         r.add_get("/panel", self.get_panel)

@@ -467,3 +467,27 @@ def test_panel_javascript_syntax():
     proc = subprocess.run(["node", "--check", path],
                           capture_output=True, text=True, timeout=30)
     assert proc.returncode == 0, proc.stderr
+
+
+def test_malformed_json_returns_400_everywhere(client):
+    srv, cl, loop = client
+    posts = ["/prompt", "/distributed/queue", "/distributed/job_complete",
+             "/distributed/heartbeat", "/distributed/submit_tiles",
+             "/distributed/request_image", "/distributed/job_status",
+             "/distributed/config/update_worker", "/distributed/check_file",
+             "/distributed/worker/clear_launching"]
+
+    async def go():
+        for p in posts:
+            r = await cl.post(p, data=b"{not json",
+                              headers={"Content-Type": "application/json"})
+            assert r.status == 400, (p, r.status)
+        # /interrupt treats the body as optional: still succeeds
+        r = await cl.post("/interrupt", data=b"{not json",
+                          headers={"Content-Type": "application/json"})
+        assert r.status == 200
+        from comfyui_distributed_amd.nodes.runtime import get_runtime
+
+        get_runtime().clear_interrupt()
+
+    run(loop, go())
