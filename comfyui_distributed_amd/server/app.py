@@ -39,6 +39,13 @@ from .orchestration import orchestrate_distributed_execution
 from .queue_request import QueueRequestError, parse_queue_request_payload
 
 
+def _int_query(request, name: str, default: int, lo: int, hi: int) -> int:
+    try:
+        return min(max(int(request.query.get(name, default)), lo), hi)
+    except (TypeError, ValueError):
+        return default
+
+
 def _err(message: str, status: int = 400):
     return web.json_response({"error": message}, status=status)
 
@@ -693,7 +700,7 @@ class DistributedServer:
         if not network.normalize_host(worker.get("host", "")):
             return _err(f"worker {wid} is local; use /distributed/worker_log",
                         status=400)
-        lines = min(max(int(request.query.get("lines", "300")), 1), 3000)
+        lines = _int_query(request, "lines", 300, 1, 3000)
         import aiohttp
 
         session = await network.get_client_session()
@@ -794,7 +801,7 @@ class DistributedServer:
         from .workers import worker_log_path
 
         wid = request.query.get("id", "")
-        lines = int(request.query.get("lines", "100"))
+        lines = _int_query(request, "lines", 100, 1, 10000)
         path = worker_log_path(wid)
         if not path.exists():
             return _err("no log for worker", status=404)
@@ -808,7 +815,7 @@ class DistributedServer:
     async def get_local_log(self, request):
         from ..utils.logging import LOG_BUFFER
 
-        lines = int(request.query.get("lines", "100"))
+        lines = _int_query(request, "lines", 100, 1, 10000)
         return web.json_response({"log": "\n".join(list(LOG_BUFFER)[-lines:])})
 
     async def post_load_image(self, request):
