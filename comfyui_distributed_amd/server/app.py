@@ -373,12 +373,16 @@ class DistributedServer:
         job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
         if job is None:
             return _err("unknown job", status=404)
-        await job.results.put({
-            "image_idx": int(data.get("image_idx", 0)),
-            "tensor": decode_png_base64(data["image"]),
-            "worker_id": str(data.get("worker_id", "")),
-            "is_last": bool(data.get("is_last", False)),
-        })
+        try:
+            item = {
+                "image_idx": int(data.get("image_idx", 0)),
+                "tensor": decode_png_base64(data["image"]),
+                "worker_id": str(data.get("worker_id", "")),
+                "is_last": bool(data.get("is_last", False)),
+            }
+        except (TypeError, ValueError, KeyError) as exc:
+            return _err(f"bad submit_image payload: {exc}")
+        await job.results.put(item)
         if data.get("is_last"):
             job.finished_workers.add(str(data.get("worker_id", "")))
         return web.json_response({"status": "ok"})

@@ -73,9 +73,12 @@ def decode_png_base64(data: str) -> torch.Tensor:
     reference api/job_routes.py:291-307)."""
     if not HAS_PIL:
         raise RuntimeError("PIL not available")
-    raw = base64.b64decode(data)
-    img = Image.open(io.BytesIO(raw))
-    img.load()
+    try:
+        raw = base64.b64decode(data)
+        img = Image.open(io.BytesIO(raw))
+        img.load()
+    except Exception as exc:  # PIL raises its own hierarchy -> one type
+        raise ValueError(f"invalid image payload: {exc}") from exc
     return pil_to_tensor(img)
 
 

@@ -491,3 +491,26 @@ def test_malformed_json_returns_400_everywhere(client):
         get_runtime().clear_interrupt()
 
     run(loop, go())
+
+
+def test_bad_media_payloads_are_client_errors(client):
+    srv, cl, loop = client
+
+    async def go():
+        # valid base64 that is not a PNG -> 400, not a PIL 500
+        r = await cl.post("/distributed/job_complete", json={
+            "job_id": "x", "worker_id": "w", "image": "QUJD",
+            "is_last": True})
+        assert r.status == 400
+        assert "invalid image payload" in (await r.text())
+        await cl.post("/distributed/prepare_job", json={"job_id": "si"})
+        # garbage image_idx on a live dynamic job -> 400
+        from comfyui_distributed_amd.nodes.runtime import get_runtime
+
+        await get_runtime().job_state.init_dynamic_job("dj", 1)
+        r = await cl.post("/distributed/submit_image", json={
+            "job_id": "dj", "worker_id": "w", "image_idx": "nope",
+            "image": "QUJD"})
+        assert r.status == 400
+
+    run(loop, go())
