@@ -327,7 +327,10 @@ class DistributedServer:
         job = await self.job_state.get_tile_job(str(data.get("job_id", "")))
         if job is None:
             return _err("unknown job", status=404)
-        items = decode_tile_submission(data)
+        try:
+            items = decode_tile_submission(data)
+        except (TypeError, ValueError, KeyError) as exc:
+            return _err(f"bad tile submission: {exc}")
         for item in items:
             await job.results.put(item)
         if data.get("is_last"):
