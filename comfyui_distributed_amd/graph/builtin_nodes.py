@@ -208,7 +208,8 @@ class WanVideoGenerate(_ContextNode):
                 "height": ("INT", {"default": 480}),
                 "frames": ("INT", {"default": 17}),
             },
-            "optional": {"negative": ("CONDITIONING",)},
+            "optional": {"negative": ("CONDITIONING",),
+                         "start_image": ("IMAGE",)},
         }
 
     RETURN_TYPES = ("IMAGE",)
@@ -216,13 +217,14 @@ class WanVideoGenerate(_ContextNode):
     CATEGORY = "video"
 
     def generate(self, model, positive, seed, steps, cfg, width, height,
-                 frames, negative=None):
+                 frames, negative=None, start_image=None):
         from ..models.video import VideoGenParams, generate_video
 
         p = VideoGenParams(seed=int(seed), steps=int(steps), cfg=float(cfg),
                            width=int(width), height=int(height),
                            frames=int(frames))
-        return (generate_video(model, positive, negative, p).cpu(),)
+        return (generate_video(model, positive, negative, p,
+                               start_image=start_image).cpu(),)
 
 
 class FluxGenerate(_ContextNode):

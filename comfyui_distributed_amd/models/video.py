@@ -172,13 +172,20 @@ class FlowCFGVelocity:
 
 def sample_flow(velocity_fn, noise: torch.Tensor, steps: int,
                 shift: float = 5.0, start_from_latent: torch.Tensor | None = None,
-                denoise: float = 1.0) -> torch.Tensor:
+                denoise: float = 1.0,
+                pin_latent: torch.Tensor | None = None,
+                pin_mask: torch.Tensor | None = None) -> torch.Tensor:
     """Euler integration of the rectified flow from noise (t=1) to data.
 
     img2img: with ``start_from_latent`` and ``denoise`` < 1 the schedule is
     truncated to [denoise, 0] and integration starts from the flow
     interpolant x_t0 = (1-t0)*x0 + t0*noise at the (shifted) strength t0 —
-    the flow-matching analogue of the SD img2img start used by USDU."""
+    the flow-matching analogue of the SD img2img start used by USDU.
+
+    pinning (i2v / flow inpainting): ``pin_mask`` (broadcastable, 1 = keep
+    pinned to ``pin_latent``, 0 = generate) resets the pinned region to
+    its flow interpolant (1-t)*pin + t*noise before every velocity call;
+    at t=0 it equals ``pin_latent`` exactly."""
     t_lin = torch.linspace(1.0, 0.0, steps + 1)
     if start_from_latent is not None and denoise < 1.0:
         t_lin = t_lin * denoise
@@ -188,25 +195,49 @@ def sample_flow(velocity_fn, noise: torch.Tensor, steps: int,
         x = (1 - t0) * start_from_latent.float() + t0 * noise.float()
     else:
         x = noise.float()
+
+    def pin(xc, t):
+        if pin_mask is None:
+            return xc
+        keep = (1 - t) * pin_latent.float() + t * noise.float()
+        return xc * (1 - pin_mask) + keep * pin_mask
+
     for i in range(steps):
         t, t_next = sig[i], sig[i + 1]
+        x = pin(x, t)
         v = velocity_fn(x, t)
         x = x + (t_next - t) * v.float()
-    return x
+    return pin(x, sig[-1])
 
 
-def generate_video(stack: WanStack, cond, uncond, p: VideoGenParams) -> torch.Tensor:
+def generate_video(stack: WanStack, cond, uncond, p: VideoGenParams,
+                   start_image: torch.Tensor | None = None) -> torch.Tensor:
     """Returns frames [T, H, W, 3] float32 in [0,1] (ComfyUI IMAGE batch
     convention for video: frames along the batch dim, so the reference's
-    ImageBatchDivider segments them directly)."""
+    ImageBatchDivider segments them directly).
+
+    i2v: with ``start_image`` ([1, H, W, 3]) the first latent frame is
+    pinned to the encoded image throughout the flow integration (temporal
+    inpainting — the real WAN i2v conditions on the first frame the same
+    way structurally, via a masked video latent)."""
     stack.validate_frames(p.frames)
     g = torch.Generator().manual_seed(p.seed)
     lat_t = stack.latent_frames(p.frames)  # 1 + (T-1)/4
     shape = (p.batch_size, stack.cfg.in_channels, lat_t, p.height // 8, p.width // 8)
     noise = torch.randn(shape, generator=g).to(stack.device)
     vel = FlowCFGVelocity(stack.model, cond, uncond, p.cfg)
+    pin_latent = pin_mask = None
+    if start_image is not None:
+        with torch.no_grad():
+            z0 = stack.vae.spatial.encode(
+                start_image.to(stack.device, stack.dtype))  # [1, Cz, h, w]
+        pin_latent = torch.zeros(shape, device=stack.device)
+        pin_latent[:, :, 0] = z0[0].float()
+        pin_mask = torch.zeros(1, 1, lat_t, 1, 1, device=stack.device)
+        pin_mask[:, :, 0] = 1.0
     with torch.no_grad():
-        lat = sample_flow(vel, noise, p.steps)
+        lat = sample_flow(vel, noise, p.steps, pin_latent=pin_latent,
+                          pin_mask=pin_mask)
         imgs = torch.cat([
             stack.vae.decode(lat[b].to(stack.dtype), frames=p.frames)
             for b in range(lat.shape[0])
