@@ -26,6 +26,7 @@ def load_wf(name):
     "distributed_audio_collect.json",
     "inpaint.json",
     "hires_fix.json",
+    "wan_image_to_video.json",
 ])
 def test_workflows_validate(name):
     validate_prompt(load_wf(name), default_registry())
