@@ -33,9 +33,13 @@ def generate_latents(stack, cond, uncond, p: GenParams) -> torch.Tensor:
         return sample(denoiser, noise, sigmas, sampler=p.sampler_name, seed=p.seed)
 
 
-def generate_latents_flux(stack, cond, uncond, p: GenParams) -> torch.Tensor:
+def generate_latents_flux(stack, cond, uncond, p: GenParams,
+                          init_latent: torch.Tensor | None = None,
+                          denoise: float = 1.0) -> torch.Tensor:
     """Rectified-flow sampling for the Flux family (velocity prediction,
-    Euler integration over the shifted time schedule)."""
+    Euler integration over the shifted time schedule). ``init_latent`` +
+    ``denoise`` < 1 does flow img2img (truncated schedule from the
+    interpolant, same math USDU uses per tile)."""
     from ..models.video import sample_flow
 
     b = p.batch_size
@@ -56,7 +60,8 @@ def generate_latents_flux(stack, cond, uncond, p: GenParams) -> torch.Tensor:
             v = vu + p.cfg * (v - vu)
         return v
 
-    return sample_flow(velocity, noise, p.steps, shift=stack.flow_shift)
+    return sample_flow(velocity, noise, p.steps, shift=stack.flow_shift,
+                       start_from_latent=init_latent, denoise=denoise)
 
 
 def generate_images(stack, cond, uncond, p: GenParams) -> torch.Tensor:

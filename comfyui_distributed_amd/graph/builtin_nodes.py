@@ -245,7 +245,10 @@ class FluxGenerate(_ContextNode):
                 "height": ("INT", {"default": 1024}),
                 "batch_size": ("INT", {"default": 1}),
             },
-            "optional": {"negative": ("CONDITIONING",)},
+            "optional": {"negative": ("CONDITIONING",),
+                         "image": ("IMAGE",),
+                         "denoise": ("FLOAT", {"default": 1.0, "min": 0.0,
+                                               "max": 1.0})},
         }
 
     RETURN_TYPES = ("IMAGE",)
@@ -253,13 +256,29 @@ class FluxGenerate(_ContextNode):
     CATEGORY = "sampling"
 
     def generate(self, model, positive, seed, steps, cfg, width, height,
-                 batch_size=1, negative=None):
-        from ..engine.generate import GenParams, generate_images
+                 batch_size=1, negative=None, image=None, denoise=1.0):
+        from ..engine.generate import GenParams, generate_latents_flux
 
+        stack = model
+        if image is not None:
+            h, w = image.shape[1], image.shape[2]
+        else:
+            h, w = int(height), int(width)
         p = GenParams(seed=int(seed), steps=int(steps), cfg=float(cfg),
-                      width=int(width), height=int(height),
-                      batch_size=int(batch_size))
-        return (generate_images(model, positive, negative, p).cpu(),)
+                      width=w, height=h, batch_size=int(batch_size))
+        init = None
+        if image is not None:
+            with torch.no_grad():
+                init = stack.vae.encode(
+                    image.to(stack.device, stack.dtype)).float()
+                if init.shape[0] != p.batch_size:
+                    init = init.expand(p.batch_size, -1, -1, -1)
+        with torch.no_grad():
+            lat = generate_latents_flux(stack, positive, negative, p,
+                                        init_latent=init,
+                                        denoise=float(denoise))
+            out = stack.vae.decode(lat.to(stack.dtype)).float()
+        return (out.cpu(),)
 
 
 class LatentUpscale(_ContextNode):

@@ -94,3 +94,23 @@ def test_flux_usdu_single_gpu_and_order_invariance():
     r2 = sample_tiles(stack, cond, None, p, canvas, plans, [0, 1])
     blend_results(canvas, {**r1, **r2}, plans, p)
     assert torch.allclose(canvas, out, atol=1e-5)
+
+
+def test_flux_img2img_node_path():
+    from comfyui_distributed_amd.graph.builtin_nodes import FluxGenerate
+
+    stack = create_diffusion_stack("flux_tiny")
+    cond = stack.make_conditioning(1)
+    src = torch.rand(1, 32, 32, 3, generator=torch.Generator().manual_seed(4))
+    out = FluxGenerate().generate(stack, cond, seed=2, steps=1, cfg=1.0,
+                                  width=64, height=64, image=src,
+                                  denoise=0.4)[0]
+    assert out.shape == (1, 32, 32, 3)  # follows the init image size
+    assert torch.isfinite(out).all()
+    # low denoise keeps the output near the init image reconstruction
+    with torch.no_grad():
+        recon = stack.vae.decode(stack.vae.encode(src))
+    out_hi = FluxGenerate().generate(stack, cond, seed=2, steps=1, cfg=1.0,
+                                     width=64, height=64, image=src,
+                                     denoise=1.0)[0]
+    assert (out - recon.float()).abs().mean() < (out_hi - recon.float()).abs().mean()
