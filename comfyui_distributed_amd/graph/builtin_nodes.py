@@ -450,14 +450,22 @@ class SaveImage(_ContextNode):
     CATEGORY = "image"
 
     def save(self, images, filename_prefix="out"):
+        import json
         from pathlib import Path
 
         out_dir = Path(self._ctx.get("output_dir", "output"))
         out_dir.mkdir(parents=True, exist_ok=True)
+        # embed the executing prompt as PNG metadata (ComfyUI convention:
+        # saved outputs carry their workflow for reproducibility)
+        meta = None
+        prompt = self._ctx.get("current_prompt")
+        if prompt is not None:
+            meta = {"prompt": json.dumps(prompt)}
         paths = []
         for i in range(images.shape[0]):
             p = out_dir / f"{filename_prefix}_{i:05d}.png"
-            p.write_bytes(encode_png_bytes(images[i : i + 1], compress_level=4))
+            p.write_bytes(encode_png_bytes(images[i : i + 1],
+                                           compress_level=4, metadata=meta))
             paths.append(str(p))
         sink = self._ctx.get("saved_images")
         if isinstance(sink, list):

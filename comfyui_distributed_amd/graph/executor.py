@@ -165,6 +165,7 @@ class Executor:
         and every executed node."""
         validate_prompt(prompt, self.registry)
         graph = PromptGraph(prompt)
+        self.context["current_prompt"] = prompt  # for output metadata
         fps = self._fingerprints(graph)
         next_run_cache: dict[str, tuple] = {}
         cache: dict[str, tuple] = {}

@@ -82,10 +82,21 @@ def decode_png_base64(data: str) -> torch.Tensor:
     return pil_to_tensor(img)
 
 
-def encode_png_bytes(t: torch.Tensor, compress_level: int = 0) -> bytes:
+def encode_png_bytes(t: torch.Tensor, compress_level: int = 0,
+                     metadata: dict | None = None) -> bytes:
+    """``metadata`` (str -> str) is embedded as PNG tEXt chunks — the
+    ComfyUI convention of storing the prompt in saved outputs."""
     img = tensor_to_pil(t)
     buf = io.BytesIO()
-    img.save(buf, format="PNG", compress_level=compress_level)
+    pnginfo = None
+    if metadata:
+        from PIL.PngImagePlugin import PngInfo
+
+        pnginfo = PngInfo()
+        for key, value in metadata.items():
+            pnginfo.add_text(str(key), str(value))
+    img.save(buf, format="PNG", compress_level=compress_level,
+             pnginfo=pnginfo)
     return buf.getvalue()
 
 

@@ -238,3 +238,27 @@ def test_latent_upscale_and_image_scale_nodes():
     # identity resize returns (numerically) the same image
     same = ImageScale().scale(img, width=24, height=16)[0]
     assert torch.allclose(same, img, atol=1e-4)
+
+
+def test_saved_png_embeds_prompt_metadata(tmp_path):
+    """ComfyUI convention: SaveImage outputs carry the workflow JSON in a
+    PNG tEXt chunk for reproducibility."""
+    import io
+    import json
+
+    from PIL import Image
+
+    from comfyui_distributed_amd.graph.executor import Executor
+
+    prompt = {
+        "1": {"class_type": "LoadImage", "inputs": {"image": "synthetic:8x8"}},
+        "2": {"class_type": "SaveImage",
+              "inputs": {"images": ["1", 0], "filename_prefix": "meta"}},
+    }
+    saved = []
+    ex = Executor(context={"output_dir": str(tmp_path),
+                           "saved_images": saved, "device": "cpu"})
+    ex.execute(prompt)
+    img = Image.open(saved[0])
+    embedded = json.loads(img.text["prompt"])
+    assert embedded == prompt
