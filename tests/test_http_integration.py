@@ -236,3 +236,64 @@ def test_simultaneous_queue_requests(tmp_config, monkeypatch):
     assert len(previews) == 2
     for p in previews:
         assert p.shape == (2, 8, 8, 3)
+
+
+@pytest.mark.timeout(120)
+def test_delegate_master_end_to_end(tmp_config, monkeypatch):
+    """delegate_master=true over real sockets: the worker generates, the
+    master executes only the post-collector subgraph with a placeholder
+    image and returns ONLY the worker's batch."""
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.2)
+
+    async def go():
+        worker_srv = DistributedServer(is_worker=True)
+        wc = TestClient(TestServer(worker_srv.build_app()))
+        await wc.start_server()
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "w", "host": "127.0.0.1",
+            "port": wc.server.port, "cuda_device": 0, "enabled": True,
+            "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+
+        prompt = {
+            "1": {"class_type": "DistributedSeed", "inputs": {"seed": 4}},
+            "2": {"class_type": "LoadImage",
+                  "inputs": {"image": "synthetic:8x8"}},
+            "3": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["2", 0], "load_balance": False}},
+            "4": {"class_type": "PreviewImage", "inputs": {"images": ["3", 0]}},
+        }
+        r = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "dm", "enabled_worker_ids": ["w1"],
+            "delegate_master": True})
+        assert r.status == 200
+        body = await r.json()
+        assert body["participants"] == ["w1"]  # master orchestrates only
+
+        for _ in range(300):
+            if previews:
+                break
+            await asyncio.sleep(0.2)
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await wc.close()
+        await mc.close()
+        return previews
+
+    previews = asyncio.run(go())
+    assert previews, "delegate job never completed"
+    # only the worker's single image — no master batch
+    assert previews[0].shape == (1, 8, 8, 3)
