@@ -138,3 +138,22 @@ def test_audio_envelope_roundtrip(b, c, n, sr):
     out = decode_audio_payload(encode_audio_payload(audio))
     assert out["sample_rate"] == sr
     assert torch.equal(out["waveform"], audio["waveform"])
+
+
+@given(st.integers(8, 256), st.integers(8, 256), st.data())
+@settings(max_examples=100, deadline=None)
+def test_expand_crop_reaches_target_when_possible(w, h, data):
+    from comfyui_distributed_amd.utils.usdu_math import expand_crop
+
+    x1 = data.draw(st.integers(0, w - 2))
+    y1 = data.draw(st.integers(0, h - 2))
+    x2 = data.draw(st.integers(x1 + 1, w))
+    y2 = data.draw(st.integers(y1 + 1, h))
+    tw = data.draw(st.integers(x2 - x1, w))
+    th = data.draw(st.integers(y2 - y1, h))
+    (rx1, ry1, rx2, ry2), _ = expand_crop((x1, y1, x2, y2), w, h, tw, th)
+    # stays in bounds and contains the original region
+    assert 0 <= rx1 <= x1 and x2 <= rx2 <= w
+    assert 0 <= ry1 <= y1 and y2 <= ry2 <= h
+    # target fits the canvas -> exact size achieved
+    assert rx2 - rx1 == tw and ry2 - ry1 == th
