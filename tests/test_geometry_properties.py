@@ -157,3 +157,31 @@ def test_expand_crop_reaches_target_when_possible(w, h, data):
     assert 0 <= ry1 <= y1 and y2 <= ry2 <= h
     # target fits the canvas -> exact size achieved
     assert rx2 - rx1 == tw and ry2 - ry1 == th
+
+
+@given(st.integers(1, 512), st.integers(1, 512), st.integers(1, 512),
+       st.integers(1, 512), st.data())
+@settings(max_examples=100, deadline=None)
+def test_resize_region_in_bounds_and_non_degenerate(iw, ih, rw, rh, data):
+    from comfyui_distributed_amd.utils.usdu_math import resize_region
+
+    x1 = data.draw(st.integers(0, iw - 1))
+    y1 = data.draw(st.integers(0, ih - 1))
+    x2 = data.draw(st.integers(x1 + 1, iw))
+    y2 = data.draw(st.integers(y1 + 1, ih))
+    nx1, ny1, nx2, ny2 = resize_region((x1, y1, x2, y2), (iw, ih), (rw, rh))
+    assert 0 <= nx1 < nx2 <= rw
+    assert 0 <= ny1 < ny2 <= rh
+    # floor/ceil mapping never shrinks a region to nothing and covers the
+    # scaled extent
+    assert nx1 <= x1 * rw / iw + 1e-9 and nx2 >= x2 * rw / iw - 1e-9
+
+
+@given(st.text(max_size=40), st.booleans(), st.integers(0, 12))
+@settings(max_examples=100, deadline=None)
+def test_distributed_value_never_crashes_on_fuzzed_json(raw, is_worker, idx):
+    from comfyui_distributed_amd.nodes.utilities import DistributedValue
+
+    out = DistributedValue().distribute(
+        "def", raw, is_worker=is_worker, worker_id=f"worker_{idx}")
+    assert isinstance(out, tuple) and len(out) == 1
