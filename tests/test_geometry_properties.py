@@ -185,3 +185,27 @@ def test_distributed_value_never_crashes_on_fuzzed_json(raw, is_worker, idx):
     out = DistributedValue().distribute(
         "def", raw, is_worker=is_worker, worker_id=f"worker_{idx}")
     assert isinstance(out, tuple) and len(out) == 1
+
+
+@given(st.dictionaries(
+    st.sampled_from(["prompt", "workflow", "client_id", "enabled_worker_ids",
+                     "workers", "delegate_master", "auto_prepare",
+                     "trace_execution_id", "junk"]),
+    st.one_of(st.none(), st.booleans(), st.integers(), st.text(max_size=10),
+              st.lists(st.text(max_size=5), max_size=3),
+              st.dictionaries(st.text(max_size=4),
+                              st.integers(), max_size=3)),
+    max_size=6))
+@settings(max_examples=150, deadline=None)
+def test_queue_request_parser_total(payload):
+    """parse_queue_request_payload is total: any JSON-shaped dict either
+    parses or raises QueueRequestError — never an uncontrolled exception."""
+    from comfyui_distributed_amd.server.queue_request import (
+        QueueRequestError, parse_queue_request_payload)
+
+    try:
+        out = parse_queue_request_payload(payload)
+        assert isinstance(out.prompt, dict)
+        assert isinstance(out.enabled_worker_ids, list)
+    except QueueRequestError:
+        pass
