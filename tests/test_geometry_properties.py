@@ -209,3 +209,28 @@ def test_queue_request_parser_total(payload):
         assert isinstance(out.enabled_worker_ids, list)
     except QueueRequestError:
         pass
+
+
+@given(st.dictionaries(
+    st.sampled_from(["job_id", "worker_id", "batch_idx", "image", "audio",
+                     "is_last", "tiles"]),
+    st.one_of(st.none(), st.booleans(), st.integers(-5, 5),
+              st.text(max_size=12),
+              st.lists(st.dictionaries(
+                  st.sampled_from(["tile_idx", "batch_idx", "image"]),
+                  st.one_of(st.integers(-2, 2), st.text(max_size=8)),
+                  max_size=3), max_size=2)),
+    max_size=6))
+@settings(max_examples=150, deadline=None)
+def test_wire_decoders_total(payload):
+    """The two wire decoders raise only (TypeError|ValueError|KeyError) on
+    junk — the route layer maps exactly those to 400s."""
+    from comfyui_distributed_amd.nodes.collector import (
+        decode_job_complete_envelope)
+    from comfyui_distributed_amd.server.usdu_http import decode_tile_submission
+
+    for fn in (decode_job_complete_envelope, decode_tile_submission):
+        try:
+            fn(dict(payload))
+        except (TypeError, ValueError, KeyError):
+            pass
