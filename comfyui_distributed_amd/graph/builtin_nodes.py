@@ -461,9 +461,19 @@ class SaveImage(_ContextNode):
         prompt = self._ctx.get("current_prompt")
         if prompt is not None:
             meta = {"prompt": json.dumps(prompt)}
+        # continue numbering after existing outputs instead of overwriting
+        # (ComfyUI's SaveImage counter behavior)
+        import re
+
+        pat = re.compile(re.escape(filename_prefix) + r"_(\d{5})\.png$")
+        start = -1
+        for existing in out_dir.glob(f"{filename_prefix}_*.png"):
+            m = pat.match(existing.name)
+            if m:
+                start = max(start, int(m.group(1)))
         paths = []
         for i in range(images.shape[0]):
-            p = out_dir / f"{filename_prefix}_{i:05d}.png"
+            p = out_dir / f"{filename_prefix}_{start + 1 + i:05d}.png"
             p.write_bytes(encode_png_bytes(images[i : i + 1],
                                            compress_level=4, metadata=meta))
             paths.append(str(p))

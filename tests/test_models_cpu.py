@@ -262,3 +262,16 @@ def test_saved_png_embeds_prompt_metadata(tmp_path):
     img = Image.open(saved[0])
     embedded = json.loads(img.text["prompt"])
     assert embedded == prompt
+
+
+def test_save_image_continues_numbering(tmp_path):
+    from comfyui_distributed_amd.graph.builtin_nodes import SaveImage
+
+    node = SaveImage()
+    node.set_context({"output_dir": str(tmp_path), "saved_images": []})
+    img = torch.rand(2, 4, 4, 3)
+    node.save(img, filename_prefix="seq")
+    node.save(img, filename_prefix="seq")  # must NOT overwrite
+    names = sorted(p.name for p in tmp_path.glob("seq_*.png"))
+    assert names == ["seq_00000.png", "seq_00001.png",
+                     "seq_00002.png", "seq_00003.png"]
