@@ -454,6 +454,13 @@ class SaveImage(_ContextNode):
         from pathlib import Path
 
         out_dir = Path(self._ctx.get("output_dir", "output"))
+        # ComfyUI allows a subfolder in the prefix ("run1/img"); keep it
+        # inside the output dir
+        prefix_path = Path(filename_prefix)
+        if prefix_path.is_absolute() or ".." in prefix_path.parts:
+            raise ValueError(f"bad filename_prefix {filename_prefix!r}")
+        out_dir = out_dir / prefix_path.parent
+        filename_prefix = prefix_path.name
         out_dir.mkdir(parents=True, exist_ok=True)
         # embed the executing prompt as PNG metadata (ComfyUI convention:
         # saved outputs carry their workflow for reproducibility)

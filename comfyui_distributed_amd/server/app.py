@@ -586,7 +586,10 @@ class DistributedServer:
             return _err("missing filename")
         kind = request.query.get("type", "output")
         key = "input_dir" if kind == "input" else "output_dir"
-        path = Path(self.executor.context.get(key, kind)) / name
+        sub = Path(str(request.query.get("subfolder", "")))
+        if sub.is_absolute() or ".." in sub.parts:
+            return _err("bad subfolder")
+        path = Path(self.executor.context.get(key, kind)) / sub / name
         if not path.is_file():
             return _err("not found", status=404)
         ctype = "image/png" if name.lower().endswith(".png") else \

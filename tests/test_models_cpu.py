@@ -275,3 +275,18 @@ def test_save_image_continues_numbering(tmp_path):
     names = sorted(p.name for p in tmp_path.glob("seq_*.png"))
     assert names == ["seq_00000.png", "seq_00001.png",
                      "seq_00002.png", "seq_00003.png"]
+
+
+def test_save_image_subfolder_prefix(tmp_path):
+    import pytest as _pytest
+
+    from comfyui_distributed_amd.graph.builtin_nodes import SaveImage
+
+    node = SaveImage()
+    node.set_context({"output_dir": str(tmp_path), "saved_images": []})
+    node.save(torch.rand(1, 4, 4, 3), filename_prefix="runA/img")
+    assert (tmp_path / "runA" / "img_00000.png").exists()
+    with _pytest.raises(ValueError):
+        node.save(torch.rand(1, 4, 4, 3), filename_prefix="../escape")
+    with _pytest.raises(ValueError):
+        node.save(torch.rand(1, 4, 4, 3), filename_prefix="/abs/path")

@@ -514,3 +514,21 @@ def test_bad_media_payloads_are_client_errors(client):
         assert r.status == 400
 
     run(loop, go())
+
+
+def test_view_subfolder(client, tmp_path):
+    srv, cl, loop = client
+
+    async def go():
+        from pathlib import Path
+
+        out = Path(srv.executor.context.setdefault("output_dir",
+                                                   str(tmp_path)))
+        (out / "runB").mkdir(parents=True, exist_ok=True)
+        (out / "runB" / "x.png").write_bytes(b"\x89PNGdata")
+        r = await cl.get("/view?filename=x.png&subfolder=runB")
+        assert r.status == 200
+        r = await cl.get("/view?filename=x.png&subfolder=../etc")
+        assert r.status == 400
+
+    run(loop, go())
