@@ -185,13 +185,25 @@ class DistributedServer:
                        "completed": status == "success",
                        "messages": [error] if error else []},
             "outputs": {"images": [
-                {"filename": os.path.basename(str(f)), "type": "output"}
-                for f in new_files
+                self._output_entry(f) for f in new_files
             ]},
         }
         self.history[str(prompt_id)] = entry
         while len(self.history) > self.HISTORY_LIMIT:
             self.history.pop(next(iter(self.history)))
+
+    def _output_entry(self, path) -> dict:
+        """filename + subfolder relative to the output dir (the pair /view
+        takes), mirroring ComfyUI's history format."""
+        out_dir = os.path.abspath(
+            str(self.executor.context.get("output_dir", "output")))
+        ap = os.path.abspath(str(path))
+        sub = ""
+        if ap.startswith(out_dir + os.sep):
+            rel = os.path.relpath(ap, out_dir)
+            sub = os.path.dirname(rel)
+        return {"filename": os.path.basename(ap), "subfolder": sub,
+                "type": "output"}
 
     async def enqueue_local(self, prompt: dict, client_id: str) -> str:
         validate_prompt(prompt, self.registry)
