@@ -340,6 +340,48 @@ class ImageScale(_ContextNode):
         return (out.clamp(0, 1),)
 
 
+class SaveAudio(_ContextNode):
+    """Persist an AUDIO dict as 16-bit PCM WAV (stdlib wave; the collector's
+    gathered audio becomes a file the way SaveImage persists images)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"audio": ("AUDIO",),
+                             "filename_prefix": ("STRING",
+                                                 {"default": "audio"})}}
+
+    RETURN_TYPES = ()
+    FUNCTION = "save"
+    OUTPUT_NODE = True
+    CATEGORY = "audio"
+
+    def save(self, audio, filename_prefix="audio"):
+        import wave
+        from pathlib import Path
+
+        out_dir = Path(self._ctx.get("output_dir", "output"))
+        out_dir.mkdir(parents=True, exist_ok=True)
+        wf = audio["waveform"]  # [B, C, N]
+        sr = int(audio["sample_rate"])
+        paths = []
+        for b in range(wf.shape[0]):
+            clip = wf[b].clamp(-1, 1)
+            pcm = (clip * 32767.0).to(torch.int16)
+            # interleave channels: [C, N] -> [N, C]
+            inter = pcm.transpose(0, 1).contiguous()
+            p = out_dir / f"{filename_prefix}_{b:05d}.wav"
+            with wave.open(str(p), "wb") as w:
+                w.setnchannels(int(wf.shape[1]))
+                w.setsampwidth(2)
+                w.setframerate(sr)
+                w.writeframes(inter.numpy().tobytes())
+            paths.append(str(p))
+        sink = self._ctx.get("saved_images")
+        if isinstance(sink, list):
+            sink.extend(paths)
+        return ()
+
+
 class ImageToMask(_ContextNode):
     """One channel of an IMAGE as a MASK [B,H,W] (ComfyUI parity node)."""
 
@@ -522,4 +564,5 @@ BUILTIN_CLASS_MAPPINGS = {
     "LatentUpscale": LatentUpscale,
     "ImageScale": ImageScale,
     "SetLatentNoiseMask": SetLatentNoiseMask,
+    "SaveAudio": SaveAudio,
 }

@@ -59,3 +59,24 @@ def test_audio_concat_along_samples():
     assert out["waveform"].shape == (1, 2, 25)
     with pytest.raises(audio.AudioPayloadError):
         audio.concat_audio([a, {"waveform": torch.randn(1, 2, 5), "sample_rate": 4000}])
+
+
+def test_save_audio_writes_valid_wav(tmp_path):
+    import wave
+
+    from comfyui_distributed_amd.graph.builtin_nodes import SaveAudio
+
+    node = SaveAudio()
+    node.set_context({"output_dir": str(tmp_path), "saved_images": []})
+    t = torch.arange(800, dtype=torch.float32) / 8000.0
+    audio = {"waveform": (0.5 * torch.sin(2 * 3.14159 * 440 * t))
+             .expand(2, -1).unsqueeze(0).contiguous(),
+             "sample_rate": 8000}
+    node.save(audio, filename_prefix="tone")
+    path = tmp_path / "tone_00000.wav"
+    assert path.exists()
+    with wave.open(str(path), "rb") as w:
+        assert w.getnchannels() == 2
+        assert w.getframerate() == 8000
+        assert w.getnframes() == 800
+        assert w.getsampwidth() == 2
