@@ -340,6 +340,46 @@ class ImageScale(_ContextNode):
         return (out.clamp(0, 1),)
 
 
+class LoadAudio(_ContextNode):
+    """Load a WAV from the input dir (or "synthetic:<seconds>@<rate>")."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {"audio": ("STRING", {"default": ""})}}
+
+    RETURN_TYPES = ("AUDIO",)
+    FUNCTION = "load"
+    CATEGORY = "audio"
+
+    def load(self, audio=""):
+        import os
+        import wave
+        from pathlib import Path
+
+        name = str(audio)
+        if name.startswith("synthetic:"):
+            spec = name.split(":", 1)[1]
+            secs, rate = spec.split("@") if "@" in spec else (spec, "44100")
+            n = max(1, int(float(secs) * int(rate)))
+            t = torch.arange(n, dtype=torch.float32) / float(rate)
+            wf = (0.5 * torch.sin(2 * torch.pi * 440.0 * t))
+            return ({"waveform": wf.expand(2, -1).unsqueeze(0).contiguous(),
+                     "sample_rate": int(rate)},)
+        path = Path(self._ctx.get("input_dir", "input")) / os.path.basename(name)
+        with wave.open(str(path), "rb") as w:
+            nch, sw, sr, nfr = (w.getnchannels(), w.getsampwidth(),
+                                w.getframerate(), w.getnframes())
+            raw = w.readframes(nfr)
+        if sw != 2:
+            raise ValueError(f"only 16-bit PCM WAV supported, got {sw*8}-bit")
+        import numpy as np
+
+        pcm = torch.from_numpy(
+            np.frombuffer(raw, dtype=np.int16).copy()).reshape(nfr, nch)
+        wf = (pcm.float() / 32767.0).transpose(0, 1).unsqueeze(0).contiguous()
+        return ({"waveform": wf, "sample_rate": sr},)
+
+
 class SaveAudio(_ContextNode):
     """Persist an AUDIO dict as 16-bit PCM WAV (stdlib wave; the collector's
     gathered audio becomes a file the way SaveImage persists images)."""
@@ -565,4 +605,5 @@ BUILTIN_CLASS_MAPPINGS = {
     "ImageScale": ImageScale,
     "SetLatentNoiseMask": SetLatentNoiseMask,
     "SaveAudio": SaveAudio,
+    "LoadAudio": LoadAudio,
 }

@@ -80,3 +80,19 @@ def test_save_audio_writes_valid_wav(tmp_path):
         assert w.getframerate() == 8000
         assert w.getnframes() == 800
         assert w.getsampwidth() == 2
+
+
+def test_load_audio_roundtrip_through_save(tmp_path):
+    from comfyui_distributed_amd.graph.builtin_nodes import LoadAudio, SaveAudio
+
+    ctx = {"output_dir": str(tmp_path), "input_dir": str(tmp_path),
+           "saved_images": []}
+    la, sa = LoadAudio(), SaveAudio()
+    la.set_context(ctx); sa.set_context(ctx)
+    src = la.load("synthetic:0.01@8000")[0]
+    assert src["waveform"].shape == (1, 2, 80)
+    sa.save(src, filename_prefix="rt")
+    back = la.load("rt_00000.wav")[0]
+    assert back["sample_rate"] == 8000
+    assert back["waveform"].shape == (1, 2, 80)
+    assert torch.allclose(back["waveform"], src["waveform"], atol=1e-4)
