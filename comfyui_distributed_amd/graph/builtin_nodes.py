@@ -340,6 +340,55 @@ class ImageScale(_ContextNode):
         return (out.clamp(0, 1),)
 
 
+class SaveAnimatedWEBP(_ContextNode):
+    """IMAGE batch -> one animated WebP (ComfyUI-core node of the same
+    name; the video workflows' frame batches become a playable file)."""
+
+    @classmethod
+    def INPUT_TYPES(cls):
+        return {"required": {
+            "images": ("IMAGE",),
+            "filename_prefix": ("STRING", {"default": "video"}),
+            "fps": ("FLOAT", {"default": 16.0, "min": 0.1, "max": 120.0}),
+            "lossless": ("BOOLEAN", {"default": False}),
+            "quality": ("INT", {"default": 80, "min": 0, "max": 100}),
+        }}
+
+    RETURN_TYPES = ()
+    FUNCTION = "save"
+    OUTPUT_NODE = True
+    CATEGORY = "image/animation"
+
+    def save(self, images, filename_prefix="video", fps=16.0,
+             lossless=False, quality=80):
+        from pathlib import Path
+
+        from PIL import Image
+
+        out_dir = Path(self._ctx.get("output_dir", "output"))
+        out_dir.mkdir(parents=True, exist_ok=True)
+        frames = []
+        for i in range(images.shape[0]):
+            arr = (images[i].clamp(0, 1) * 255.0).round().to(torch.uint8)
+            frames.append(Image.fromarray(arr.cpu().numpy(), mode="RGB"))
+        if not frames:
+            raise ValueError("no frames to save")
+        p = out_dir / f"{filename_prefix}_00000.webp"
+        n = 0
+        while p.exists():
+            n += 1
+            p = out_dir / f"{filename_prefix}_{n:05d}.webp"
+        frames[0].save(
+            p, save_all=True, append_images=frames[1:], format="WEBP",
+            duration=int(round(1000.0 / float(fps))), loop=0,
+            lossless=bool(lossless), quality=int(quality),
+        )
+        sink = self._ctx.get("saved_images")
+        if isinstance(sink, list):
+            sink.append(str(p))
+        return ()
+
+
 class LoadAudio(_ContextNode):
     """Load a WAV from the input dir (or "synthetic:<seconds>@<rate>")."""
 
@@ -606,4 +655,5 @@ BUILTIN_CLASS_MAPPINGS = {
     "SetLatentNoiseMask": SetLatentNoiseMask,
     "SaveAudio": SaveAudio,
     "LoadAudio": LoadAudio,
+    "SaveAnimatedWEBP": SaveAnimatedWEBP,
 }

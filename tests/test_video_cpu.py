@@ -141,3 +141,22 @@ def test_i2v_pins_first_latent_frame():
     out = generate_video(stack, cond, None, p, start_image=start)
     assert out.shape == (5, 32, 32, 3)
     assert torch.isfinite(out).all()
+
+
+def test_save_animated_webp(tmp_path):
+    from PIL import Image
+
+    from comfyui_distributed_amd.graph.builtin_nodes import SaveAnimatedWEBP
+
+    node = SaveAnimatedWEBP()
+    node.set_context({"output_dir": str(tmp_path), "saved_images": []})
+    frames = torch.rand(5, 16, 16, 3)
+    node.save(frames, filename_prefix="clip", fps=8.0)
+    path = tmp_path / "clip_00000.webp"
+    assert path.exists()
+    im = Image.open(path)
+    assert getattr(im, "n_frames", 1) == 5
+    assert im.size == (16, 16)
+    # second save does not overwrite
+    node.save(frames, filename_prefix="clip", fps=8.0)
+    assert (tmp_path / "clip_00001.webp").exists()
