@@ -53,5 +53,7 @@ def test_wan_workflow_executes_downsized(tmp_path):
     ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
                            "device": "cpu"})
     ex.execute(wf)
-    # 5 frames split 3/2 across the two dividers' save nodes
-    assert len(saved) == 5
+    # 5 frames split 3/2 across the two dividers' save nodes, plus the
+    # animated webp of the full clip
+    assert len(saved) == 6
+    assert sum(1 for p in saved if p.endswith(".webp")) == 1
