@@ -604,8 +604,10 @@ class DistributedServer:
         path = Path(self.executor.context.get(key, kind)) / sub / name
         if not path.is_file():
             return _err("not found", status=404)
-        ctype = "image/png" if name.lower().endswith(".png") else \
-            "application/octet-stream"
+        ext = name.lower().rsplit(".", 1)[-1]
+        ctype = {"png": "image/png", "webp": "image/webp",
+                 "wav": "audio/wav", "jpg": "image/jpeg",
+                 "jpeg": "image/jpeg"}.get(ext, "application/octet-stream")
         return web.Response(body=path.read_bytes(), content_type=ctype)
 
     async def get_network_info(self, request):
