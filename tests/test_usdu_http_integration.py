@@ -414,3 +414,82 @@ def test_zombie_worker_image_requeued_dynamic_mode(tmp_config, monkeypatch):
     canvas = previews[0]
     assert canvas.shape == (2, 32, 32, 3)
     assert torch.isfinite(canvas).all()
+
+
+@pytest.mark.timeout(180)
+def test_video_upscale_pipeline_over_http(tmp_config, monkeypatch):
+    """The distributed_upscale_video.json composition, downsized: WAN
+    frames ride the batch dim into USDU, cross the dynamic threshold, and
+    the frames come back image-parallel via a real worker server."""
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+    monkeypatch.setattr(constants, "JOB_READY_POLL_INTERVAL", 0.2)
+
+    async def go():
+        worker_srv = DistributedServer(is_worker=True)
+        wc = TestClient(TestServer(worker_srv.build_app()))
+        await wc.start_server()
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "w", "host": "127.0.0.1",
+            "port": wc.server.port, "cuda_device": 0, "enabled": True,
+            "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+
+        prompt = {
+            "1": {"class_type": "CheckpointLoader",
+                  "inputs": {"ckpt_name": "wan_tiny"}},
+            "2": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "clip", "clip": ["1", 1]}},
+            "3": {"class_type": "WanVideoGenerate", "inputs": {
+                "model": ["1", 0], "positive": ["2", 0], "seed": 5,
+                "steps": 1, "cfg": 1.0, "width": 32, "height": 32,
+                "frames": 5}},
+            "4": {"class_type": "CheckpointLoader",
+                  "inputs": {"ckpt_name": "tiny"}},
+            "5": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "sharp", "clip": ["4", 1]}},
+            "6": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "", "clip": ["4", 1]}},
+            "7": {"class_type": "UltimateSDUpscaleDistributed", "inputs": {
+                "upscaled_image": ["3", 0], "model": ["4", 0],
+                "positive": ["5", 0], "negative": ["6", 0], "vae": ["4", 2],
+                "seed": 3, "steps": 1, "cfg": 1.0, "sampler_name": "euler",
+                "scheduler": "karras", "denoise": 0.4, "tile_width": 16,
+                "tile_height": 16, "padding": 16, "mask_blur": 2,
+                "force_uniform_tiles": True, "tiled_decode": False,
+                "dynamic_threshold": 2}},
+            "8": {"class_type": "DistributedCollector",
+                  "inputs": {"images": ["7", 0], "load_balance": False}},
+            "9": {"class_type": "PreviewImage", "inputs": {"images": ["8", 0]}},
+        }
+        r = await mc.post("/distributed/queue", json={
+            "prompt": prompt, "client_id": "vu", "enabled_worker_ids": ["w1"]})
+        assert r.status == 200
+
+        for _ in range(600):
+            if previews:
+                break
+            await asyncio.sleep(0.25)
+        from comfyui_distributed_amd.server.network import close_client_session
+
+        await close_client_session()
+        await wc.close()
+        await mc.close()
+        return previews
+
+    previews = asyncio.run(go())
+    assert previews, "video upscale pipeline never completed"
+    frames = previews[0]
+    assert frames.shape == (5, 32, 32, 3)
+    assert torch.isfinite(frames).all()
