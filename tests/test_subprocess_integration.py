@@ -117,3 +117,112 @@ def test_real_subprocess_worker_roundtrip(tmp_config, monkeypatch):
     # both be finite and in range
     assert torch.isfinite(combined).all()
     assert not torch.allclose(combined[0], combined[1])
+
+
+@pytest.mark.timeout(300)
+def test_subprocess_worker_usdu_tiles(tmp_config, monkeypatch):
+    """Tile-parallel USDU with a REAL worker subprocess: separate
+    interpreter, separate model init, tiles crossing the wire as PNG —
+    the strongest single determinism/protocol proof in the suite."""
+    from comfyui_distributed_amd.server import workers as workers_mod
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.2)
+    monkeypatch.setattr(constants, "JOB_READY_POLL_INTERVAL", 0.2)
+    wport = free_port()
+
+    async def go():
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = [{
+            "id": "w1", "name": "sub", "host": "127.0.0.1", "port": wport,
+            "cuda_device": 0, "enabled": True, "type": "remote",
+        }]
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+        os.environ["DISTGPU_LOG_DIR"] = os.path.dirname(
+            os.environ["DISTGPU_CONFIG"])
+        handle = workers_mod.launch_worker(cfg["workers"][0], monitor=False)
+        try:
+            import aiohttp
+
+            up = False
+            async with aiohttp.ClientSession() as s:
+                for _ in range(120):
+                    try:
+                        async with s.get(f"http://127.0.0.1:{wport}/prompt",
+                                         timeout=aiohttp.ClientTimeout(total=2)) as r:
+                            if r.status == 200:
+                                up = True
+                                break
+                    except Exception:
+                        pass
+                    await asyncio.sleep(0.5)
+            assert up, "worker subprocess never came up"
+
+            prompt = {
+                "1": {"class_type": "CheckpointLoader",
+                      "inputs": {"ckpt_name": "tiny"}},
+                "2": {"class_type": "CLIPTextEncode",
+                      "inputs": {"text": "detail", "clip": ["1", 1]}},
+                "3": {"class_type": "CLIPTextEncode",
+                      "inputs": {"text": "", "clip": ["1", 1]}},
+                "4": {"class_type": "LoadImage",
+                      "inputs": {"image": "synthetic:48x48"}},
+                "5": {"class_type": "UltimateSDUpscaleDistributed", "inputs": {
+                    "upscaled_image": ["4", 0], "model": ["1", 0],
+                    "positive": ["2", 0], "negative": ["3", 0],
+                    "vae": ["1", 2], "seed": 3, "steps": 1, "cfg": 1.0,
+                    "sampler_name": "euler", "scheduler": "karras",
+                    "denoise": 0.5, "tile_width": 16, "tile_height": 16,
+                    "padding": 16, "mask_blur": 2,
+                    "force_uniform_tiles": True, "tiled_decode": False}},
+                "6": {"class_type": "DistributedCollector",
+                      "inputs": {"images": ["5", 0], "load_balance": False}},
+                "7": {"class_type": "PreviewImage",
+                      "inputs": {"images": ["6", 0]}},
+            }
+            r = await mc.post("/distributed/queue", json={
+                "prompt": prompt, "client_id": "su",
+                "enabled_worker_ids": ["w1"]})
+            assert r.status == 200
+
+            for _ in range(600):
+                if previews:
+                    break
+                await asyncio.sleep(0.25)
+            return previews
+        finally:
+            workers_mod.stop_worker(handle, "w1")
+            from comfyui_distributed_amd.server.network import close_client_session
+
+            await close_client_session()
+            await mc.close()
+
+    previews = asyncio.run(go())
+    assert previews, "subprocess USDU job never completed"
+    canvas = previews[0]
+    assert canvas.shape == (1, 48, 48, 3)
+
+    from comfyui_distributed_amd.engine.usdu import USDUParams, process_single_gpu
+    from comfyui_distributed_amd.graph.builtin_nodes import (
+        _STACK_CACHE, stable_text_seed)
+
+    stack = _STACK_CACHE[("tiny", "cpu")]
+    cond = stack.make_conditioning(stable_text_seed("detail"))
+    uncond = stack.make_conditioning(stable_text_seed(""))
+    p = USDUParams(seed=3, steps=1, cfg=1.0, sampler_name="euler",
+                   scheduler="karras", denoise=0.5, tile_width=16,
+                   tile_height=16, padding=16, mask_blur=2)
+    img = torch.rand(1, 48, 48, 3, generator=torch.Generator().manual_seed(0))
+    ref = process_single_gpu(stack, cond, uncond, p, img)
+    # PNG-quantized wire tolerance, cross-PROCESS
+    assert torch.allclose(canvas, ref, atol=0.02), (
+        (canvas - ref).abs().max().item()
+    )
