@@ -234,3 +234,29 @@ def test_wire_decoders_total(payload):
             fn(dict(payload))
         except (TypeError, ValueError, KeyError):
             pass
+
+
+@given(st.recursive(
+    st.one_of(st.none(), st.booleans(), st.integers(-5, 5),
+              st.text(max_size=6)),
+    lambda children: st.dictionaries(st.text(max_size=6), children,
+                                     max_size=4),
+    max_leaves=12).filter(lambda v: isinstance(v, dict)))
+@settings(max_examples=100, deadline=None)
+def test_config_default_merge_properties(data):
+    """_merge_defaults: every default key present; unknown user keys
+    survive; user scalars win over defaults."""
+    from comfyui_distributed_amd.utils.config import (
+        DEFAULT_CONFIG, _merge_defaults)
+
+    merged = _merge_defaults(DEFAULT_CONFIG, data)
+    for key in DEFAULT_CONFIG:
+        assert key in merged
+    for key, value in data.items():
+        assert key in merged
+        if not isinstance(value, dict):
+            assert merged[key] == value  # user scalar wins
+    # defaults not clobbered when user section is a dict
+    if isinstance(data.get("settings"), dict):
+        for k in DEFAULT_CONFIG["settings"]:
+            assert k in merged["settings"]
