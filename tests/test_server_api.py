@@ -532,3 +532,32 @@ def test_view_subfolder(client, tmp_path):
         assert r.status == 400
 
     run(loop, go())
+
+
+def test_history_subfolder_outputs_fetchable(client, tmp_path):
+    srv, cl, loop = client
+
+    async def go():
+        srv.executor.context["output_dir"] = str(tmp_path)
+        srv.executor.context["device"] = "cpu"
+        prompt = {
+            "1": {"class_type": "LoadImage",
+                  "inputs": {"image": "synthetic:8x8"}},
+            "2": {"class_type": "SaveImage",
+                  "inputs": {"images": ["1", 0],
+                             "filename_prefix": "runX/img"}},
+        }
+        r = await cl.post("/prompt", json={"prompt": prompt, "client_id": "s"})
+        pid = (await r.json())["prompt_id"]
+        for _ in range(100):
+            body = await (await cl.get(f"/history/{pid}")).json()
+            if body:
+                break
+            await asyncio.sleep(0.05)
+        im = body[pid]["outputs"]["images"][0]
+        assert im["subfolder"] == "runX"
+        r = await cl.get(f"/view?filename={im['filename']}"
+                         f"&subfolder={im['subfolder']}")
+        assert r.status == 200
+
+    run(loop, go())
