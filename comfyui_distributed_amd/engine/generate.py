@@ -23,6 +23,12 @@ class GenParams:
 
 
 def generate_latents(stack, cond, uncond, p: GenParams) -> torch.Tensor:
+    if not hasattr(stack, "schedule"):
+        raise ValueError(
+            f"model family {getattr(stack, 'family', '?')!r} is not an "
+            "image-generation stack: video models go through "
+            "models.video.generate_video"
+        )
     ds = stack.vae.downscale
     shape = (p.batch_size, stack.cfg.unet.in_channels, p.height // ds, p.width // ds)
     g = torch.Generator(device="cpu").manual_seed(p.seed)

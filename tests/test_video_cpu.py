@@ -160,3 +160,13 @@ def test_save_animated_webp(tmp_path):
     # second save does not overwrite
     node.save(frames, filename_prefix="clip", fps=8.0)
     assert (tmp_path / "clip_00001.webp").exists()
+
+
+def test_generate_images_rejects_video_stack():
+    from comfyui_distributed_amd.engine.generate import GenParams, generate_images
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+
+    stack = create_diffusion_stack("wan_tiny")
+    with pytest.raises(ValueError, match="generate_video"):
+        generate_images(stack, stack.make_conditioning(0), None,
+                        GenParams(seed=1, steps=1, width=16, height=16))
