@@ -48,6 +48,10 @@ def generate_latents_flux(stack, cond, uncond, p: GenParams,
     interpolant, same math USDU uses per tile)."""
     from ..models.video import sample_flow
 
+    if getattr(stack, "family", "") != "flux":
+        raise ValueError(
+            f"generate_latents_flux needs a flux stack, got "
+            f"{getattr(stack, 'family', '?')!r}")
     b = p.batch_size
     shape = (b, stack.cfg.in_channels, p.height // 8, p.width // 8)
     g = torch.Generator(device="cpu").manual_seed(p.seed)

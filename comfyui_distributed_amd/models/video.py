@@ -220,6 +220,11 @@ def generate_video(stack: WanStack, cond, uncond, p: VideoGenParams,
     pinned to the encoded image throughout the flow integration (temporal
     inpainting — the real WAN i2v conditions on the first frame the same
     way structurally, via a masked video latent)."""
+    if getattr(stack, "family", "") != "wan":
+        raise ValueError(
+            f"model family {getattr(stack, 'family', '?')!r} is not a video "
+            "stack: image models generate via engine.generate_images"
+        )
     stack.validate_frames(p.frames)
     g = torch.Generator().manual_seed(p.seed)
     lat_t = stack.latent_frames(p.frames)  # 1 + (T-1)/4

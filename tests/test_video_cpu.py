@@ -170,3 +170,14 @@ def test_generate_images_rejects_video_stack():
     with pytest.raises(ValueError, match="generate_video"):
         generate_images(stack, stack.make_conditioning(0), None,
                         GenParams(seed=1, steps=1, width=16, height=16))
+
+
+def test_generate_video_rejects_image_stack():
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+    from comfyui_distributed_amd.models.video import VideoGenParams, generate_video
+
+    stack = create_diffusion_stack("tiny")
+    with pytest.raises(ValueError, match="not a video"):
+        generate_video(stack, stack.make_conditioning(0), None,
+                       VideoGenParams(seed=1, steps=1, width=16, height=16,
+                                      frames=5))
