@@ -204,3 +204,27 @@ def test_gather_variable_counts_world4():
     assert sorted(res["meta"]) == sorted(
         [(0, i) for i in range(2)]
         + [(r, i) for r in range(1, 4) for i in range(3)])
+
+
+def _body_flux_seed_parallel(ctx, port):
+    from comfyui_distributed_amd.engine.generate import GenParams
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+    from comfyui_distributed_amd.parallel.collector import seed_parallel_generate
+
+    stack = create_diffusion_stack("flux_tiny", seed=7)
+    cond = stack.make_conditioning(0)
+    p = GenParams(seed=11, steps=1, cfg=1.0, width=16, height=16,
+                  batch_size=1)
+    out = seed_parallel_generate(ctx, stack, cond, None, p)
+    if ctx.is_master:
+        return {"images": out}
+    return None
+
+
+def test_flux_seed_parallel_world2():
+    out = _spawn("_body_flux_seed_parallel", port=PORT_BASE + 8)
+    images = out[0]["images"]
+    assert images.shape == (2, 16, 16, 3)
+    assert torch.isfinite(images).all()
+    # per-rank seed offsets -> different images
+    assert not torch.allclose(images[0], images[1])
