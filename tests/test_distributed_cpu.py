@@ -228,3 +228,35 @@ def test_flux_seed_parallel_world2():
     assert torch.isfinite(images).all()
     # per-rank seed offsets -> different images
     assert not torch.allclose(images[0], images[1])
+
+
+def _body_wan_seed_parallel(ctx, port):
+    from dataclasses import replace
+
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+    from comfyui_distributed_amd.models.video import VideoGenParams, generate_video
+    from comfyui_distributed_amd.parallel.collector import seed_for_rank
+    from comfyui_distributed_amd.parallel.dist import gather_tensor_lists
+
+    stack = create_diffusion_stack("wan_tiny", seed=7)
+    cond = stack.make_conditioning(0)
+    p = VideoGenParams(seed=3, steps=1, cfg=1.0, width=16, height=16,
+                       frames=5)
+    local = generate_video(stack, cond, None,
+                           replace(p, seed=seed_for_rank(p.seed, ctx.rank)))
+    tensors = [local[i] for i in range(local.shape[0])]
+    meta = [(ctx.rank, i) for i in range(local.shape[0])]
+    res = gather_tensor_lists(ctx, tensors, meta)
+    if ctx.is_master:
+        all_t, all_m = res
+        return {"n": len(all_t), "meta": sorted(all_m)}
+    return None
+
+
+def test_wan_seed_parallel_frame_gather_world2():
+    """The bench.py wan-t2v gather shape: every rank's 5 frames arrive on
+    rank 0 with (rank, frame) metadata intact."""
+    out = _spawn("_body_wan_seed_parallel", port=PORT_BASE + 9)
+    res = out[0]
+    assert res["n"] == 10
+    assert res["meta"] == [(r, i) for r in range(2) for i in range(5)]
