@@ -40,3 +40,21 @@ def test_bench_json_contract_tiny():
     cfg = data["config"]
     for key in ("model", "global_batch", "parallelism"):
         assert key in cfg, key
+
+
+@pytest.mark.timeout(600)
+def test_bench_generation_branch_contract():
+    """The seed-parallel generation branch (gen-sdxl / gen-flux / wan-t2v
+    presets) also honors the JSON contract."""
+    proc = subprocess.run(
+        [sys.executable, "bench.py", "--config", "gen-flux", "--model",
+         "flux_tiny", "--steps", "1", "--warmup", "0",
+         "--sampler-steps", "1"],
+        cwd=ROOT, capture_output=True, text=True, timeout=540,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    json_lines = [ln for ln in proc.stdout.splitlines() if ln.startswith("{")]
+    assert len(json_lines) == 1, proc.stdout
+    data = json.loads(json_lines[0])
+    assert REQUIRED_KEYS.issubset(data)
+    assert data["unit"] == "images/s" and data["value"] > 0
