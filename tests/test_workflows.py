@@ -57,3 +57,35 @@ def test_wan_workflow_executes_downsized(tmp_path):
     # animated webp of the full clip
     assert len(saved) == 6
     assert sum(1 for p in saved if p.endswith(".webp")) == 1
+
+
+def test_hires_fix_workflow_executes_downsized(tmp_path):
+    wf = load_wf("hires_fix.json")
+    wf["1"]["inputs"]["ckpt_name"] = "tiny"
+    wf["4"]["inputs"].update(width=16, height=16)
+    wf["5"]["inputs"].update(steps=1)
+    wf["6"]["inputs"].update(width=32, height=32)
+    wf["7"]["inputs"].update(steps=1)
+    saved = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "device": "cpu"})
+    ex.execute(wf)
+    assert len(saved) == 1
+    import torch as _t
+
+    from comfyui_distributed_amd.utils.image import decode_png_bytes
+
+    img = decode_png_bytes(Path(saved[0]).read_bytes())
+    assert img.shape == (1, 32, 32, 3)  # the upscaled size
+
+
+def test_inpaint_workflow_executes_downsized(tmp_path):
+    wf = load_wf("inpaint.json")
+    wf["1"]["inputs"]["ckpt_name"] = "tiny"
+    wf["4"]["inputs"]["image"] = "synthetic:32x32"
+    wf["8"]["inputs"].update(steps=1)
+    saved = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "device": "cpu"})
+    ex.execute(wf)
+    assert len(saved) == 1
