@@ -89,3 +89,26 @@ def test_inpaint_workflow_executes_downsized(tmp_path):
                            "device": "cpu"})
     ex.execute(wf)
     assert len(saved) == 1
+
+
+def test_flux_workflow_executes_downsized(tmp_path):
+    wf = load_wf("distributed_flux_txt2img.json")
+    wf["1"]["inputs"]["ckpt_name"] = "flux_tiny"
+    wf["4"]["inputs"].update(width=16, height=16, steps=1)
+    saved = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "device": "cpu"})
+    ex.execute(wf)
+    assert len(saved) == 1
+
+
+def test_audio_workflow_executes_downsized(tmp_path):
+    wf = load_wf("distributed_audio_collect.json")
+    wf["2"]["inputs"].update(seconds=0.01, sample_rate=8000)
+    saved = []
+    previews = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "preview_images": previews, "device": "cpu"})
+    ex.execute(wf)
+    assert any(p.endswith(".wav") for p in saved)
+    assert previews
