@@ -112,3 +112,16 @@ def test_audio_workflow_executes_downsized(tmp_path):
     ex.execute(wf)
     assert any(p.endswith(".wav") for p in saved)
     assert previews
+
+
+def test_i2v_workflow_executes_downsized(tmp_path):
+    wf = load_wf("wan_image_to_video.json")
+    wf["1"]["inputs"]["ckpt_name"] = "wan_tiny"
+    wf["3"]["inputs"]["image"] = "synthetic:16x16"
+    wf["4"]["inputs"].update(width=16, height=16, frames=5, steps=1)
+    saved = []
+    previews = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "preview_images": previews, "device": "cpu"})
+    ex.execute(wf)
+    assert any(p.endswith(".webp") for p in saved)
