@@ -561,3 +561,21 @@ def test_history_subfolder_outputs_fetchable(client, tmp_path):
         assert r.status == 200
 
     run(loop, go())
+
+
+def test_view_content_types(client, tmp_path):
+    srv, cl, loop = client
+
+    async def go():
+        from pathlib import Path
+
+        out = Path(srv.executor.context.setdefault("output_dir",
+                                                   str(tmp_path)))
+        out.mkdir(parents=True, exist_ok=True)
+        for name, ctype in (("a.png", "image/png"), ("b.webp", "image/webp"),
+                            ("c.wav", "audio/wav")):
+            (out / name).write_bytes(b"data")
+            r = await cl.get(f"/view?filename={name}")
+            assert r.status == 200 and r.content_type == ctype, name
+
+    run(loop, go())
