@@ -236,3 +236,13 @@ def test_probe_concurrency_is_bounded(monkeypatch):
     res = asyncio.run(orchestration.probe_workers(workers, concurrency=3))
     assert len(res) == 12 and all(v is not None for v in res.values())
     assert peak["max"] <= 3
+
+
+def test_unknown_worker_ids_fall_back_to_master(two_worker_config, monkeypatch):
+    payload = QueueRequestPayload(prompt=dist_prompt(), client_id="c",
+                                  enabled_worker_ids=["ghost1", "ghost2"])
+    probes = {"w1": {"exec_info": {"queue_remaining": 0}}}
+    result, calls, _ = run_orch(payload, monkeypatch, probes)
+    assert result["participants"] == ["master"]
+    assert calls["dispatched"] == []
+    assert len(calls["local"]) == 1
