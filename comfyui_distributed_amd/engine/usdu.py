@@ -134,8 +134,14 @@ def sample_tiles(
     # share a sampler batch (reference processes per tile for the same
     # reason: tile_ops.py crop context + crop_model_patch.py)
     step = 1 if has_spatial else max(params.tile_batch, 1)
+    from ..nodes.runtime import get_runtime
+
     results: dict[tuple[int, int], torch.Tensor] = {}
     for i in range(0, len(work), step):
+        # user interrupt aborts between chunks in EVERY mode (the
+        # distributed loops also check; this covers single-GPU/local —
+        # reference tile loops check comfy.model_management per tile)
+        get_runtime().throw_if_interrupted()
         chunk = work[i : i + step]
         chunk_cond, chunk_uncond = cond, uncond
         if has_spatial:

@@ -493,3 +493,69 @@ def test_video_upscale_pipeline_over_http(tmp_config, monkeypatch):
     frames = previews[0]
     assert frames.shape == (5, 32, 32, 3)
     assert torch.isfinite(frames).all()
+
+
+@pytest.mark.timeout(120)
+def test_interrupt_aborts_running_usdu(tmp_config, monkeypatch):
+    """POST /interrupt mid-job: the master's collection loop aborts, the
+    tile job is cleaned up (finally path), and no preview is produced."""
+    from comfyui_distributed_amd.utils import constants
+
+    monkeypatch.setattr(constants, "COLLECTOR_SLICE_TIMEOUT", 0.1)
+
+    async def go():
+        master_srv = DistributedServer()
+        previews: list = []
+        master_srv.executor.context["preview_images"] = previews
+        mc = TestClient(TestServer(master_srv.build_app()))
+        await mc.start_server()
+
+        cfg = load_config()
+        cfg["workers"] = []
+        cfg["master"]["host"] = "127.0.0.1"
+        cfg["master"]["port"] = mc.server.port
+        save_config(cfg)
+
+        # a deliberately slow local job: 64 tiles x 4 steps on the tiny
+        # stack keeps the executor busy long enough to interrupt
+        prompt = {
+            "1": {"class_type": "CheckpointLoader", "inputs": {"ckpt_name": "tiny"}},
+            "2": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "detail", "clip": ["1", 1]}},
+            "3": {"class_type": "CLIPTextEncode",
+                  "inputs": {"text": "", "clip": ["1", 1]}},
+            "4": {"class_type": "LoadImage",
+                  "inputs": {"image": "synthetic:128x128"}},
+            "5": {"class_type": "UltimateSDUpscaleDistributed", "inputs": {
+                "upscaled_image": ["4", 0], "model": ["1", 0],
+                "positive": ["2", 0], "negative": ["3", 0], "vae": ["1", 2],
+                "seed": 3, "steps": 4, "cfg": 2.0, "sampler_name": "euler",
+                "scheduler": "karras", "denoise": 0.5, "tile_width": 16,
+                "tile_height": 16, "padding": 16, "mask_blur": 2,
+                "force_uniform_tiles": True, "tiled_decode": False}},
+            "6": {"class_type": "PreviewImage", "inputs": {"images": ["5", 0]}},
+        }
+        r = await mc.post("/prompt", json={"prompt": prompt, "client_id": "i"})
+        assert r.status == 200
+        pid = (await r.json())["prompt_id"]
+        await asyncio.sleep(0.8)  # let the job start chewing tiles
+        r = await mc.post("/interrupt")
+        assert r.status == 200
+        # the prompt must finish (as an error) promptly, not run all tiles
+        done = False
+        for _ in range(200):
+            body = await (await mc.get(f"/history/{pid}")).json()
+            if body:
+                done = True
+                break
+            await asyncio.sleep(0.1)
+        assert done, "interrupted prompt never resolved"
+        assert body[pid]["status"]["completed"] is False
+        assert previews == []
+        from comfyui_distributed_amd.nodes.runtime import get_runtime
+
+        get_runtime().clear_interrupt()
+        await mc.close()
+        return True
+
+    assert asyncio.run(go())
