@@ -239,6 +239,20 @@ class _MaskedDenoiser:
         return self.inner(x, sigma)
 
 
+class _InterruptibleDenoiser:
+    """Checks the user-interrupt flag before every model call, so a long
+    sampler loop aborts promptly in every mode (ComfyUI checks
+    model_management interrupts per step the same way)."""
+
+    def __init__(self, inner, runtime):
+        self.inner = inner
+        self.runtime = runtime
+
+    def __call__(self, x, sigma):
+        self.runtime.throw_if_interrupted()
+        return self.inner(x, sigma)
+
+
 def sample(denoiser, noise_or_latent: torch.Tensor, sigmas: torch.Tensor,
            sampler: str = "euler", seed: int | None = None,
            start_from_latent: torch.Tensor | None = None,
@@ -260,6 +274,9 @@ def sample(denoiser, noise_or_latent: torch.Tensor, sigmas: torch.Tensor,
     gen = None
     if seed is not None:
         gen = torch.Generator(device="cpu").manual_seed(seed)
+    from ..nodes.runtime import get_runtime
+
+    denoiser = _InterruptibleDenoiser(denoiser, get_runtime())
     if sampler == "euler":
         out = _sample_euler(denoiser, x, sigmas)
     elif sampler == "euler_ancestral":

@@ -202,7 +202,11 @@ def sample_flow(velocity_fn, noise: torch.Tensor, steps: int,
         keep = (1 - t) * pin_latent.float() + t * noise.float()
         return xc * (1 - pin_mask) + keep * pin_mask
 
+    from ..nodes.runtime import get_runtime
+
+    rt = get_runtime()
     for i in range(steps):
+        rt.throw_if_interrupted()  # user interrupt aborts per flow step
         t, t_next = sig[i], sig[i + 1]
         x = pin(x, t)
         v = velocity_fn(x, t)
