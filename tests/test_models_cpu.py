@@ -290,3 +290,27 @@ def test_save_image_subfolder_prefix(tmp_path):
         node.save(torch.rand(1, 4, 4, 3), filename_prefix="../escape")
     with _pytest.raises(ValueError):
         node.save(torch.rand(1, 4, 4, 3), filename_prefix="/abs/path")
+
+
+def test_sampler_and_flow_loops_abort_on_interrupt():
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+    from comfyui_distributed_amd.models.sampling import sample
+    from comfyui_distributed_amd.models.video import sample_flow
+    from comfyui_distributed_amd.nodes.runtime import get_runtime
+
+    rt = get_runtime()
+    rt.interrupt()
+    try:
+        stack = create_diffusion_stack("tiny")
+        from comfyui_distributed_amd.models.sampling import (
+            CFGDenoiser, NoiseSchedule)
+
+        den = CFGDenoiser(stack.unet, stack.schedule,
+                          stack.make_conditioning(0), None, 1.0)
+        sig = stack.schedule.sigmas(2, "karras")
+        with pytest.raises(InterruptedError):
+            sample(den, torch.randn(1, 4, 4, 4), sig)
+        with pytest.raises(InterruptedError):
+            sample_flow(lambda x, t: x, torch.randn(1, 4, 2, 2, 2), steps=2)
+    finally:
+        rt.clear_interrupt()
