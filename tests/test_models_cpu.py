@@ -1,5 +1,6 @@
 """Shape/NaN sanity of the model stack on CPU (tiny config)."""
 
+import pytest
 import torch
 
 from comfyui_distributed_amd.models import create_diffusion_stack
