@@ -125,3 +125,30 @@ def test_i2v_workflow_executes_downsized(tmp_path):
                            "preview_images": previews, "device": "cpu"})
     ex.execute(wf)
     assert any(p.endswith(".webp") for p in saved)
+
+
+def test_parameter_sweep_workflow_executes_downsized(tmp_path):
+    wf = load_wf("parameter_sweep.json")
+    wf["1"]["inputs"]["ckpt_name"] = "tiny"
+    wf["5"]["inputs"].update(width=16, height=16)
+    wf["6"]["inputs"].update(steps=1)
+    saved = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "device": "cpu"})
+    ex.execute(wf)
+    assert len(saved) == 1
+
+
+def test_upscale_video_workflow_executes_downsized(tmp_path):
+    wf = load_wf("distributed_upscale_video.json")
+    wf["1"]["inputs"]["ckpt_name"] = "wan_tiny"
+    wf["3"]["inputs"].update(width=16, height=16, frames=5, steps=1)
+    wf["4"]["inputs"]["ckpt_name"] = "tiny"
+    wf["7"]["inputs"].update(steps=1, tile_width=16, tile_height=16,
+                             padding=16, mask_blur=2)
+    saved = []
+    previews = []
+    ex = Executor(context={"output_dir": str(tmp_path), "saved_images": saved,
+                           "preview_images": previews, "device": "cpu"})
+    ex.execute(wf)
+    assert len(saved) == 5  # five upscaled frames
