@@ -77,6 +77,10 @@ def build_worker_url(worker: dict) -> str:
     without an explicit port; everything else http://host:port."""
     host = normalize_host(worker.get("host") or "localhost")
     port = worker.get("port")
+    # a host entered as "1.2.3.4:9000" carries its own port
+    bare, embedded = split_host_port(host)
+    if embedded is not None:
+        host, port = bare, embedded
     if is_cloud_host(host):
         return f"https://{host}"
     if not host:

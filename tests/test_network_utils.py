@@ -36,3 +36,8 @@ def test_master_url_tunnel():
     assert (network.build_master_url({"host": "t.trycloudflare.com", "port": 8188})
             == "https://t.trycloudflare.com")
     assert network.build_master_url({"host": "", "port": 9000}) == "http://127.0.0.1:9000"
+
+
+def test_build_worker_url_host_with_embedded_port():
+    assert (network.build_worker_url({"host": "10.0.0.2:9000", "port": 8190})
+            == "http://10.0.0.2:9000")
