@@ -67,11 +67,14 @@ class TemporalVAE(nn.Module):
         return 1 + (frames - 1) // self.t_down
 
     def _time_apply(self, conv, z):
-        # z [C, T, h, w] -> conv over T batched across pixels
+        # z [C, T, h, w] -> conv over T batched across pixels.
+        # Inputs are cast to the conv's own weight dtype: when the module is
+        # .to(bf16) a caller-side fp32 tensor would otherwise be a dtype
+        # mismatch inside Conv1d/ConvTranspose1d on GPU.
         c, t, h, w = z.shape
-        seq = z.permute(2, 3, 0, 1).reshape(h * w, c, t)
+        seq = z.permute(2, 3, 0, 1).reshape(h * w, c, t).to(conv.weight.dtype)
         out = conv(seq)  # [h*w, C, T']
-        return out.reshape(h, w, c, -1).permute(2, 3, 0, 1)
+        return out.reshape(h, w, c, -1).permute(2, 3, 0, 1).to(z.dtype)
 
     def encode(self, frames: torch.Tensor) -> torch.Tensor:
         t = frames.shape[0]
@@ -82,7 +85,7 @@ class TemporalVAE(nn.Module):
         # length T + (k-1) with k = t_down+1 gives exactly 1+(T-1)/4
         pad = self.tdown.kernel_size[0] - 1
         z = torch.cat([z[:, :1].expand(-1, pad, -1, -1), z], dim=1)
-        z = self._time_apply(self.tdown, z.float())
+        z = self._time_apply(self.tdown, z)
         assert z.shape[1] == n_lat
         return z.to(frames.dtype)
 
@@ -91,7 +94,7 @@ class TemporalVAE(nn.Module):
         4n+1 clip length T_lat encodes."""
         t_lat = z.shape[1]
         t = frames if frames is not None else 1 + (t_lat - 1) * self.t_down
-        zs = self._time_apply(self.tup, z.float())  # [Cz, 4*T_lat, h, w]
+        zs = self._time_apply(self.tup, z)  # [Cz, 4*T_lat, h, w]
         zs = zs[:, :t]
         return self.spatial.decode(
             zs.permute(1, 0, 2, 3).to(z.dtype)

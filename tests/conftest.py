@@ -23,6 +23,10 @@ def pytest_collection_modifyitems(config, items):
     except Exception:
         has_gpu = False
     if has_gpu:
+        # Kernel-numerics suites run FIRST on a GPU box so a model-level
+        # failure under -x can never hide them (driver runs -x -m gpu).
+        order = {"test_ops_gpu": 0, "test_conv_gpu": 1}
+        items.sort(key=lambda it: order.get(it.module.__name__.rsplit(".", 1)[-1], 2))
         return
     skip = pytest.mark.skip(reason="no GPU available")
     for item in items:

@@ -97,6 +97,24 @@ def test_temporal_vae_compression_and_causality():
         vae.encode(torch.rand(8, 32, 32, 3))  # not 4n+1
 
 
+def test_temporal_vae_bf16_dtype_mix():
+    """Regression: a bf16 TemporalVAE fed fp32/bf16 frames must not crash
+    with a Conv1d dtype mismatch (round-1 GPU gate failure, video.py:73)."""
+    from comfyui_distributed_amd.models.registry import create_diffusion_stack
+
+    stack = create_diffusion_stack("wan_tiny")
+    vae = stack.vae.to(torch.bfloat16)
+    frames = torch.rand(5, 32, 32, 3, dtype=torch.bfloat16)
+    z = vae.encode(frames)
+    assert z.dtype == torch.bfloat16
+    out = vae.decode(z, frames=5)
+    assert out.shape == (5, 32, 32, 3)
+    assert torch.isfinite(out.float()).all()
+    # fp32 latent into a bf16 decoder must also work (the exact crash mode)
+    out2 = vae.decode(z.float(), frames=5)
+    assert out2.shape == (5, 32, 32, 3)
+
+
 def test_generate_video_latent_shrinks_4x():
     from comfyui_distributed_amd.models.registry import create_diffusion_stack
     from comfyui_distributed_amd.models.video import VideoGenParams, generate_video
