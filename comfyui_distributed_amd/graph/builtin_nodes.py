@@ -556,16 +556,46 @@ class LoadImage(_ContextNode):
     FUNCTION = "load"
     CATEGORY = "image"
 
+    @staticmethod
+    def _resolve_input_path(input_dir, name):
+        """Resolve ``name`` strictly inside the input dir. Prompts arrive
+        over an unauthenticated endpoint, so an absolute path or a ``..``
+        escape must not read arbitrary files (the reference confines loads
+        via ComfyUI's folder_paths the same way)."""
+        import os
+        from pathlib import Path
+
+        base = Path(input_dir).resolve()
+        cand = (base / str(name)).resolve()
+        if os.path.commonpath([str(base), str(cand)]) != str(base):
+            raise ValueError(f"image path {name!r} escapes the input dir")
+        return cand
+
+    @classmethod
+    def IS_CHANGED(cls, context=None, image=""):
+        """Content mark folded into the execution-cache fingerprint: a
+        re-uploaded file under the same name (media sync) re-executes the
+        load instead of returning the stale cached tensor."""
+        name = str(image)
+        if name.startswith("synthetic:"):
+            return name
+        try:
+            ctx = context or {}
+            path = cls._resolve_input_path(ctx.get("input_dir", "input"), name)
+            st = path.stat()
+            return (st.st_mtime_ns, st.st_size)
+        except Exception:
+            return float("nan")  # unreadable -> always re-execute (and fail there)
+
     def load(self, image=""):
         name = str(image)
         if name.startswith("synthetic:"):
             w, h = (int(v) for v in name.split(":", 1)[1].split("x"))
             g = torch.Generator().manual_seed(0)
             return (torch.rand(1, h, w, 3, generator=g),)
-        from pathlib import Path
-
-        base = Path(self._ctx.get("input_dir", "input"))
-        raw = (base / name).read_bytes()
+        path = self._resolve_input_path(
+            self._ctx.get("input_dir", "input"), name)
+        raw = path.read_bytes()
         return (decode_png_bytes(raw),)
 
 

@@ -8,10 +8,16 @@ framework executes prompt graphs itself: topological evaluation of the
 
 from __future__ import annotations
 
+import itertools
 from typing import Any
 
 from ..utils.errors import PromptValidationError
 from .prompt import PromptGraph, is_link
+
+#: sentinel for "this node must never be cross-run cached" (IS_CHANGED nan)
+_VOLATILE = object()
+#: monotone source for volatile fingerprints (never repeats across runs)
+_volatile_seq = itertools.count()
 
 
 class NodeRegistry:
@@ -132,6 +138,27 @@ class Executor:
         #: run so memory stays bounded by one workflow's outputs
         self._run_cache: dict[str, tuple] = {}
 
+    def _is_changed_mark(self, class_type: str, widgets: dict):
+        """Evaluate a node class's IS_CHANGED hook over its widget values.
+
+        Returns ``None`` (no hook), ``_VOLATILE`` (nan / failing hook =>
+        always re-execute, ComfyUI parity), or a hashable content mark to
+        fold into the fingerprint (e.g. a file's mtime+size)."""
+        cls = self.registry.get(class_type)
+        if cls is None or not hasattr(cls, "IS_CHANGED"):
+            return None
+        try:
+            fn = cls.IS_CHANGED
+            try:
+                mark = fn(context=self.context, **widgets)
+            except TypeError:
+                mark = fn(**widgets)
+        except Exception:
+            return _VOLATILE
+        if isinstance(mark, float) and mark != mark:  # nan
+            return _VOLATILE
+        return mark
+
     def _fingerprints(self, graph: PromptGraph) -> dict[str, str]:
         import hashlib
         import json
@@ -149,8 +176,17 @@ class Executor:
                     links[name] = (fp(str(value[0])), int(value[1]))
                 else:
                     widgets[name] = value
+            # ComfyUI's IS_CHANGED convention: the class may report a
+            # content mark (e.g. LoadImage's file mtime/size) that is folded
+            # into the fingerprint so a re-uploaded file under the same name
+            # busts the cross-run cache; a NaN (or a failing hook) means
+            # "always changed" and poisons this node AND its downstream.
+            mark = self._is_changed_mark(node["class_type"], widgets)
+            if mark is _VOLATILE:
+                fps[nid] = f"volatile:{nid}:{next(_volatile_seq)}"
+                return fps[nid]
             spec = json.dumps(
-                [node["class_type"], widgets, links],
+                [node["class_type"], widgets, links, repr(mark)],
                 sort_keys=True, default=repr,
             )
             fps[nid] = hashlib.md5(spec.encode()).hexdigest()
@@ -175,11 +211,12 @@ class Executor:
                 return cache[nid]
             node = graph.node(nid)
             cls = self.registry.get(node["class_type"])
-            # OUTPUT_NODEs and nodes declaring IS_CHANGED (ComfyUI's
-            # always-re-execute convention, e.g. the USDU node's nan)
-            # never reuse a cross-run cached result
-            is_output = bool(getattr(cls, "OUTPUT_NODE", False)) or \
-                hasattr(cls, "IS_CHANGED")
+            # OUTPUT_NODEs never reuse a cross-run cached result; IS_CHANGED
+            # volatility (the USDU node's nan) is handled in the fingerprint
+            # itself (a volatile fingerprint never matches across runs), so
+            # content-marking nodes like LoadImage can still cache when their
+            # file is unchanged
+            is_output = bool(getattr(cls, "OUTPUT_NODE", False))
             if not is_output and fps[nid] in self._run_cache:
                 out = self._run_cache[fps[nid]]
                 cache[nid] = out

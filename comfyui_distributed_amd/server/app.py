@@ -518,12 +518,20 @@ class DistributedServer:
         data = await request.json()
         candidates = []
         name = data.get("filename")
+        input_dir = str(self.executor.context.get("input_dir", "input"))
+        output_dir = str(self.executor.context.get("output_dir", "output"))
         if name:
             candidates.append(os.path.join(
-                str(self.executor.context.get("input_dir", "input")),
-                os.path.basename(str(name))))
+                input_dir, os.path.basename(str(name))))
         if data.get("path"):
-            candidates.append(str(data["path"]))
+            # same-filesystem fallback, but never an arbitrary-path md5
+            # oracle: the path must resolve inside the input or output dir
+            p = os.path.realpath(str(data["path"]))
+            for base in (input_dir, output_dir):
+                rbase = os.path.realpath(base)
+                if os.path.commonpath([rbase, p]) == rbase:
+                    candidates.append(p)
+                    break
         path = next((p for p in candidates if os.path.isfile(p)), None)
         if path is None:
             return web.json_response({"exists": False})

@@ -225,3 +225,51 @@ def test_auto_populate_creates_workers_for_fake_gpus(tmp_config, monkeypatch):
             await cl.close()
 
     asyncio.run(go())
+
+
+def test_load_image_rejects_path_escape(tmp_path):
+    """LoadImage must not read files outside the input dir."""
+    from comfyui_distributed_amd.graph.builtin_nodes import LoadImage
+
+    secret = tmp_path / "secret.png"
+    secret.write_bytes(b"x")
+    input_dir = tmp_path / "input"
+    input_dir.mkdir()
+    node = LoadImage()
+    node.set_context({"input_dir": str(input_dir)})
+    for name in ("../secret.png", str(secret), "a/../../secret.png"):
+        with pytest.raises(ValueError, match="escapes"):
+            node.load(image=name)
+
+
+def test_load_image_cache_busts_on_file_change(tmp_path):
+    """a re-uploaded file under the same name re-executes the
+    load (IS_CHANGED content mark in the fingerprint)."""
+    import os
+    import time
+
+    import torch
+
+    from comfyui_distributed_amd.graph.executor import Executor
+    from comfyui_distributed_amd.utils.image import encode_png_bytes
+
+    input_dir = tmp_path / "input"
+    input_dir.mkdir()
+    img1 = torch.zeros(1, 4, 4, 3)
+    img2 = torch.ones(1, 4, 4, 3)
+    p = input_dir / "x.png"
+    p.write_bytes(encode_png_bytes(img1))
+    ex = Executor(context={"input_dir": str(input_dir),
+                           "output_dir": str(tmp_path / "out")})
+    prompt = {"1": {"class_type": "LoadImage", "inputs": {"image": "x.png"}},
+              "2": {"class_type": "PreviewImage", "inputs": {"images": ["1", 0]}}}
+    out1 = ex.execute(prompt)["1"][0]
+    assert float(out1.sum()) == 0.0
+    # unchanged file: cached (same tensor object back)
+    assert ex.execute(prompt)["1"][0] is out1
+    # replace content under the same name -> must re-read
+    p.write_bytes(encode_png_bytes(img2))
+    st = p.stat()
+    os.utime(p, ns=(st.st_atime_ns, st.st_mtime_ns + 1_000_000))
+    out2 = ex.execute(prompt)["1"][0]
+    assert float(out2.mean()) > 0.9

@@ -96,3 +96,40 @@ def test_is_local_worker_heuristics(tmp_config):
         assert run(media_sync.is_local_worker(w)) is False
     finally:
         media_sync.fetch_worker_system_info = orig
+
+
+def test_check_file_rejects_out_of_tree_path(tmp_config, tmp_path):
+    """the raw `path` fallback must not be an arbitrary-file
+    md5 oracle — only paths inside the input/output dirs are answered."""
+    import asyncio
+
+    from aiohttp.test_utils import TestClient, TestServer
+
+    from comfyui_distributed_amd.server.app import DistributedServer
+
+    secret = tmp_path / "secret.bin"
+    secret.write_bytes(b"topsecret")
+    input_dir = tmp_path / "input"
+    input_dir.mkdir()
+    inside = input_dir / "ok.png"
+    inside.write_bytes(b"pngdata")
+
+    async def go():
+        srv = DistributedServer()
+        srv.executor.context["input_dir"] = str(input_dir)
+        srv.executor.context["output_dir"] = str(tmp_path / "out")
+        cl = TestClient(TestServer(srv.build_app()))
+        await cl.start_server()
+        try:
+            # outside path: existence/hash not disclosed
+            r = await cl.post("/distributed/check_file",
+                              json={"path": str(secret)})
+            assert (await r.json())["exists"] is False
+            # inside path still works (same-filesystem fast path)
+            r = await cl.post("/distributed/check_file",
+                              json={"path": str(inside)})
+            assert (await r.json())["exists"] is True
+        finally:
+            await cl.close()
+
+    asyncio.run(go())
