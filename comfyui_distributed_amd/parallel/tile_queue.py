@@ -11,8 +11,9 @@ State per job (namespaced by job id):
   total            number of tasks
   next             atomic pull cursor (store.add)
   requeue          JSON list of task ids pushed back by the monitor
-  assigned:<idx>   rank that pulled task idx
-  done:<idx>       completion marker
+  alist:<rank>     JSON list of task ids rank pulled (written only by rank)
+  dlist:<rank>     JSON list of task ids rank completed
+  done:<idx>       completion marker (idempotence across requeue races)
   completed        atomic completion counter
   hb:<rank>        wall-clock heartbeat stamp
 """
@@ -121,8 +122,19 @@ class TileQueue:
                     return None
             else:
                 idx = cand
-        self.store.set(self._k(f"assigned:{idx}"), str(self.rank))
+        self._append_own(f"alist:{self.rank}", idx)
         return idx
+
+    def _append_own(self, name: str, idx: int) -> None:
+        """Append to a per-rank JSON list. Only THIS rank ever writes its
+        own list (sequentially), so a plain read-modify-write is safe — no
+        CAS loop, and the monitor's assigned_incomplete becomes two store
+        reads instead of one per task in the job (round-1 O(total) scan)."""
+        key = self._k(name)
+        cur = self._get(key)
+        lst = json.loads(cur) if cur else []
+        lst.append(int(idx))
+        self.store.set(key, json.dumps(lst))
 
     def _pop_requeue(self) -> int | None:
         key = self._k("requeue")
@@ -140,6 +152,7 @@ class TileQueue:
     def mark_done(self, idx: int) -> int:
         """Mark a task complete; returns the completion count. Idempotent
         across requeue races (only the first completion counts)."""
+        self._append_own(f"dlist:{self.rank}", idx)
         n = self.store.add(self._k(f"done:{idx}"), 1)
         if n == 1:
             return self.store.add(self._k("completed"), 1)
@@ -176,13 +189,24 @@ class TileQueue:
         return now - float(raw)
 
     def assigned_incomplete(self, rank: int) -> list[int]:
+        """Task ids ``rank`` pulled but never completed. Two store reads
+        plus one done-check per *candidate* (a requeued task the rank lost
+        may have been finished elsewhere — skip those)."""
+        raw_a = self._get(self._k(f"alist:{rank}"))
+        raw_d = self._get(self._k(f"dlist:{rank}"))
+        assigned = json.loads(raw_a) if raw_a else []
+        done = set(json.loads(raw_d)) if raw_d else set()
         out = []
-        for idx in range(self.total()):
-            raw_done = self._get(self._k(f"done:{idx}"))
-            if raw_done and int(raw_done) > 0:
+        for idx in assigned:
+            if idx in done:
                 continue
-            if self._get(self._k(f"assigned:{idx}")) == str(rank).encode():
-                out.append(idx)
+            # add(key, 0) is the non-creating-as-"" existence probe: _get's
+            # compare_set would create the key as an EMPTY string, which a
+            # later mark_done add() rejects server-side (stoll) and the
+            # TCPStore then drops the connection
+            if self.store.add(self._k(f"done:{idx}"), 0) > 0:
+                continue
+            out.append(idx)
         return out
 
 
@@ -206,6 +230,23 @@ class TileScheduler:
         self.timeout = timeout if timeout is not None else constants.ACTOR_HEARTBEAT_TIMEOUT
         self.probe = probe  # callable rank -> bool ("is the worker busy/alive")
         self.clock = clock
+        #: job-start grace origin: a rank that NEVER heartbeats (crashed
+        #: before its first chunk) is aged from here instead of being
+        #: skipped forever (round-1 advisor finding)
+        self._started = clock()
+
+    def force_drop(self, rank: int) -> list[int]:
+        """Immediately requeue a rank's incomplete work and drop it (used
+        when its result transfer failed — no probe/grace)."""
+        if rank not in self.active:
+            return []
+        tasks = self.queue.assigned_incomplete(rank)
+        if tasks:
+            self.queue.requeue_tasks(tasks)
+        self.active.discard(rank)
+        self.dropped.add(rank)
+        log(f"scheduler: rank {rank} force-dropped — requeued {len(tasks)} tasks")
+        return tasks
 
     def check_and_requeue(self) -> list[int]:
         """Run one monitor pass; returns task ids requeued this pass."""
@@ -213,7 +254,9 @@ class TileScheduler:
         requeued: list[int] = []
         for rank in sorted(self.active):
             age = self.queue.heartbeat_age(rank, now)
-            if age is None or age <= self.timeout:
+            if age is None:
+                age = now - self._started
+            if age <= self.timeout:
                 continue
             # probe outside any lock (reference job_timeout.py:53-56)
             if self.probe is not None and self.probe(rank):

@@ -1,23 +1,39 @@
-"""Distributed USDU: tile pull-queue over all ranks, RCCL result gather,
-canonical blend on rank 0.
+"""Distributed USDU: tile pull-queue over all ranks, streamed RCCL result
+transport, canonical blend on rank 0 — with live fault tolerance.
 
 This is the intra-node re-architecture of the reference's static mode
 (upscale/modes/static.py): the HTTP request_image/submit_tiles loop becomes
-a TCPStore pull queue + one batched p2p gather over xGMI; the master
+a TCPStore pull queue + per-chunk p2p sends over xGMI; the master
 participates exactly like the reference master does (pulls tiles from its
-own queue). Determinism: extraction reads the original canvas and blending
-runs in ascending (tile, batch) order on rank 0, so the result is
-bit-identical to the single-GPU run regardless of tile assignment.
+own queue) while ALSO running the monitor loop — heartbeat age -> grace ->
+requeue -> drop (reference upscale/job_timeout.py:17-150) — and finally
+takes over any tile whose result never arrived (reference master-takeover,
+upscale/modes/static.py:354-363,469-513). Determinism: extraction reads the
+original canvas and blending runs in ascending (tile, batch) order on
+rank 0, so the result is bit-identical to the single-GPU run regardless of
+tile assignment, requeues, or duplicated work after a crash.
 """
 
 from __future__ import annotations
 
+import time
+
 import torch
 
 from ..engine.usdu import USDUParams, blend_results, plan_for_image, sample_tiles
-from ..utils.logging import debug_log
-from .dist import DistContext, broadcast_tensor, gather_tensor_lists
-from .tile_queue import TileQueue
+from ..utils.logging import debug_log, log
+from .dist import DistContext, ResultMailbox, broadcast_tensor
+from .tile_queue import TileQueue, TileScheduler
+
+
+def _pop_chunk(queue: TileQueue, n: int) -> list[int]:
+    ids: list[int] = []
+    while len(ids) < n:
+        idx = queue.pop()
+        if idx is None:
+            break
+        ids.append(idx)
+    return ids
 
 
 def run_distributed_usdu(
@@ -30,11 +46,20 @@ def run_distributed_usdu(
     image: torch.Tensor | None,
     job_id: str = "usdu",
     broadcast_input: bool = False,
+    scheduler_timeout: float | None = None,
+    recv_timeout: float = 30.0,
 ) -> torch.Tensor | None:
     """All ranks call this with the same input image (constructed
     identically per rank), or — with ``broadcast_input=True`` — only rank 0
     provides it and it is broadcast over RCCL/xGMI to the others. Returns
-    the blended canvas on rank 0, None on workers."""
+    the blended canvas on rank 0, None on workers.
+
+    Fault tolerance (rank > 0 crash at ANY point after the job starts):
+    the master's scheduler requeues a silent rank's unfinished tiles, the
+    mailbox bounds every receive, and tiles whose results were lost in
+    flight (marked done but never delivered) are reprocessed locally before
+    the blend. A worker crash therefore costs time, never correctness.
+    """
     if broadcast_input:
         image = broadcast_tensor(ctx, image if ctx.is_master else None)
     assert image is not None, "non-master ranks need broadcast_input=True"
@@ -47,40 +72,88 @@ def run_distributed_usdu(
     queue = TileQueue(store, job_id, ctx.rank)
     if ctx.is_master:
         queue.init_job(len(plans))
-    ctx.barrier()
+    else:
+        # job-ready poll instead of a barrier: one fewer collective that a
+        # dead rank could hang (reference workers poll job_status the same
+        # way, upscale/modes/static.py:33-47)
+        deadline = time.monotonic() + recv_timeout
+        while not queue.is_ready():
+            if time.monotonic() > deadline:
+                raise TimeoutError(f"job {job_id} never became ready")
+            time.sleep(0.01)
+    queue.heartbeat()  # initial stamp: a rank is monitorable from the start
 
-    tensors: list[torch.Tensor] = []
-    meta: list[tuple[int, int]] = []
-    done = 0
     # pop enough tile ids per iteration to fill the sampler batch
     # (each tile id covers all B batch images)
     ids_per_iter = max(1, params.tile_batch // max(B, 1))
-    while True:
-        ids: list[int] = []
-        while len(ids) < ids_per_iter:
-            idx = queue.pop()
-            if idx is None:
-                break
-            ids.append(idx)
-        if not ids:
-            break
-        res = sample_tiles(stack, cond, uncond, params, canvas, plans, ids)
-        for (t, b), img in sorted(res.items()):
-            tensors.append(img[0])
-            meta.append((t, b))
-        for idx in ids:
-            queue.mark_done(idx)
-        queue.heartbeat()
-        done += len(ids)
-    debug_log(f"rank {ctx.rank}: processed {done} tiles")
+    mailbox = ResultMailbox(ctx, store, job_id, recv_timeout=recv_timeout)
 
-    gathered = gather_tensor_lists(ctx, tensors, meta)
     if not ctx.is_master:
+        done = 0
+        while True:
+            ids = _pop_chunk(queue, ids_per_iter)
+            if not ids:
+                break
+            res = sample_tiles(stack, cond, uncond, params, canvas, plans, ids)
+            items = sorted(res.items())
+            # post the chunk's sends BEFORE marking done: a crash between
+            # the two leaves tiles un-done -> the scheduler requeues them
+            mailbox.send_chunk([img[0] for (_tb, img) in items],
+                               [tb for (tb, _img) in items])
+            for idx in ids:
+                queue.mark_done(idx)
+            queue.heartbeat()
+            done += len(ids)
+        mailbox.finish()
+        queue.heartbeat()
+        debug_log(f"rank {ctx.rank}: processed {done} tiles")
         return None
-    all_tensors, all_meta = gathered
-    results = {
-        (int(t), int(b)): tensor[None].float().to(canvas.device)
-        for tensor, (t, b) in zip(all_tensors, all_meta)
-    }
+
+    # ---- master: process + drain + monitor --------------------------------
+    workers = list(range(1, ctx.world_size))
+    scheduler = TileScheduler(queue, workers, timeout=scheduler_timeout)
+    results: dict[tuple[int, int], torch.Tensor] = {}
+
+    def _absorb(tensors, meta):
+        for tensor, (t, b) in zip(tensors, meta):
+            results[(int(t), int(b))] = tensor[None].float().to(canvas.device)
+
+    done = 0
+    while True:
+        ids = _pop_chunk(queue, ids_per_iter)
+        if ids:
+            res = sample_tiles(stack, cond, uncond, params, canvas, plans, ids)
+            _absorb([img[0] for (_tb, img) in sorted(res.items())],
+                    [tb for (tb, _img) in sorted(res.items())])
+            for idx in ids:
+                queue.mark_done(idx)
+            queue.heartbeat()
+            done += len(ids)
+        _absorb(*mailbox.drain())
+        for r in list(mailbox.failed_ranks):
+            scheduler.force_drop(r)
+        scheduler.check_and_requeue()
+        if not ids:
+            if all(r in scheduler.dropped or mailbox.rank_finished(r)
+                   for r in workers):
+                _absorb(*mailbox.drain())
+                break
+            time.sleep(0.02)
+    debug_log(f"rank 0: processed {done} tiles locally")
+
+    # ---- takeover: reprocess anything whose result never arrived ----------
+    missing_tiles = sorted({
+        t for t in range(len(plans))
+        if any((t, b) not in results for b in range(B))
+    })
+    if missing_tiles:
+        log(f"master takeover: {len(missing_tiles)} tiles missing — "
+            "reprocessing locally")
+    for i in range(0, len(missing_tiles), ids_per_iter):
+        ids = missing_tiles[i : i + ids_per_iter]
+        res = sample_tiles(stack, cond, uncond, params, canvas, plans, ids)
+        _absorb([img[0] for (_tb, img) in sorted(res.items())],
+                [tb for (tb, _img) in sorted(res.items())])
+
     blend_results(canvas, results, plans, params)
     return canvas
