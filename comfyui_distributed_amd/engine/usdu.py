@@ -26,6 +26,45 @@ from ..ops import dispatch as ops
 from ..utils import usdu_math
 from ..utils.trace import trace_range
 
+# --- wall-gap diagnosis (DISTGPU_PHASE_TIMING=1): host wall per phase with
+# device syncs at the boundaries; prints a summary every flush() -------------
+import os as _os
+import time as _time
+
+
+class _PhaseTimer:
+    enabled = _os.environ.get("DISTGPU_PHASE_TIMING", "") == "1"
+
+    def __init__(self):
+        self.acc: dict[str, float] = {}
+        self._t0 = None
+        self._cur = None
+
+    def mark(self, phase: str | None):
+        if not self.enabled:
+            return
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        now = _time.perf_counter()
+        if self._cur is not None:
+            self.acc[self._cur] = self.acc.get(self._cur, 0.0) + (now - self._t0)
+        self._cur = phase
+        self._t0 = now
+
+    def flush(self, label: str = ""):
+        if not self.enabled or not self.acc:
+            return
+        self.mark(None)
+        total = sum(self.acc.values())
+        parts = " ".join(f"{k}={v*1e3:.0f}ms" for k, v in
+                         sorted(self.acc.items(), key=lambda kv: -kv[1]))
+        print(f"[phase-timing]{label} total={total*1e3:.0f}ms {parts}",
+              flush=True)
+        self.acc.clear()
+
+
+phase_timer = _PhaseTimer()
+
 
 @dataclass
 class USDUParams:
@@ -161,6 +200,7 @@ def sample_tiles(
                                  chunk_uncond, params.cfg)
             )
         # ---- extract + resample each (tile, batch) crop to process size ----
+        phase_timer.mark("extract")
         crops = []
         for t, b in chunk:
             plan = plans[t]
@@ -170,8 +210,10 @@ def sample_tiles(
         batch_img = torch.cat(crops, dim=0)
         # ---- encode -> img2img sample -> decode ----
         with torch.no_grad():
+            phase_timer.mark("vae_encode")
             with trace_range("usdu.vae_encode"):
                 latents = stack.vae.encode(batch_img)
+            phase_timer.mark("noise")
             noise = torch.stack(
                 [
                     _tile_noise(params.seed, t, b + batch_offset, latents.shape[1:])
@@ -179,6 +221,7 @@ def sample_tiles(
                 ]
             ).to(latents.device)
             plan0 = plans[chunk[0][0]]
+            phase_timer.mark("sample")
             with trace_range("usdu.sample"), crop_model_patches(
                 stack, plan0.crop_region, plan0.canvas_size, plan0.process_size
             ):
@@ -203,13 +246,16 @@ def sample_tiles(
                         seed=params.seed,
                         start_from_latent=latents.float(),
                     )
+            phase_timer.mark("vae_decode")
             with trace_range("usdu.vae_decode"):
                 if params.tiled_decode:
                     out_img = stack.vae.decode_tiled(latent_out.to(stack.dtype))
                 else:
                     out_img = stack.vae.decode(latent_out.to(stack.dtype))
+        phase_timer.mark("collect")
         for j, (t, b) in enumerate(chunk):
             results[(t, b)] = out_img[j : j + 1].float()
+        phase_timer.mark(None)
     return results
 
 
@@ -222,8 +268,10 @@ def blend_results(
     """One canonical blend pass: ascending (tile_idx, batch_idx) — the
     deterministic order the reference enforces for worker tiles
     (upscale/modes/static.py:521-527)."""
+    phase_timer.mark("blend")
     for (t, b) in sorted(results.keys()):
         blend_processed_tile(canvas, results[(t, b)], plans[t], params, batch_index=b)
+    phase_timer.mark(None)
 
 
 def process_tiles(

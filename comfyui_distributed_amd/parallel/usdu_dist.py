@@ -156,4 +156,6 @@ def run_distributed_usdu(
                 [tb for (tb, _img) in sorted(res.items())])
 
     blend_results(canvas, results, plans, params)
+    from ..engine.usdu import phase_timer
+    phase_timer.flush(f" job={job_id}")
     return canvas

@@ -58,7 +58,7 @@ def bench_attn(iters):
 
 
 def bench_gn(iters):
-    print("== fused GroupNorm+SiLU (bf16) ==")
+    print("== fused GroupNorm+SiLU NCHW (bf16) ==")
     from comfyui_distributed_amd.ops import ext
 
     mod = ext.get_ext(True)
@@ -76,6 +76,32 @@ def bench_gn(iters):
         te = timeit(lambda: F.silu(F.group_norm(xf.float(), 32, w, b, 1e-5)).to(torch.bfloat16), iters)
         print(f"{str(shape):24s} {t*1e3:7.3f} ms  {gb/t:7.0f} GB/s   "
               f"eager {te*1e3:7.3f} ms ({te/t:4.1f}x)")
+
+
+def bench_gn_nhwc(iters):
+    print("== fused GroupNorm+SiLU NHWC/channels-last (bf16) — the hot one ==")
+    import torch.nn.functional as F
+
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    # (B, H, W, C): UNet trunk levels at tile_batch=16, VAE decode stages
+    for shape in [(16, 68, 68, 320), (16, 34, 34, 640), (16, 17, 17, 1280),
+                  (16, 136, 136, 512), (16, 272, 272, 512),
+                  (16, 544, 544, 128), (1, 1088, 1088, 128)]:
+        x = torch.randn(*shape, device="cuda", dtype=torch.bfloat16)
+        C = shape[3]
+        w = torch.randn(C, device="cuda")
+        b = torch.randn(C, device="cuda")
+        y = mod.group_norm_nhwc(x, 32, w, b, 1e-5, True)
+        # reference: fp32 NCHW group_norm on the same data
+        xf = x.permute(0, 3, 1, 2).float()
+        ref = F.silu(F.group_norm(xf, 32, w, b, 1e-5)).permute(0, 2, 3, 1)
+        err = (y.float() - ref).abs().max().item()
+        t = timeit(lambda: mod.group_norm_nhwc(x, 32, w, b, 1e-5, True), iters)
+        gb = 3 * x.numel() * 2 / 1e9  # 2 reads + 1 write bf16
+        print(f"{str(shape):24s} {t*1e3:7.3f} ms  {gb/t:7.0f} GB/s  "
+              f"maxerr {err:.4f}")
 
 
 def bench_conv(iters):
@@ -131,6 +157,8 @@ def main():
         bench_attn(args.iters)
     if args.op in ("gn", "all"):
         bench_gn(args.iters)
+    if args.op in ("gn_nhwc", "gn", "all"):
+        bench_gn_nhwc(args.iters)
     if args.op in ("conv", "all"):
         bench_conv(args.iters)
     if args.op in ("tiles", "all"):
