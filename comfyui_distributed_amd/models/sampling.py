@@ -170,6 +170,213 @@ class GraphedModel:
         return (sx, st, sc, sy, sout, g)
 
 
+class GraphedSamplerLoop:
+    """Captures an ENTIRE deterministic sampler loop — every UNet forward,
+    the CFG combine and the sampler update math — as ONE hipGraph.
+
+    Round-1 profiling (profiles/r01_prof_final_summary.csv header) showed a
+    27% wall-over-kernel gap: per-step eager math (timestep interpolation,
+    CFG cat/chunk, the x update) plus per-replay launch overhead dominate
+    between the UNet graphs. One replay per CHUNK removes all of it. Sigma
+    and timestep values are baked in as constants — the cache key includes
+    the full sigma tuple — and conditioning/noise/latent are static
+    copy-in buffers. Only deterministic samplers (no per-step randn) are
+    captured; ancestral/SDE run the eager path.
+    """
+
+    SUPPORTED = ("euler", "heun", "dpm_2", "dpmpp_2m")
+
+    def __init__(self, model, schedule: NoiseSchedule):
+        self.model = model  # the RAW UNet module (not GraphedModel)
+        self.schedule = schedule
+        self.graphs: dict = {}
+        self._failed = False
+
+    def _denoise_const(self, x, i, entry_consts, s_ctx, s_y, cfg, need_cfg,
+                       sig, t_tensor):
+        """One CFG denoise with python-float sigma (mirrors CFGDenoiser)."""
+        c_in = 1.0 / math.sqrt(1.0 + sig * sig)
+        if need_cfg:
+            x_in = torch.cat([x * c_in] * 2)
+            eps = (self.model(x_in, t_tensor, s_ctx, y=s_y)
+                   if s_y is not None else self.model(x_in, t_tensor, s_ctx))
+            eps_c, eps_u = eps.chunk(2)
+            eps = eps_u + cfg * (eps_c - eps_u)
+        else:
+            eps = (self.model(x * c_in, t_tensor, s_ctx, y=s_y)
+                   if s_y is not None else self.model(x * c_in, t_tensor, s_ctx))
+        return x - eps.float() * sig
+
+    def _loop_body(self, sampler, x, sigs, denoise_at):
+        """The sampler recurrence with float sigmas; ``denoise_at(x, i,
+        sig)`` runs the model at schedule position i."""
+        n = len(sigs) - 1
+        if sampler == "euler":
+            for i in range(n):
+                sig, nxt = sigs[i], sigs[i + 1]
+                denoised = denoise_at(x, i, sig)
+                d = (x - denoised) / sig
+                x = x + d * (nxt - sig)
+            return x
+        if sampler == "heun":
+            for i in range(n):
+                sig, nxt = sigs[i], sigs[i + 1]
+                denoised = denoise_at(x, i, sig)
+                d = (x - denoised) / sig
+                dt = nxt - sig
+                if nxt == 0:
+                    x = x + d * dt
+                    continue
+                x2 = x + d * dt
+                denoised2 = denoise_at(x2, i, nxt, aux=True)
+                d2 = (x2 - denoised2) / nxt
+                x = x + (d + d2) / 2 * dt
+            return x
+        if sampler == "dpm_2":
+            for i in range(n):
+                sig, nxt = sigs[i], sigs[i + 1]
+                denoised = denoise_at(x, i, sig)
+                d = (x - denoised) / sig
+                if nxt == 0:
+                    x = x + d * (nxt - sig)
+                    continue
+                sig_mid = math.exp(
+                    math.log(sig) + (math.log(nxt) - math.log(sig)) * 0.5)
+                x2 = x + d * (sig_mid - sig)
+                denoised2 = denoise_at(x2, i, sig_mid, aux=True)
+                d2 = (x2 - denoised2) / sig_mid
+                x = x + d2 * (nxt - sig)
+            return x
+        if sampler == "dpmpp_2m":
+            old_denoised = None
+            for i in range(n):
+                sig, nxt = sigs[i], sigs[i + 1]
+                denoised = denoise_at(x, i, sig)
+                if nxt == 0:
+                    x = denoised
+                elif old_denoised is None:
+                    h = -math.log(nxt) + math.log(sig)
+                    x = (nxt / sig) * x - (math.expm1(-h)) * denoised
+                else:
+                    h = -math.log(nxt) + math.log(sig)
+                    h_last = -math.log(sig) + math.log(sigs[i - 1])
+                    r = h_last / h
+                    denoised_d = (1 + 1 / (2 * r)) * denoised - \
+                        (1 / (2 * r)) * old_denoised
+                    x = (nxt / sig) * x - (math.expm1(-h)) * denoised_d
+                old_denoised = denoised
+            return x
+        raise ValueError(sampler)
+
+    def run(self, cond, uncond, cfg_scale, noise, sigmas, sampler,
+            start_from_latent):
+        """Returns the sampled latent, or None when this config can't be
+        captured (caller falls back to eager)."""
+        if self._failed or sampler not in self.SUPPORTED or not noise.is_cuda:
+            return None
+        b = noise.shape[0]
+        need_cfg = cfg_scale != 1.0 and uncond is not None
+        sigs = [float(s) for s in sigmas]
+        ctx_c = cond["context"]
+        y_c = cond.get("y")
+        key = (
+            sampler, tuple(noise.shape), tuple(sigs), bool(need_cfg),
+            float(cfg_scale), tuple(ctx_c.shape),
+            None if y_c is None else tuple(y_c.shape),
+            start_from_latent is not None,
+        )
+        entry = self.graphs.get(key)
+        if entry is None:
+            try:
+                entry = self._capture(key, cond, uncond, cfg_scale, noise,
+                                      sigs, sampler, start_from_latent)
+            except Exception as exc:  # noqa: BLE001
+                import warnings
+
+                warnings.warn(
+                    f"sampler-loop hipGraph capture failed, eager: {exc!r}")
+                self._failed = True
+                return None
+            self.graphs[key] = entry
+        (s_noise, s_lat, s_ctx, s_y, s_out, g) = entry
+        s_noise.copy_(noise)
+        if s_lat is not None:
+            s_lat.copy_(start_from_latent)
+        s_ctx.copy_(self._cat_ctx(cond, uncond, b,
+                                  cfg_scale != 1.0 and uncond is not None))
+        if s_y is not None:
+            s_y.copy_(self._cat_y(cond, uncond, b,
+                                  cfg_scale != 1.0 and uncond is not None))
+        g.replay()
+        # clone: the static output buffer is overwritten by the next replay
+        return s_out.clone()
+
+    def _cat_ctx(self, cond, uncond, b, need_cfg):
+        ctx = cond["context"].expand(b, -1, -1)
+        if need_cfg:
+            ctx = torch.cat([ctx, uncond["context"].expand(b, -1, -1)])
+        return ctx
+
+    def _cat_y(self, cond, uncond, b, need_cfg):
+        y = cond.get("y")
+        if y is None:
+            return None
+        y = y.expand(b, -1)
+        if need_cfg:
+            y = torch.cat([y, uncond["y"].expand(b, -1)])
+        return y
+
+    def _capture(self, key, cond, uncond, cfg_scale, noise, sigs, sampler,
+                 start_from_latent):
+        b = noise.shape[0]
+        need_cfg = cfg_scale != 1.0 and uncond is not None
+        s_noise = noise.detach().float().clone()
+        s_lat = (None if start_from_latent is None
+                 else start_from_latent.detach().float().clone())
+        s_ctx = self._cat_ctx(cond, uncond, b, need_cfg).detach().clone()
+        s_y = self._cat_y(cond, uncond, b, need_cfg)
+        s_y = None if s_y is None else s_y.detach().clone()
+        # timestep tensors per schedule position: constants of the capture
+        dev = noise.device
+        tb = 2 * b if need_cfg else b
+        sig_t = torch.tensor(sigs[:-1], dtype=torch.float32)
+        t_vals = self.schedule.timestep(sig_t)
+        t_ts = [torch.full((tb,), float(t), device=dev) for t in t_vals]
+        # heun/dpm_2 evaluate at a second sigma inside the step; map those
+        # to their own timestep tensors lazily
+        aux_cache: dict[float, torch.Tensor] = {}
+
+        def t_for(sig_val: float):
+            t = aux_cache.get(sig_val)
+            if t is None:
+                tv = self.schedule.timestep(
+                    torch.tensor([sig_val], dtype=torch.float32))
+                t = torch.full((tb,), float(tv[0]), device=dev)
+                aux_cache[sig_val] = t
+            return t
+
+        def denoise_at(x, i, sig, aux=False):
+            t_tensor = t_for(sig) if aux else t_ts[i]
+            return self._denoise_const(x, i, None, s_ctx, s_y, cfg_scale,
+                                       need_cfg, sig, t_tensor)
+
+        def fwd():
+            x = (s_lat + s_noise * sigs[0] if s_lat is not None
+                 else s_noise * sigs[0])
+            return self._loop_body(sampler, x, sigs, denoise_at)
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                fwd()
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            s_out = fwd()
+        return (s_noise, s_lat, s_ctx, s_y, s_out, g)
+
+
 class CFGDenoiser:
     """eps-model + classifier-free guidance -> denoised prediction x0."""
 
@@ -275,6 +482,31 @@ def sample(denoiser, noise_or_latent: torch.Tensor, sigmas: torch.Tensor,
     if seed is not None:
         gen = torch.Generator(device="cpu").manual_seed(seed)
     from ..nodes.runtime import get_runtime
+
+    # whole-loop hipGraph fast path: deterministic samplers on GPU with a
+    # plain CFGDenoiser replay the full loop as one graph (interrupt is
+    # checked once per call — per-chunk abort granularity)
+    import os as _os
+
+    if (
+        isinstance(denoiser, CFGDenoiser)
+        and denoise_mask is None
+        and noise_or_latent.is_cuda
+        and sampler in GraphedSamplerLoop.SUPPORTED
+        and _os.environ.get("DISTGPU_GRAPH_LOOP", "1") == "1"
+    ):
+        get_runtime().throw_if_interrupted()
+        raw = getattr(denoiser.unet, "model", denoiser.unet)
+        loop = getattr(raw, "_loop_graph", None)
+        if loop is None or loop.model is not raw:
+            loop = GraphedSamplerLoop(raw, denoiser.schedule)
+            object.__setattr__(raw, "_loop_graph", loop)
+        out = loop.run(denoiser.cond, denoiser.uncond, denoiser.cfg_scale,
+                       noise_or_latent.float(), sigmas, sampler,
+                       None if start_from_latent is None
+                       else start_from_latent.float())
+        if out is not None:
+            return out
 
     denoiser = _InterruptibleDenoiser(denoiser, get_runtime())
     if sampler == "euler":

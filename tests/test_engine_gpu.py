@@ -110,3 +110,38 @@ def test_flux_usdu_tile_gpu():
     out = process_single_gpu(stack, cond, None, p, img)
     assert out.shape == (1, 128, 128, 3)
     assert torch.isfinite(out).all()
+
+
+def test_graphed_sampler_loop_matches_eager():
+    """The whole-loop hipGraph fast path must match the eager sampler loop
+    for every captured sampler (same seeds, same conditioning)."""
+    import os
+
+    from comfyui_distributed_amd.engine.usdu import (
+        USDUParams,
+        plan_for_image,
+        sample_tiles,
+    )
+    from comfyui_distributed_amd.models import create_diffusion_stack
+
+    stack = create_diffusion_stack("tiny", device="cuda:0",
+                                   dtype=torch.bfloat16, seed=3)
+    cond = stack.make_conditioning(0)
+    uncond = stack.make_conditioning(1)
+    canvas = torch.rand(1, 128, 128, 3, device="cuda:0",
+                        generator=torch.Generator("cuda:0").manual_seed(7))
+    for sampler in ("euler", "heun", "dpm_2", "dpmpp_2m"):
+        p = USDUParams(seed=1, steps=3, cfg=7.0, denoise=0.6, tile_width=64,
+                       tile_height=64, padding=16, mask_blur=4, tile_batch=4,
+                       sampler_name=sampler)
+        plans = plan_for_image(128, 128, p)
+        os.environ["DISTGPU_GRAPH_LOOP"] = "0"
+        try:
+            eager = sample_tiles(stack, cond, uncond, p, canvas, plans, [0, 1])
+        finally:
+            os.environ["DISTGPU_GRAPH_LOOP"] = "1"
+        graphed = sample_tiles(stack, cond, uncond, p, canvas, plans, [0, 1])
+        for key in eager:
+            a, b = eager[key], graphed[key]
+            err = (a - b).abs().max().item()
+            assert err < 0.05, f"{sampler} {key}: graphed vs eager err {err}"
