@@ -57,6 +57,31 @@ def timestep_embedding(t: torch.Tensor, dim: int, max_period: float = 10000.0):
     return torch.cat([torch.cos(args), torch.sin(args)], dim=-1).to(t.dtype)
 
 
+class MfmaLinear(nn.Linear):
+    """nn.Linear routed through the 256-tile MFMA GEMM on GPU (gemm.hip);
+    parameter layout/init identical to nn.Linear so seeded CPU/GPU model
+    construction stays in lockstep."""
+
+    def forward(self, x):
+        return ops.linear_mfma(x, self.weight, self.bias)
+
+
+class MfmaConv2d(nn.Conv2d):
+    """nn.Conv2d routed through the implicit-GEMM MFMA kernels for the
+    supported shapes (stride 1, 3x3 pad1 / 1x1, C%32==0); MIOpen
+    otherwise. Replaces the round-1 MIOpen igemm path (~33% of flagship
+    kernel time, profiles/r01_prof_final_summary.csv)."""
+
+    def forward(self, x):
+        if (
+            x.is_cuda
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and ops.conv_supported(self)
+        ):
+            return ops.conv2d_mfma(x, self)
+        return super().forward(x)
+
+
 class FusedGroupNorm(nn.Module):
     """GroupNorm with optional fused SiLU via the HIP kernel."""
 
@@ -88,12 +113,12 @@ class ResBlock(nn.Module):
         super().__init__()
         out_channels = out_channels or channels
         self.norm1 = FusedGroupNorm(channels, silu=True)
-        self.conv1 = nn.Conv2d(channels, out_channels, 3, padding=1)
+        self.conv1 = MfmaConv2d(channels, out_channels, 3, padding=1)
         self.emb_proj = nn.Linear(emb_dim, out_channels)
         self.norm2 = FusedGroupNorm(out_channels, silu=True)
-        self.conv2 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        self.conv2 = MfmaConv2d(out_channels, out_channels, 3, padding=1)
         self.skip = (
-            nn.Conv2d(channels, out_channels, 1) if out_channels != channels else nn.Identity()
+            MfmaConv2d(channels, out_channels, 1) if out_channels != channels else nn.Identity()
         )
 
     def forward(self, x, emb):
@@ -110,10 +135,10 @@ class CrossAttention(nn.Module):
         context_dim = context_dim or dim
         self.heads = heads
         self.head_dim = head_dim
-        self.to_q = nn.Linear(dim, inner, bias=False)
-        self.to_k = nn.Linear(context_dim, inner, bias=False)
-        self.to_v = nn.Linear(context_dim, inner, bias=False)
-        self.to_out = nn.Linear(inner, dim)
+        self.to_q = MfmaLinear(dim, inner, bias=False)
+        self.to_k = MfmaLinear(context_dim, inner, bias=False)
+        self.to_v = MfmaLinear(context_dim, inner, bias=False)
+        self.to_out = MfmaLinear(inner, dim)
 
     def _fused_weight(self, name: str, parts: list[torch.Tensor]) -> torch.Tensor:
         cached = getattr(self, name, None)
@@ -130,11 +155,11 @@ class CrossAttention(nn.Module):
             w = self._fused_weight(
                 "_wqkv", [self.to_q.weight, self.to_k.weight, self.to_v.weight]
             )
-            o = ops.attention_qkv(F.linear(x, w), heads=self.heads)
+            o = ops.attention_qkv(ops.linear_mfma(x, w), heads=self.heads)
         else:  # cross-attention: fused KV
             w = self._fused_weight("_wkv", [self.to_k.weight, self.to_v.weight])
             o = ops.attention_q_kv(
-                self.to_q(x), F.linear(context, w), heads=self.heads
+                self.to_q(x), ops.linear_mfma(context, w), heads=self.heads
             )
         return self.to_out(o)
 
@@ -143,8 +168,8 @@ class GEGLUFeedForward(nn.Module):
     def __init__(self, dim: int, mult: int = 4):
         super().__init__()
         inner = dim * mult
-        self.proj_in = nn.Linear(dim, inner * 2)
-        self.proj_out = nn.Linear(inner, dim)
+        self.proj_in = MfmaLinear(dim, inner * 2)
+        self.proj_out = MfmaLinear(inner, dim)
 
     def forward(self, x):
         a, gate = self.proj_in(x).chunk(2, dim=-1)
@@ -173,11 +198,11 @@ class SpatialTransformer(nn.Module):
                  depth: int = 1):
         super().__init__()
         self.norm = FusedGroupNorm(channels, silu=False)
-        self.proj_in = nn.Linear(channels, channels)
+        self.proj_in = MfmaLinear(channels, channels)
         self.blocks = nn.ModuleList(
             [TransformerBlock(channels, context_dim, heads, head_dim) for _ in range(depth)]
         )
-        self.proj_out = nn.Linear(channels, channels)
+        self.proj_out = MfmaLinear(channels, channels)
 
     def forward(self, x, context):
         b, c, h, w = x.shape
@@ -217,7 +242,7 @@ class Downsample(nn.Module):
 class Upsample(nn.Module):
     def __init__(self, channels):
         super().__init__()
-        self.conv = nn.Conv2d(channels, channels, 3, padding=1)
+        self.conv = MfmaConv2d(channels, channels, 3, padding=1)
 
     def forward(self, x, output_shape=None):
         # odd latent sizes (e.g. 68->34->17->9) need the exact skip shape on

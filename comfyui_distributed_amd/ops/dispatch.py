@@ -10,6 +10,7 @@ the kernels (documented per-op).
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 import torch.nn.functional as F
@@ -113,9 +114,14 @@ def _repacked_weight(conv) -> torch.Tensor:
     return wt
 
 
+_CONV256 = os.environ.get("DISTGPU_CONV256", "1") == "1"
+_GEMM256 = os.environ.get("DISTGPU_GEMM256", "1") == "1"
+
+
 def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
     """x: NCHW tensor in channels_last memory format (bf16, on GPU) ->
-    same layout. Dispatches to the implicit-GEMM NHWC kernel."""
+    same layout. Dispatches to the implicit-GEMM NHWC kernel: the 256-tile
+    glds template (gemm.hip) when shapes allow, else the v1/v2 kernels."""
     assert x.is_cuda
     b, c, h, w = x.shape
     nhwc = x.permute(0, 2, 3, 1)  # view: contiguous when x is channels_last
@@ -124,11 +130,42 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
     wt = _repacked_weight(conv)
     rs = 9 if conv.kernel_size == (3, 3) else 1
     bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
+    if _CONV256 and c % 64 == 0 and b * h * w >= 256:
+        y = ext.get_ext(True).conv256_nhwc(
+            nhwc.to(torch.bfloat16), wt, bias, b, h, w, c,
+            conv.out_channels, rs, fuse_silu,
+        )
+        return y.permute(0, 3, 1, 2)
     y = ext.get_ext(True).conv_nhwc(
         nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
         fuse_silu,
     )
     return y.permute(0, 3, 1, 2)  # NCHW semantic, channels_last storage
+
+
+def linear_mfma(x: torch.Tensor, weight: torch.Tensor,
+                bias: torch.Tensor | None = None,
+                fuse_silu: bool = False) -> torch.Tensor:
+    """torch-Linear on the 256-tile MFMA GEMM when the shape fills it
+    (tokens >= 1024, K % 64 == 0); hipBLASLt otherwise. x [..., K]."""
+    K = x.shape[-1]
+    M = x.numel() // K
+    if (
+        _GEMM256
+        and _on_gpu(x)
+        and M >= 1024
+        and K % 64 == 0
+        and weight.shape[0] >= 64
+    ):
+        x2 = x.reshape(M, K)
+        if not x2.is_contiguous():
+            x2 = x2.contiguous()
+        b = bias if bias is not None else torch.empty(0, device=x.device)
+        y = ext.get_ext(True).gemm256_bf16(
+            x2.to(torch.bfloat16), weight.to(torch.bfloat16), b, fuse_silu)
+        return y.reshape(*x.shape[:-1], weight.shape[0])
+    y = F.linear(x, weight, bias)
+    return F.silu(y) if fuse_silu else y
 
 
 def conv2d_smallc(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:

@@ -33,6 +33,7 @@ SOURCES = [
     "tile_ops.hip",
     "mfma_selftest.hip",
     "conv.hip",
+    "gemm.hip",
 ]
 
 _ext = None

@@ -31,6 +31,12 @@ torch::Tensor group_norm_nhwc(torch::Tensor x, int64_t groups,
 torch::Tensor conv_smallc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
                           int64_t B, int64_t H, int64_t W, int64_t C,
                           int64_t K, int64_t rs, bool fuse_silu);
+torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
+                           torch::Tensor bias, bool fuse_silu);
+torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
+                           torch::Tensor bias, int64_t B, int64_t H,
+                           int64_t W, int64_t C, int64_t K, int64_t rs,
+                           bool fuse_silu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_fused", &group_norm_fused,
@@ -56,4 +62,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused GroupNorm(+SiLU), NHWC bf16");
   m.def("conv_smallc", &conv_smallc,
         "small-C NHWC conv (stem convs, C <= 8)");
+  m.def("gemm256_bf16", &gemm256_bf16,
+        "256x256x64 glds-staged MFMA GEMM, torch-Linear layout (+SiLU)");
+  m.def("conv256_nhwc", &conv256_nhwc,
+        "256-tile implicit-GEMM 3x3/1x1 NHWC conv on the same template");
 }

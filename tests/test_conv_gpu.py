@@ -99,3 +99,52 @@ def test_vae_decode_nhwc_path_consistent(extmod):
     err = (img_gpu - img_cpu).abs().max().item()
     assert err < 0.12, f"decode mismatch {err}"  # bf16 conv chain tolerance
     assert torch.isfinite(img_gpu).all()
+
+
+def test_gemm256_matches_fp32_linear():
+    """256-tile glds GEMM vs fp32 torch reference, incl. partial M/N tiles
+    and the SiLU fusion."""
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    torch.manual_seed(0)
+    for (M, N, K) in [(4624, 320, 320), (4624, 2560, 320), (1156, 640, 768),
+                      (300, 77, 128), (256, 256, 64)]:
+        x = (torch.randn(M, K, device="cuda") / 4).to(torch.bfloat16)
+        w = (torch.randn(N, K, device="cuda") / 4).to(torch.bfloat16)
+        b = torch.randn(N, device="cuda").to(torch.bfloat16)
+        y = mod.gemm256_bf16(x, w, b, False)
+        ref = torch.nn.functional.linear(x.float(), w.float(), b.float())
+        err = (y.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 0.02, f"gemm {M}x{N}x{K}: rel err {err/scale}"
+        # SiLU fusion
+        y2 = mod.gemm256_bf16(x, w, b, True)
+        ref2 = torch.nn.functional.silu(ref)
+        err2 = (y2.float() - ref2).abs().max().item()
+        assert err2 / max(scale, 1e-6) < 0.02
+
+
+def test_conv256_matches_fp32_conv():
+    """Implicit-GEMM 256-tile conv (3x3 pad1 + 1x1) vs fp32 reference —
+    exercises the zero-page OOB tap redirect at image borders."""
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    torch.manual_seed(1)
+    for (B, C, H, W, K, rs) in [(2, 64, 17, 19, 96, 9), (1, 128, 34, 34, 320, 9),
+                                (2, 320, 20, 20, 320, 9), (2, 64, 32, 32, 48, 1)]:
+        x = (torch.randn(B, H, W, C, device="cuda") / 4).to(torch.bfloat16)
+        wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
+        if rs == 1:
+            wkern = wkern[:, :, :1, :1].contiguous()
+        wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
+        b = torch.randn(K, device="cuda").to(torch.bfloat16)
+        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, rs, False)
+        xf = x.permute(0, 3, 1, 2).float()
+        ref = torch.nn.functional.conv2d(
+            xf, wkern.float(), b.float(), padding=1 if rs == 9 else 0)
+        ref = ref.permute(0, 2, 3, 1)
+        err = (y.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 0.02, f"conv {B}x{C}x{H}x{W}->{K} rs{rs}: {err/scale}"

@@ -132,6 +132,65 @@ def bench_conv(iters):
               f"| miopen-cl {t_cl*1e3:7.3f}")
 
 
+def bench_gemm(iters):
+    print("== gemm256 bf16 (glds 256-tile) vs hipBLASLt ==")
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    shapes = [
+        # (M, N, K) — UNet Linear hot shapes at tile_batch 16 (2x CFG batch)
+        (32 * 4624, 320, 320),    # qkv proj level0 (N=960 fused)
+        (32 * 4624, 960, 320),    # fused qkv
+        (32 * 4624, 2560, 320),   # GEGLU in
+        (32 * 4624, 320, 1280),   # GEGLU out
+        (32 * 1156, 1920, 640),
+        (32 * 289, 3840, 1280),
+        (8192, 8192, 8192),       # reference square
+    ]
+    for (M, N, K) in shapes:
+        x = (torch.randn(M, K, device="cuda") / 4).to(torch.bfloat16)
+        w = (torch.randn(N, K, device="cuda") / 4).to(torch.bfloat16)
+        b = torch.empty(0, device="cuda")
+        t = timeit(lambda: mod.gemm256_bf16(x, w, b, False), iters)
+        tl = timeit(lambda: torch.nn.functional.linear(x, w), iters)
+        fl = 2.0 * M * N * K
+        print(f"M{M} N{N} K{K}: ours {t*1e3:7.3f} ms ({fl/t/1e12:6.1f} TF) | "
+              f"blaslt {tl*1e3:7.3f} ms ({fl/tl/1e12:6.1f} TF)")
+
+
+def bench_conv256(iters):
+    print("== conv256 (glds template) vs conv v2 vs MIOpen-CL ==")
+    import os
+
+    from comfyui_distributed_amd.ops import dispatch
+
+    shapes = [
+        (32, 320, 68, 68, 320),
+        (32, 640, 34, 34, 640),
+        (32, 1280, 17, 17, 1280),
+        (16, 512, 136, 136, 512),
+        (16, 256, 272, 272, 256),
+        (16, 128, 544, 544, 128),
+    ]
+    for b, c, h, w, k in shapes:
+        conv = torch.nn.Conv2d(c, k, 3, padding=1).cuda().to(torch.bfloat16)
+        x = (torch.randn(b, c, h, w) / 4).cuda().to(torch.bfloat16)
+        xcl = x.contiguous(memory_format=torch.channels_last)
+        conv_cl = torch.nn.Conv2d(c, k, 3, padding=1).cuda().to(torch.bfloat16) \
+            .to(memory_format=torch.channels_last)
+        dispatch._CONV256 = True
+        t256 = timeit(lambda: dispatch.conv2d_mfma(xcl, conv), iters)
+        dispatch._CONV256 = False
+        tv2 = timeit(lambda: dispatch.conv2d_mfma(xcl, conv), iters)
+        dispatch._CONV256 = True
+        tmi = timeit(lambda: conv_cl(xcl), iters)
+        flops = 2.0 * b * h * w * k * c * 9
+        print(f"B{b} C{c} {h}x{w} K{k}: 256 {t256*1e3:7.3f} ms "
+              f"({flops/t256/1e12:6.1f} TF) | v2 {tv2*1e3:7.3f} "
+              f"({flops/tv2/1e12:6.1f} TF) | miopen-cl {tmi*1e3:7.3f} "
+              f"({flops/tmi/1e12:6.1f} TF)")
+
+
 def bench_tiles(iters):
     print("== tile ops (f32) ==")
     from comfyui_distributed_amd.ops import ext
@@ -161,6 +220,10 @@ def main():
         bench_gn_nhwc(args.iters)
     if args.op in ("conv", "all"):
         bench_conv(args.iters)
+    if args.op in ("gemm", "all"):
+        bench_gemm(args.iters)
+    if args.op in ("conv256", "all"):
+        bench_conv256(args.iters)
     if args.op in ("tiles", "all"):
         bench_tiles(args.iters)
 
