@@ -115,7 +115,10 @@ def _repacked_weight(conv) -> torch.Tensor:
 
 
 _CONV256 = os.environ.get("DISTGPU_CONV256", "1") == "1"
-_GEMM256 = os.environ.get("DISTGPU_GEMM256", "1") == "1"
+# hipBLASLt wins the plain-Linear shapes (measured gpurun_out/call3: 368-1614
+# TF vs our 205-1009) — the 256-tile GEMM stays available for fused uses and
+# future skinny-N tiles but is opt-in for nn.Linear routing
+_GEMM256 = os.environ.get("DISTGPU_GEMM256", "0") == "1"
 
 
 def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:

@@ -57,7 +57,20 @@ __device__ __forceinline__ void glds16(const void* gsrc, void* lds_dst) {
 struct ConvGeo {
   int H, W, C;     // input spatial + channels
   int RS;          // 9 for 3x3 (pad 1), 1 for 1x1
+  int tile2d;      // 1: blocks cover 16x16 pixel tiles (H,W % 16 == 0) —
+                   // a 3x3 tap re-reads an 18x18 halo (1.27x) instead of
+                   // the scanline tile's 9x; the VAE-decode C=128/256
+                   // shapes are HBM-bound on exactly that re-read
+  int xcd_swz;     // 1: bijective blockIdx.x -> XCD-major remap (T1)
 };
+
+// bijective XCD-aware remap (guide §5 "XCD swizzle must be bijective"):
+// consecutive original ids spread across the 8 XCDs' L2s
+__device__ __forceinline__ unsigned xcd_remap(unsigned id, unsigned nwg) {
+  const unsigned q = nwg / 8u, r = nwg % 8u;
+  const unsigned xcd = id % 8u, pos = id / 8u;
+  return (xcd < r ? xcd * (q + 1u) : r * (q + 1u) + (xcd - r) * q) + pos;
+}
 
 // Per-thread precomputed source descriptors for the 8 glds chunk slots
 // (2 instrs x 2 halves x {A,B}).
@@ -84,7 +97,20 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
   // 128 KiB total (A images then B images), byte offsets per the macros
   __shared__ __align__(16) uint16_t lds[65536];
 
-  const long long m_blk = (long long)blockIdx.x * GBM;
+  unsigned bx = blockIdx.x;
+  if (IS_CONV && geo.xcd_swz) bx = xcd_remap(bx, gridDim.x);
+  long long m_blk = 0;
+  int t_b = 0, t_py0 = 0, t_px0 = 0;
+  if (IS_CONV && geo.tile2d) {
+    const int tx_n = geo.W >> 4;
+    const int per_img = tx_n * (geo.H >> 4);
+    t_b = (int)(bx / per_img);
+    const int rem = (int)(bx % per_img);
+    t_py0 = (rem / tx_n) << 4;
+    t_px0 = (rem % tx_n) << 4;
+  } else {
+    m_blk = (long long)bx * GBM;
+  }
   const int n_blk = blockIdx.y * GBN;
 
   // ---- per-thread glds chunk descriptors (K-invariant) -----------------
@@ -104,10 +130,16 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       // A row
-      const long long m = m_blk + (long long)h * 128 + row;
       AChunk d;
-      d.ok = m < M;
-      if (IS_CONV) {
+      if (IS_CONV && geo.tile2d) {
+        const int idx = h * 128 + row;  // 0..255 within the 16x16 tile
+        d.ok = true;
+        d.px_y = t_py0 + (idx >> 4);
+        d.px_x = t_px0 + (idx & 15);
+        d.row_base = A + (long long)t_b * geo.H * geo.W * geo.C;
+      } else if (IS_CONV) {
+        const long long m = m_blk + (long long)h * 128 + row;
+        d.ok = m < M;
         const long long HW = (long long)geo.H * geo.W;
         const long long mm = d.ok ? m : 0;
         const int pb = (int)(mm / HW);
@@ -116,6 +148,8 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
         d.px_x = rem % geo.W;
         d.row_base = A + ((long long)pb * HW) * geo.C;
       } else {
+        const long long m = m_blk + (long long)h * 128 + row;
+        d.ok = m < M;
         d.row_base = A + (d.ok ? m * (long long)Kdim : 0);
         d.px_y = d.px_x = 0;
       }
@@ -251,8 +285,15 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
       const float bval = bias ? bias[n] : 0.f;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        const long long m = m_wave + mf * 16 + kgrp * 4 + rr;
-        if (m >= M) continue;
+        long long m;
+        if (IS_CONV && geo.tile2d) {
+          const int idx = wm * 128 + mf * 16 + kgrp * 4 + rr;
+          m = ((long long)t_b * geo.H + t_py0 + (idx >> 4)) * geo.W +
+              t_px0 + (idx & 15);
+        } else {
+          m = m_wave + mf * 16 + kgrp * 4 + rr;
+          if (m >= M) continue;
+        }
         float v = acc[mf][nf][rr] + bval;
         if (FUSE_SILU) v = silu_f(v);
         Y[m * N + n] = f32_to_bf16_bits(v);
@@ -292,7 +333,7 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
   }
   dim3 grid((unsigned)((M + GBM - 1) / GBM), (unsigned)((N + GBN - 1) / GBN));
   auto stream = at::hip::getCurrentHIPStream();
-  ConvGeo geo{0, 0, 0, 0};
+  ConvGeo geo{0, 0, 0, 0, 0, 0};
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<false, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
@@ -328,9 +369,17 @@ torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
     bf32 = bias.contiguous().to(at::kFloat);
     bptr = bf32.data_ptr<float>();
   }
-  dim3 grid((unsigned)((M + GBM - 1) / GBM), (unsigned)((K + GBN - 1) / GBN));
+  static const int xcd_swz = [] {
+    const char* e = getenv("DISTGPU_CONV_XCDSWZ");
+    return (!e || e[0] == '1') ? 1 : 0;
+  }();
+  const int tile2d = (H % 16 == 0 && W % 16 == 0) ? 1 : 0;
+  const unsigned gx = tile2d
+      ? (unsigned)(B * (H / 16) * (W / 16))
+      : (unsigned)((M + GBM - 1) / GBM);
+  dim3 grid(gx, (unsigned)((K + GBN - 1) / GBN));
   auto stream = at::hip::getCurrentHIPStream();
-  ConvGeo geo{(int)H, (int)W, (int)C, (int)rs};
+  ConvGeo geo{(int)H, (int)W, (int)C, (int)rs, tile2d, xcd_swz};
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<true, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),

@@ -133,7 +133,9 @@ def test_conv256_matches_fp32_conv():
     mod = ext.get_ext(True)
     torch.manual_seed(1)
     for (B, C, H, W, K, rs) in [(2, 64, 17, 19, 96, 9), (1, 128, 34, 34, 320, 9),
-                                (2, 320, 20, 20, 320, 9), (2, 64, 32, 32, 48, 1)]:
+                                (2, 320, 20, 20, 320, 9), (2, 64, 32, 32, 48, 1),
+                                # H,W % 16 == 0 -> the 16x16 tile2d mode
+                                (1, 128, 32, 48, 64, 9), (2, 64, 16, 16, 96, 9)]:
         x = (torch.randn(B, H, W, C, device="cuda") / 4).to(torch.bfloat16)
         wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
         if rs == 1:
