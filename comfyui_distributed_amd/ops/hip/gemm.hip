@@ -114,49 +114,63 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
   const int n_blk = blockIdx.y * GBN;
 
   // ---- per-thread glds chunk descriptors (K-invariant) -----------------
-  // chunk slot (instr i, half h): chunk = tid + i*512 within half h's
-  // [128][64] image; logical byte = swz(chunk*16); row = logical>>7,
-  // k-byte = logical&127 -> channel offset c_local = (logical&127)/2.
-  AChunk a_desc[2][2];   // [half][instr]
+  // Staging is 4 pieces of 16 KiB per K-tile, 2 glds each, issued in the
+  // order [B0, B1, A0, A1] so counted vmcnt waits drain exactly what the
+  // next quadrant phase reads (B fully + A's mq=0 rows at phase 0; A's
+  // mq=1 rows only by phase 2). The A image is row-permuted so piece A0
+  // holds the mq=0 rows of BOTH wave halves:
+  //   image quarter qq = mq*2 + wm  ->  m_local = (qq&1)*128+(qq>>1)*64+r.
+  // chunk slot (piece, instr i): chunk = tid + i*512; logical byte within
+  // the 32 KiB image = swz(piece_off + chunk*16).
+  AChunk a_desc[2][2];   // [a_piece][instr]
   const uint16_t* b_src[2][2];
-  int c_local[2];        // per instr (same for both halves)
-  int a_cloc[2];
+  int c_local[2][2];     // [a_piece][instr] channel offset
+  int cb_local[2][2];    // [b_piece][instr]
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
-    const unsigned chunk = (unsigned)tid + (unsigned)i * 512u;
-    const unsigned lo = swz(chunk * 16u);
-    const int row = (int)(lo >> 7);
-    a_cloc[i] = c_local[i] = (int)((lo & 127u) >> 1);
+    const unsigned chunk16 = ((unsigned)tid + (unsigned)i * 512u) * 16u;
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      // A row
-      AChunk d;
-      if (IS_CONV && geo.tile2d) {
-        const int idx = h * 128 + row;  // 0..255 within the 16x16 tile
-        d.ok = true;
-        d.px_y = t_py0 + (idx >> 4);
-        d.px_x = t_px0 + (idx & 15);
-        d.row_base = A + (long long)t_b * geo.H * geo.W * geo.C;
-      } else if (IS_CONV) {
-        const long long m = m_blk + (long long)h * 128 + row;
-        d.ok = m < M;
-        const long long HW = (long long)geo.H * geo.W;
-        const long long mm = d.ok ? m : 0;
-        const int pb = (int)(mm / HW);
-        const int rem = (int)(mm - (long long)pb * HW);
-        d.px_y = rem / geo.W;
-        d.px_x = rem % geo.W;
-        d.row_base = A + ((long long)pb * HW) * geo.C;
-      } else {
-        const long long m = m_blk + (long long)h * 128 + row;
-        d.ok = m < M;
-        d.row_base = A + (d.ok ? m * (long long)Kdim : 0);
-        d.px_y = d.px_x = 0;
+    for (int pc = 0; pc < 2; ++pc) {
+      // ---- A piece pc: image rows pc*128..+128 (quarter-permuted) ----
+      {
+        const unsigned y = swz((unsigned)pc * 16384u + chunk16);
+        const int img_row = (int)(y >> 7);
+        const int qq = img_row >> 6;
+        const int m_local = ((qq & 1) << 7) | ((qq >> 1) << 6) |
+                            (img_row & 63);
+        c_local[pc][i] = (int)((y & 127u) >> 1);
+        AChunk d;
+        if (IS_CONV && geo.tile2d) {
+          d.ok = true;
+          d.px_y = t_py0 + (m_local >> 4);
+          d.px_x = t_px0 + (m_local & 15);
+          d.row_base = A + (long long)t_b * geo.H * geo.W * geo.C;
+        } else if (IS_CONV) {
+          const long long m = m_blk + m_local;
+          d.ok = m < M;
+          const long long HW = (long long)geo.H * geo.W;
+          const long long mm = d.ok ? m : 0;
+          const int pb = (int)(mm / HW);
+          const int rem = (int)(mm - (long long)pb * HW);
+          d.px_y = rem / geo.W;
+          d.px_x = rem % geo.W;
+          d.row_base = A + ((long long)pb * HW) * geo.C;
+        } else {
+          const long long m = m_blk + m_local;
+          d.ok = m < M;
+          d.row_base = A + (d.ok ? m * (long long)Kdim : 0);
+          d.px_y = d.px_x = 0;
+        }
+        a_desc[pc][i] = d;
       }
-      a_desc[h][i] = d;
-      // B row
-      const int n = n_blk + h * 128 + row;
-      b_src[h][i] = (n < N) ? Bw + (long long)n * Kdim : nullptr;
+      // ---- B piece pc: image rows pc*128..+128 (linear) --------------
+      {
+        const unsigned y = swz((unsigned)pc * 16384u + chunk16);
+        const int row = (int)(y >> 7);
+        cb_local[pc][i] = (int)((y & 127u) >> 1);
+        const int n = n_blk + row;
+        b_src[pc][i] = (n < N) ? Bw + (long long)n * Kdim : nullptr;
+      }
     }
   }
 
@@ -172,8 +186,10 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
 
   const int n_ktiles = Kdim / GBK;
 
-  // ---- glds stage of one K-tile into buffer p --------------------------
-  auto stage = [&](int kt, int p) {
+  // ---- glds stage of one 16 KiB piece (2 instrs) into buffer p ---------
+  // piece 0 = B rows 0..127, 1 = B rows 128..255, 2 = A quarters 0-1
+  // (mq=0 rows of both wave halves), 3 = A quarters 2-3 (mq=1 rows)
+  auto stage_piece = [&](int kt, int p, int piece) {
     const int kbase = kt * GBK;
     int tap_dy = 0, tap_dx = 0, cbase = kbase;
     if (IS_CONV && geo.RS == 9) {
@@ -181,96 +197,104 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
       cbase = kbase - tap * geo.C;
       tap_dy = tap / 3 - 1;
       tap_dx = tap % 3 - 1;
-    } else if (IS_CONV) {
-      cbase = kbase;  // 1x1: k == c
     }
+    const bool is_b = piece < 2;
+    const int pc = piece & 1;
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        // A half h, instr i
-        const AChunk& d = a_desc[h][i];
-        const uint16_t* src;
+    for (int i = 0; i < 2; ++i) {
+      const uint16_t* src;
+      unsigned base;
+      if (is_b) {
+        src = b_src[pc][i] ? b_src[pc][i] + kbase + cb_local[pc][i]
+                           : zero_page;
+        base = (unsigned)G_BBUF + p * G_TILE + pc * 16384u;
+      } else {
+        const AChunk& d = a_desc[pc][i];
         if (IS_CONV) {
           const int sy = d.px_y + tap_dy;
           const int sx = d.px_x + tap_dx;
           const bool ok = d.ok && sy >= 0 && sy < geo.H && sx >= 0 &&
                           sx < geo.W;
           src = ok ? d.row_base + ((long long)sy * geo.W + sx) * geo.C +
-                         cbase + a_cloc[i]
+                         cbase + c_local[pc][i]
                    : zero_page;
         } else {
-          src = d.ok ? d.row_base + kbase + a_cloc[i] : zero_page;
+          src = d.ok ? d.row_base + kbase + c_local[pc][i] : zero_page;
         }
-        // wave-uniform LDS base; hardware adds lane*16
-        const unsigned lds_off = (unsigned)G_ABUF + p * G_TILE + h * G_HALF +
-                                 (wave * 64u + i * 512u) * 16u;
-        glds16(src, reinterpret_cast<uint8_t*>(lds) + lds_off);
-        // B half h, instr i
-        const uint16_t* bsrc =
-            b_src[h][i] ? b_src[h][i] + kbase + c_local[i] : zero_page;
-        const unsigned lds_off_b = (unsigned)G_BBUF + p * G_TILE +
-                                   h * G_HALF + (wave * 64u + i * 512u) * 16u;
-        glds16(bsrc, reinterpret_cast<uint8_t*>(lds) + lds_off_b);
+        base = (unsigned)G_ABUF + p * G_TILE + pc * 16384u;
       }
+      // wave-uniform LDS base; hardware adds lane*16
+      glds16(src, reinterpret_cast<uint8_t*>(lds) + base +
+                      (wave * 64u + i * 512u) * 16u);
     }
   };
 
-  // ---- compute one K-tile from buffer p --------------------------------
-  auto compute = [&](int p) {
-    const unsigned a_base = (unsigned)G_ABUF + p * G_TILE + wm * G_HALF;
-    // quadrants: (mq, nq) -> m-frags mq*4..+4, n-frags nq*2..+2
+  // ---- one quadrant (16 MFMA over the K-tile's 64-deep K) --------------
+  auto compute_quadrant = [&](int p, int mq, int nq) {
+    const unsigned a_base = (unsigned)G_ABUF + p * G_TILE;
+    const unsigned b_base = (unsigned)G_BBUF + p * G_TILE;
+    const int qq = mq * 2 + wm;  // A image quarter for this wave
+    short8_g af[4][2], bfr[2][2];
 #pragma unroll
-    for (int mq = 0; mq < 2; ++mq) {
+    for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-      for (int nq = 0; nq < 2; ++nq) {
-        short8_g af[4][2], bf[2][2];
-#pragma unroll
-        for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks) {
-            const unsigned lo =
-                ((mq * 4 + mf) * 16 + fr) * 128u + ks * 64u + kgrp * 16u;
-            af[mf][ks] = *reinterpret_cast<const short8_g*>(
-                lds + ((a_base + swz(lo)) >> 1));
-          }
-#pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks) {
-            const int brow = wn * 64 + (nq * 2 + nf) * 16 + fr;
-            const unsigned b_base =
-                (unsigned)G_BBUF + p * G_TILE + (brow >> 7) * G_HALF;
-            const unsigned lo = (brow & 127) * 128u + ks * 64u + kgrp * 16u;
-            bf[nf][ks] = *reinterpret_cast<const short8_g*>(
-                lds + ((b_base + swz(lo)) >> 1));
-          }
-        __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-        for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-          for (int nf = 0; nf < 2; ++nf)
-#pragma unroll
-            for (int ks = 0; ks < 2; ++ks)
-              acc[mq * 4 + mf][nq * 2 + nf] =
-                  __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                      af[mf][ks], bf[nf][ks], acc[mq * 4 + mf][nq * 2 + nf],
-                      0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+      for (int ks = 0; ks < 2; ++ks) {
+        const unsigned lo =
+            (qq * 64 + mf * 16 + fr) * 128u + ks * 64u + kgrp * 16u;
+        af[mf][ks] = *reinterpret_cast<const short8_g*>(
+            lds + ((a_base + swz(lo)) >> 1));
       }
-    }
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int brow = wn * 64 + (nq * 2 + nf) * 16 + fr;
+        const unsigned lo = brow * 128u + ks * 64u + kgrp * 16u;
+        bfr[nf][ks] = *reinterpret_cast<const short8_g*>(
+            lds + ((b_base + swz(lo)) >> 1));
+      }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[mq * 4 + mf][nq * 2 + nf] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[mf][ks], bfr[nf][ks], acc[mq * 4 + mf][nq * 2 + nf],
+                  0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
   };
 
-  // ---- main loop: double-buffered glds, one barrier pair per K-tile ----
-  stage(0, 0);
+  // ---- main loop ---------------------------------------------------------
+  // Raw s_barrier + counted vmcnt (guide §5: __syncthreads with glds in
+  // flight drains vmcnt(0) — the ~20% structural stall of the simple
+  // variant). One barrier per K-tile; the next tile's pieces are issued
+  // between quadrant clusters; vmcnt(2) at phase 0 drains this tile's
+  // B0,B1,A0 (A1 may still fly), vmcnt(4) before the mq=1 quadrants
+  // drains A1.
+#pragma unroll
+  for (int pc = 0; pc < 4; ++pc) stage_piece(0, 0, pc);
   for (int kt = 0; kt < n_ktiles; ++kt) {
     const int p = kt & 1;
-    // __syncthreads() with glds in flight emits vmcnt(0): all of buffer
-    // p's DMA (issued last iteration) has landed for every wave after the
-    // barrier (guide §5 "glds, 2 LDS buffers, BK=64" row)
-    __syncthreads();
-    if (kt + 1 < n_ktiles) stage(kt + 1, 1 - p);
-    compute(p);
+    const bool more = kt + 1 < n_ktiles;
+    if (more)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (more) stage_piece(kt + 1, 1 - p, 0);
+    compute_quadrant(p, 0, 0);
+    if (more) stage_piece(kt + 1, 1 - p, 1);
+    compute_quadrant(p, 0, 1);
+    if (more) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      stage_piece(kt + 1, 1 - p, 2);
+    }
+    compute_quadrant(p, 1, 0);
+    if (more) stage_piece(kt + 1, 1 - p, 3);
+    compute_quadrant(p, 1, 1);
   }
 
   // ---- epilogue: bias + optional SiLU, bf16 store ----------------------
