@@ -18,7 +18,7 @@ import torch.nn.functional as F
 from . import ext
 
 LANCZOS_A = 3
-ATTN_DPADS = (64, 96, 128, 160)
+ATTN_DPADS = (48, 64, 96, 128, 160)
 
 
 def hip_available() -> bool:

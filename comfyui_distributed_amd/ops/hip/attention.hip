@@ -32,7 +32,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define ZERO8 short8{0, 0, 0, 0, 0, 0, 0, 0}
 
-template <int D_PAD>
+template <int D_PAD>  // multiple of 16; QK^T K-depth rounds up to 32s
 __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ o, int Nq, int Nk,
@@ -40,7 +40,7 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     long long q_hstride, long long q_rstride, long long k_bstride,
     long long k_hstride, long long k_rstride, long long o_bstride,
     long long o_hstride, long long o_rstride) {
-  constexpr int DK = D_PAD / 32;
+  constexpr int DK = (D_PAD + 31) / 32;
   constexpr int DN = D_PAD / 16;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -242,7 +242,10 @@ static torch::Tensor launch_attn_raw(const uint16_t* qp, const uint16_t* kp,
                                      float scale, long long qb, long long qh,
                                      long long qr, long long kb, long long kh,
                                      long long kr, int batch) {
-  const int dpad = D <= 64 ? 64 : (D <= 96 ? 96 : (D <= 128 ? 128 : 160));
+  // D=40 (SD1.5's hot dim) pads to 48, not 64: PV runs 3 d-fragments
+  // instead of 4 and the V^T stage shrinks 25% on the dominant kernel
+  const int dpad = D <= 48 ? 48
+                 : (D <= 64 ? 64 : (D <= 96 ? 96 : (D <= 128 ? 128 : 160)));
   TORCH_CHECK(D % 8 == 0 && D <= 160, "head dim must be %8 and <=160, got ", D);
   // output is always a fresh packed [B, Nq, H*D] tensor
   const long long ob = (long long)Nq * H * D, oh = D, orr = (long long)H * D;
@@ -254,6 +257,7 @@ static torch::Tensor launch_attn_raw(const uint16_t* qp, const uint16_t* kp,
                      vp, (uint16_t*)o.data_ptr(), Nq, Nk, D, H, Hkv, scale,   \
                      qb, qh, qr, kb, kh, kr, ob, oh, orr)
   switch (dpad) {
+    case 48: LAUNCH_D(48); break;
     case 64: LAUNCH_D(64); break;
     case 96: LAUNCH_D(96); break;
     case 128: LAUNCH_D(128); break;
