@@ -97,6 +97,7 @@ class DistributedServer:
         r = app.router
         r.add_get("/", self.get_panel)
         r.add_get("/panel", self.get_panel)
+        r.add_get("/static/{tail:.+}", self.get_static)
         r.add_get("/prompt", self.get_prompt)
         r.add_post("/prompt", self.post_prompt)
         r.add_get("/object_info", self.get_object_info)
@@ -221,6 +222,18 @@ class DistributedServer:
 
         path = Path(__file__).parent / "static" / "panel.html"
         return web.Response(text=path.read_text(), content_type="text/html")
+
+    async def get_static(self, request):
+        """Panel assets (static/js/*). Confined to the static dir."""
+        from pathlib import Path
+
+        base = (Path(__file__).parent / "static").resolve()
+        cand = (base / request.match_info["tail"]).resolve()
+        if not str(cand).startswith(str(base) + os.sep) or not cand.is_file():
+            raise web.HTTPNotFound
+        ctype = ("application/javascript" if cand.suffix == ".js"
+                 else "text/css" if cand.suffix == ".css" else "text/plain")
+        return web.Response(text=cand.read_text(), content_type=ctype)
 
     async def get_prompt(self, request):
         return web.json_response(
