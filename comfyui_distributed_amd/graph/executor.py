@@ -229,6 +229,23 @@ class Executor:
                     kwargs[name] = src_out[int(value[1])]
                 else:
                     kwargs[name] = value
+            # ComfyUI hidden-input convention: UNIQUE_ID injects the node's
+            # own id, EXTRA_PNGINFO the executing workflow (the prompt dict
+            # here — it is what SaveImage embeds into output metadata), so
+            # nodes like DistributedModelName can write resolved values
+            # back (reference nodes/utilities.py:164-224)
+            hidden = {}
+            try:
+                hidden = cls.INPUT_TYPES().get("hidden", {})
+            except Exception:
+                pass
+            for hname, hkind in hidden.items():
+                if hname in kwargs:
+                    continue
+                if hkind == "UNIQUE_ID":
+                    kwargs[hname] = nid
+                elif hkind == "EXTRA_PNGINFO":
+                    kwargs[hname] = {"workflow": prompt}
             inst = cls()
             if hasattr(inst, "set_context"):
                 inst.set_context(self.context)

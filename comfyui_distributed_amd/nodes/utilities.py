@@ -147,12 +147,19 @@ class DistributedValue:
 
 
 class DistributedModelName:
-    """Output node: stringifies a model path so workers can substitute their
-    own local model files (reference :164-224)."""
+    """Output node: stringifies a model path so workers can substitute
+    their own local model files, AND writes the resolved value back into
+    the executing workflow's node entry (reference :164-224 updates
+    widgets_values the same way — the resolved name then travels in saved
+    outputs' embedded workflow metadata)."""
 
     @classmethod
     def INPUT_TYPES(cls):
-        return {"required": {"model_name": (any_type,)}}
+        return {
+            "required": {"model_name": (any_type,)},
+            "hidden": {"unique_id": "UNIQUE_ID",
+                       "extra_pnginfo": "EXTRA_PNGINFO"},
+        }
 
     RETURN_TYPES = ("STRING",)
     RETURN_NAMES = ("name",)
@@ -160,8 +167,29 @@ class DistributedModelName:
     OUTPUT_NODE = True
     CATEGORY = "utils"
 
-    def log_input(self, model_name):
-        return (str(model_name),)
+    @staticmethod
+    def _stringify(value):
+        if isinstance(value, str):
+            return value
+        if isinstance(value, (int, float, bool)):
+            return str(value)
+        try:
+            return json.dumps(value, indent=4)
+        except (TypeError, ValueError):
+            return str(value)
+
+    def log_input(self, model_name, unique_id=None, extra_pnginfo=None):
+        values = ([self._stringify(v) for v in model_name]
+                  if isinstance(model_name, list)
+                  else [self._stringify(model_name)])
+        # write-back: keep the resolved display value in the workflow
+        workflow = (extra_pnginfo or {}).get("workflow") \
+            if isinstance(extra_pnginfo, dict) else None
+        if workflow is not None and unique_id is not None:
+            node = workflow.get(str(unique_id))
+            if isinstance(node, dict):
+                node["widgets_values"] = list(values)
+        return (values[0] if len(values) == 1 else values,)
 
 
 class ImageBatchDivider:

@@ -27,11 +27,76 @@ TUNNEL_URL_RE = re.compile(r"https://[a-z0-9-]+\.trycloudflare\.com")
 START_TIMEOUT = float(os.environ.get("TUNNEL_START_TIMEOUT", "30"))
 
 
-def find_cloudflared() -> str | None:
+#: GitHub latest-release asset per platform (reference binary.py:13-46)
+_CF_ASSETS = {
+    ("linux", "x86_64"): "cloudflared-linux-amd64",
+    ("linux", "aarch64"): "cloudflared-linux-arm64",
+    ("darwin", "x86_64"): "cloudflared-darwin-amd64.tgz",
+    ("darwin", "arm64"): "cloudflared-darwin-amd64.tgz",
+    ("windows", "amd64"): "cloudflared-windows-amd64.exe",
+}
+_CF_RELEASE_URL = ("https://github.com/cloudflare/cloudflared/releases/"
+                   "latest/download/{asset}")
+
+
+def _download_dir() -> str:
+    return os.path.join(os.path.dirname(__file__), "..", "..", ".cloudflared")
+
+
+def download_cloudflared(fetch=None, dest_dir: str | None = None,
+                         timeout: float = 60.0) -> str | None:
+    """Fetch the platform's cloudflared binary from the GitHub latest
+    release (reference utils/cloudflare/binary.py:47-83). ``fetch(url,
+    timeout) -> bytes`` is injectable for tests; the default uses urllib
+    and simply returns None when the host has no egress (this build
+    environment does not — the tunnel then requires CLOUDFLARED_PATH)."""
+    import platform
+
+    key = (platform.system().lower(), platform.machine().lower())
+    asset = _CF_ASSETS.get(key)
+    if asset is None or asset.endswith((".tgz", ".exe")):
+        return None  # only the plain linux binaries are auto-installed
+    url = _CF_RELEASE_URL.format(asset=asset)
+    if fetch is None:
+        def fetch(u, t):
+            import urllib.request
+
+            with urllib.request.urlopen(u, timeout=t) as resp:
+                return resp.read()
+    try:
+        data = fetch(url, timeout)
+    except Exception as exc:  # no egress / fetch failure -> graceful None
+        log(f"cloudflared download failed ({exc!r}); set CLOUDFLARED_PATH")
+        return None
+    if not data:
+        return None
+    dest_dir = dest_dir or _download_dir()
+    os.makedirs(dest_dir, exist_ok=True)
+    dest = os.path.join(dest_dir, "cloudflared")
+    tmp = dest + ".part"
+    with open(tmp, "wb") as fh:
+        fh.write(data)
+    os.chmod(tmp, 0o755)
+    os.replace(tmp, dest)
+    log(f"cloudflared downloaded to {dest}")
+    return dest
+
+
+def find_cloudflared(allow_download: bool = False, fetch=None) -> str | None:
+    """Resolution order (reference binary.py:13-83): CLOUDFLARED_PATH ->
+    PATH -> previously-downloaded copy -> (optional) fresh download."""
     override = os.environ.get("CLOUDFLARED_PATH")
     if override and os.path.isfile(override):
         return override
-    return shutil.which("cloudflared")
+    on_path = shutil.which("cloudflared")
+    if on_path:
+        return on_path
+    cached = os.path.join(_download_dir(), "cloudflared")
+    if os.path.isfile(cached) and os.access(cached, os.X_OK):
+        return cached
+    if allow_download:
+        return download_cloudflared(fetch=fetch)
+    return None
 
 
 class ProcessReader:
@@ -74,7 +139,7 @@ class TunnelManager:
         async with self._lock:
             if self.proc is not None and self.proc.poll() is None:
                 return self.url or ""
-            binary = find_cloudflared()
+            binary = find_cloudflared(allow_download=True)
             if binary is None:
                 raise TunnelError(
                     "cloudflared binary not found (no egress to download it; "

@@ -108,3 +108,36 @@ def test_tunnel_manager_with_fake_binary(tmp_path, tmp_config, monkeypatch):
         assert mgr.status()["active"] is False
 
     asyncio.run(go())
+
+
+def test_cloudflared_download_with_faked_fetch(tmp_path, monkeypatch):
+    """Reference utils/cloudflare/binary.py:47-83 parity: the binary is
+    fetched from the GitHub latest release, chmod +x'd, and found by the
+    resolver afterwards; a fetch failure (no egress) degrades to None."""
+    import os
+
+    from comfyui_distributed_amd.server import tunnel as tmod
+
+    monkeypatch.delenv("CLOUDFLARED_PATH", raising=False)
+    fetched = {}
+
+    def fake_fetch(url, timeout):
+        fetched["url"] = url
+        return b"#!/bin/sh\necho fake-cloudflared\n"
+
+    dest = tmod.download_cloudflared(fetch=fake_fetch, dest_dir=str(tmp_path))
+    assert dest and os.path.isfile(dest)
+    assert os.access(dest, os.X_OK)
+    assert "cloudflared/releases/latest/download" in fetched["url"]
+
+    # resolver picks up the downloaded copy
+    monkeypatch.setattr(tmod, "_download_dir", lambda: str(tmp_path))
+    monkeypatch.setattr(tmod.shutil, "which", lambda name: None)
+    assert tmod.find_cloudflared() == dest
+
+    # fetch failure (no egress) -> graceful None
+    def broken_fetch(url, timeout):
+        raise OSError("no route to host")
+
+    assert tmod.download_cloudflared(
+        fetch=broken_fetch, dest_dir=str(tmp_path / "x")) is None

@@ -72,3 +72,27 @@ def test_zero_is_a_legitimate_worker_override():
     vals2 = json.dumps({"_type": "FLOAT", "1": ""})
     out = node.distribute("3.5", vals2, is_worker=True, worker_id="worker_0")
     assert out == (3.5,)
+
+
+def test_model_name_writes_back_into_workflow():
+    """Reference nodes/utilities.py:164-224: the resolved model name is
+    written into the executing workflow's node entry (widgets_values)."""
+    from comfyui_distributed_amd.graph.executor import Executor
+
+    prompt = {
+        "1": {"class_type": "DistributedModelName",
+              "inputs": {"model_name": "sd15.safetensors"}},
+    }
+    ex = Executor()
+    out = ex.execute(prompt)
+    assert out["1"][0] == "sd15.safetensors"
+    assert prompt["1"]["widgets_values"] == ["sd15.safetensors"]
+
+
+def test_model_name_stringifies_nonstring():
+    from comfyui_distributed_amd.nodes.utilities import DistributedModelName
+
+    node = DistributedModelName()
+    assert node.log_input(7)[0] == "7"
+    assert node.log_input({"a": 1})[0] == '{\n    "a": 1\n}'
+    assert node.log_input(["a", "b"])[0] == ["a", "b"]
