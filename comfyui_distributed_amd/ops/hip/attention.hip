@@ -35,7 +35,14 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 template <int D_PAD>  // multiple of 16; QK^T K-depth rounds up to 32s
 __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
-    const uint16_t* __restrict__ v, uint16_t* __restrict__ o, int Nq, int Nk,
+    const uint16_t* __restrict__ v, uint16_t* __restrict__ o,
+    const uint16_t* __restrict__ zp,  // >=16B zeros: OOB loads redirect
+                                      // here so every load is UNconditional
+                                      // (a per-lane branchy load makes
+                                      // hipcc emit vmcnt(0) per cluster —
+                                      // guide §5 trap (c) — serializing
+                                      // the whole prefetch pipeline)
+    int Nq, int Nk,
     int D, int H, int Hkv, float scale, long long q_bstride,
     long long q_hstride, long long q_rstride, long long k_bstride,
     long long k_hstride, long long k_rstride, long long o_bstride,
@@ -60,7 +67,7 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
 
   const int kd0 = (lane >> 4) * 8;
 
-  // ---- Q fragments resident (masked rows / masked d) --------------------
+  // ---- Q fragments resident (OOB rows/d -> zero page) -------------------
   short8 qfrag[DK];
   {
     const int row = q_row0 + (lane & 15);
@@ -68,9 +75,9 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
       const int d = kk * 32 + kd0;
-      qfrag[kk] = (row_ok && d + 8 <= D)
-          ? *reinterpret_cast<const short8*>(qbase + (long long)row * q_rstride + d)
-          : ZERO8;
+      const uint16_t* src = (row_ok && d + 8 <= D)
+          ? qbase + (long long)row * q_rstride + d : zp;
+      qfrag[kk] = *reinterpret_cast<const short8*>(src);
     }
   }
 
@@ -91,28 +98,52 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
 #pragma unroll
       for (int f = 0; f < KFRAG; ++f) {
         const int key = key0 + f * 16 + (lane & 15);
-        dst[kk][f] = (d_ok && key < Nk)
-            ? *reinterpret_cast<const short8*>(kbase + (long long)key * k_rstride + d)
-            : ZERO8;
+        const uint16_t* src = (d_ok && key < Nk)
+            ? kbase + (long long)key * k_rstride + d : zp;
+        dst[kk][f] = *reinterpret_cast<const short8*>(src);
       }
     }
   };
   load_kfrags(0, kf);
 
+  // ---- V register double-buffer: tile t+1's V rows are loaded during
+  // tile t's compute and written to LDS at tile start — the old
+  // load-then-__syncthreads stage exposed the full L2/HBM latency at
+  // every tile's barrier (vmcnt(0) drain). One short8 slot per thread
+  // covers KT*D_PAD/8 elements; big D needs 2-3 slots.
+  constexpr int VSLOTS = (KT * (D_PAD / 8) + NWAVES * 64 - 1) / (NWAVES * 64);
+  short8 vreg[VSLOTS];
+  auto load_v = [&](int key0) {
+#pragma unroll
+    for (int s = 0; s < VSLOTS; ++s) {
+      const int c = (int)threadIdx.x + s * NWAVES * 64;
+      const int key = c & (KT - 1);
+      const int d0 = (c / KT) * 8;
+      // unconditional load (zero page covers c overflow, key tail and
+      // d beyond D) — a lane-divergent branch would serialize the cluster
+      const bool ok = c < KT * (D_PAD / 8) && key0 + key < Nk && d0 + 8 <= D;
+      const uint16_t* src =
+          ok ? vbase + (long long)(key0 + key) * k_rstride + d0 : zp;
+      vreg[s] = *reinterpret_cast<const short8*>(src);
+    }
+  };
+  load_v(0);
+
   for (int t = 0; t < ntiles; ++t) {
     const int key0 = t * KT;
 
-    // ---- stage V^T cooperatively (masked) --------------------------------
+    // ---- commit the prefetched V^T rows (barrier-bracketed, no global
+    // wait: the loads were issued one full tile ago) ----------------------
     __syncthreads();
-    for (int c = threadIdx.x; c < KT * (D_PAD / 8); c += NWAVES * 64) {
-      const int key = c & (KT - 1);
-      const int d0 = (c / KT) * 8;
-      short8 vv = (key0 + key < Nk && d0 + 8 <= D)
-          ? *reinterpret_cast<const short8*>(
-                vbase + (long long)(key0 + key) * k_rstride + d0)
-          : ZERO8;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) v_t[d0 + j][key] = (uint16_t)vv[j];
+    for (int s = 0; s < VSLOTS; ++s) {
+      const int c = (int)threadIdx.x + s * NWAVES * 64;
+      if (c < KT * (D_PAD / 8)) {
+        const int key = c & (KT - 1);
+        const int d0 = (c / KT) * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v_t[d0 + j][key] = (uint16_t)vreg[s][j];
+      }
     }
     __syncthreads();
 
@@ -131,7 +162,12 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
         s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[kk][f], qfrag[kk],
                                                        s[f], 0, 0, 0);
 
-    if (t + 1 < ntiles) load_kfrags(key0 + KT, kf);
+    // t+1 prefetches issue HERE — after the QK^T MFMAs' kf-wait, so the
+    // compiler's vmcnt(0) before the MFMAs never drains the fresh loads
+    if (t + 1 < ntiles) {
+      load_kfrags(key0 + KT, kf);
+      load_v(key0 + KT);
+    }
 
 #pragma unroll
     for (int f = 0; f < KFRAG; ++f) {
@@ -236,12 +272,23 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
   }
 }
 
+static torch::Tensor g_attn_zero_page;
+
+static const uint16_t* attn_zero_page(const torch::Tensor& like) {
+  if (!g_attn_zero_page.defined() ||
+      g_attn_zero_page.device() != like.device())
+    g_attn_zero_page =
+        torch::zeros({64}, like.options().dtype(at::kBFloat16));
+  return (const uint16_t*)g_attn_zero_page.data_ptr();
+}
+
 static torch::Tensor launch_attn_raw(const uint16_t* qp, const uint16_t* kp,
                                      const uint16_t* vp, torch::Tensor o,
                                      int H, int Hkv, int Nq, int Nk, int D,
                                      float scale, long long qb, long long qh,
                                      long long qr, long long kb, long long kh,
                                      long long kr, int batch) {
+  const uint16_t* zp = attn_zero_page(o);
   // D=40 (SD1.5's hot dim) pads to 48, not 64: PV runs 3 d-fragments
   // instead of 4 and the V^T stage shrinks 25% on the dominant kernel
   const int dpad = D <= 48 ? 48
@@ -254,8 +301,8 @@ static torch::Tensor launch_attn_raw(const uint16_t* qp, const uint16_t* kp,
   auto stream = at::hip::getCurrentHIPStream();
 #define LAUNCH_D(DP)                                                          \
   hipLaunchKernelGGL((attn_fwd_kernel<DP>), grid, block, 0, stream, qp, kp,   \
-                     vp, (uint16_t*)o.data_ptr(), Nq, Nk, D, H, Hkv, scale,   \
-                     qb, qh, qr, kb, kh, kr, ob, oh, orr)
+                     vp, (uint16_t*)o.data_ptr(), zp, Nq, Nk, D, H, Hkv,      \
+                     scale, qb, qh, qr, kb, kh, kr, ob, oh, orr)
   switch (dpad) {
     case 48: LAUNCH_D(48); break;
     case 64: LAUNCH_D(64); break;
