@@ -102,18 +102,25 @@ def gather_tensor_lists(
         payload = None
         meta_t = None
 
-    # shape agreement: broadcast reference shape from the first non-empty rank
+    # shape+dtype agreement: broadcast from the first non-empty rank.
+    # Payloads travel in their SOURCE dtype — bf16 tile results cost half
+    # the xGMI bytes the old fp32 upcast did (r1 verdict item 7).
+    dtype_codes = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2,
+                   torch.uint8: 3, torch.int64: 4}
+    code_dtypes = {v: k for k, v in dtype_codes.items()}
     shape_src = next((r for r, c in enumerate(counts) if c > 0), None)
     if shape_src is None:
         return ([], []) if ctx.is_master else None
-    shape_t = torch.zeros(8, dtype=torch.int64, device=device)
+    shape_t = torch.zeros(10, dtype=torch.int64, device=device)
     if ctx.rank == shape_src and shape is not None:
         dims = torch.tensor(shape, dtype=torch.int64, device=device)
         shape_t[0] = len(shape)
         shape_t[1 : 1 + len(shape)] = dims
+        shape_t[9] = dtype_codes.get(payload.dtype, 0)
     dist.broadcast(shape_t, src=shape_src)
     ndim = int(shape_t[0].item())
     ref_shape = tuple(int(x) for x in shape_t[1 : 1 + ndim])
+    wire_dtype = code_dtypes[int(shape_t[9].item())]
 
     # One batched p2p group: every source's isend pairs with a
     # pre-posted irecv on rank 0, so all sources stream CONCURRENTLY over
@@ -126,7 +133,7 @@ def gather_tensor_lists(
             n = counts[src]
             if n == 0:
                 continue
-            buf = torch.empty((n, *ref_shape), dtype=torch.float32, device=device)
+            buf = torch.empty((n, *ref_shape), dtype=wire_dtype, device=device)
             mbuf = torch.empty((n, meta_width), dtype=torch.int64, device=device)
             bufs[src] = (buf, mbuf)
             ops.append(dist.P2POp(dist.irecv, buf, src))
@@ -145,7 +152,7 @@ def gather_tensor_lists(
                 all_meta.append(tuple(int(x) for x in mbuf[i]))
         return all_tensors, all_meta
     if counts[ctx.rank] > 0:
-        ops = [dist.P2POp(dist.isend, payload.to(torch.float32), 0),
+        ops = [dist.P2POp(dist.isend, payload.to(wire_dtype), 0),
                dist.P2POp(dist.isend, meta_t, 0)]
         for work in dist.batch_isend_irecv(ops):
             work.wait()
