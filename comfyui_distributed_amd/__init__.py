@@ -24,6 +24,16 @@ with torch.distributed over RCCL.
 
 __version__ = "0.1.0"
 
+import os as _os
+
+# The handful of convs still on MIOpen (stride-2 downsamples, the K=4
+# out_conv) trigger MIOpen's exhaustive kernel search on first use: ~220 s
+# of naive-conv candidate runs in the flagship warmup canvas
+# (profiles/r02_results.md). FAST find mode picks from heuristics instead;
+# those convs are <1% of steady-state kernel time, so the tuned-vs-
+# heuristic delta is noise while the warmup drops by minutes.
+_os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 # Node maps at package root (reference __init__.py:1-29 exposes the same
 # names so a host app can discover the node classes; kept lazy so plain
 # `import comfyui_distributed_amd` stays light).

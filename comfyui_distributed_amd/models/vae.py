@@ -15,7 +15,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import dispatch as ops
-from .unet import FusedGroupNorm, StemConv
+from .unet import FusedGroupNorm, MfmaConv2d, StemConv
 
 
 @dataclass
@@ -104,7 +104,7 @@ class VAEEncoder(nn.Module):
                 downs.append(VAEResBlock(cin, cout))
                 cin = cout
             if level != len(cfg.channel_mult) - 1:
-                downs.append(nn.Conv2d(cin, cin, 3, stride=2, padding=1))
+                downs.append(MfmaConv2d(cin, cin, 3, stride=2, padding=1))
         self.down = nn.ModuleList(downs)
         self.mid = nn.ModuleList(
             [VAEResBlock(cin, cin), VAEAttention(cin), VAEResBlock(cin, cin)]

@@ -92,8 +92,13 @@ def act_mul(a: torch.Tensor, b: torch.Tensor, gelu: bool = False) -> torch.Tenso
 def conv_supported(conv) -> bool:
     """True when the MFMA conv kernel covers this nn.Conv2d."""
     k = conv.kernel_size
+    stride_ok = conv.stride == (1, 1) or (
+        # stride-2 3x3 (UNet/VAE Downsample) runs on the 256-tile kernel;
+        # keeping it off MIOpen also kills its exhaustive-find warmup
+        conv.stride == (2, 2) and k == (3, 3) and conv.in_channels % 64 == 0
+    )
     return (
-        conv.stride == (1, 1)
+        stride_ok
         and conv.in_channels % 32 == 0
         and conv.out_channels % 16 == 0
         and ((k == (3, 3) and conv.padding == (1, 1))
@@ -133,12 +138,14 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
     wt = _repacked_weight(conv)
     rs = 9 if conv.kernel_size == (3, 3) else 1
     bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
+    stride = conv.stride[0]
     if _CONV256 and c % 64 == 0 and b * h * w >= 256:
         y = ext.get_ext(True).conv256_nhwc(
             nhwc.to(torch.bfloat16), wt, bias, b, h, w, c,
-            conv.out_channels, rs, fuse_silu,
+            conv.out_channels, rs, stride, fuse_silu,
         )
         return y.permute(0, 3, 1, 2)
+    assert stride == 1, "strided conv needs the conv256 path"
     y = ext.get_ext(True).conv_nhwc(
         nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
         fuse_silu,

@@ -55,7 +55,9 @@ __device__ __forceinline__ void glds16(const void* gsrc, void* lds_dst) {
 }
 
 struct ConvGeo {
-  int H, W, C;     // input spatial + channels
+  int H, W, C;     // INPUT spatial + channels
+  int Ho, Wo;      // output spatial (Ho = H/stride for 3x3 pad1)
+  int stride;      // 1 or 2 (the UNet/VAE Downsample convs)
   int RS;          // 9 for 3x3 (pad 1), 1 for 1x1
   int tile2d;      // 1: blocks cover 16x16 pixel tiles (H,W % 16 == 0) —
                    // a 3x3 tap re-reads an 18x18 halo (1.27x) instead of
@@ -102,8 +104,8 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
   long long m_blk = 0;
   int t_b = 0, t_py0 = 0, t_px0 = 0;
   if (IS_CONV && geo.tile2d) {
-    const int tx_n = geo.W >> 4;
-    const int per_img = tx_n * (geo.H >> 4);
+    const int tx_n = geo.Wo >> 4;
+    const int per_img = tx_n * (geo.Ho >> 4);
     t_b = (int)(bx / per_img);
     const int rem = (int)(bx % per_img);
     t_py0 = (rem / tx_n) << 4;
@@ -148,13 +150,13 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
         } else if (IS_CONV) {
           const long long m = m_blk + m_local;
           d.ok = m < M;
-          const long long HW = (long long)geo.H * geo.W;
+          const long long HWo = (long long)geo.Ho * geo.Wo;
           const long long mm = d.ok ? m : 0;
-          const int pb = (int)(mm / HW);
-          const int rem = (int)(mm - (long long)pb * HW);
-          d.px_y = rem / geo.W;
-          d.px_x = rem % geo.W;
-          d.row_base = A + ((long long)pb * HW) * geo.C;
+          const int pb = (int)(mm / HWo);
+          const int rem = (int)(mm - (long long)pb * HWo);
+          d.px_y = rem / geo.Wo;   // OUTPUT pixel coords; stride applied
+          d.px_x = rem % geo.Wo;   // at the tap address
+          d.row_base = A + (long long)pb * geo.H * geo.W * geo.C;
         } else {
           const long long m = m_blk + m_local;
           d.ok = m < M;
@@ -211,8 +213,8 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
       } else {
         const AChunk& d = a_desc[pc][i];
         if (IS_CONV) {
-          const int sy = d.px_y + tap_dy;
-          const int sx = d.px_x + tap_dx;
+          const int sy = d.px_y * geo.stride + tap_dy;
+          const int sx = d.px_x * geo.stride + tap_dx;
           const bool ok = d.ok && sy >= 0 && sy < geo.H && sx >= 0 &&
                           sx < geo.W;
           src = ok ? d.row_base + ((long long)sy * geo.W + sx) * geo.C +
@@ -312,7 +314,7 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
         long long m;
         if (IS_CONV && geo.tile2d) {
           const int idx = wm * 128 + mf * 16 + kgrp * 4 + rr;
-          m = ((long long)t_b * geo.H + t_py0 + (idx >> 4)) * geo.W +
+          m = ((long long)t_b * geo.Ho + t_py0 + (idx >> 4)) * geo.Wo +
               t_px0 + (idx & 15);
         } else {
           m = m_wave + mf * 16 + kgrp * 4 + rr;
@@ -357,7 +359,7 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
   }
   dim3 grid((unsigned)((M + GBM - 1) / GBM), (unsigned)((N + GBN - 1) / GBN));
   auto stream = at::hip::getCurrentHIPStream();
-  ConvGeo geo{0, 0, 0, 0, 0, 0};
+  ConvGeo geo{0, 0, 0, 0, 0, 1, 0, 0, 0};
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<false, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
@@ -377,16 +379,21 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
                            torch::Tensor bias, int64_t B, int64_t H,
                            int64_t W, int64_t C, int64_t K, int64_t rs,
-                           bool fuse_silu) {
-  // x [B,H,W,C] bf16 NHWC, wt [K_out, rs*C] repacked -> y [B,H,W,K]
+                           int64_t stride, bool fuse_silu) {
+  // x [B,H,W,C] bf16 NHWC, wt [K_out, rs*C] repacked -> y [B,Ho,Wo,K]
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
               x.is_contiguous());
   TORCH_CHECK(wt.is_cuda() && wt.is_contiguous());
   TORCH_CHECK(rs == 9 || rs == 1);
+  TORCH_CHECK(stride == 1 || (stride == 2 && rs == 9));
   TORCH_CHECK(C % GBK == 0, "C must be a multiple of 64");
-  const long long M = B * H * W;
+  // 3x3 always pad 1: Ho = ceil(H/stride) (stride 2 needs even dims for
+  // the torch formula (H+2-3)/2+1 = H/2 when H even)
+  const int64_t Ho = stride == 1 ? H : (H + 2 - 3) / 2 + 1;
+  const int64_t Wo = stride == 1 ? W : (W + 2 - 3) / 2 + 1;
+  const long long M = B * Ho * Wo;
   const int Kdim = (int)(rs * C);
-  auto y = torch::empty({B, H, W, K}, x.options());
+  auto y = torch::empty({B, Ho, Wo, K}, x.options());
   const float* bptr = nullptr;
   torch::Tensor bf32;
   if (bias.defined() && bias.numel() > 0) {
@@ -397,13 +404,14 @@ torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
     const char* e = getenv("DISTGPU_CONV_XCDSWZ");
     return (!e || e[0] == '1') ? 1 : 0;
   }();
-  const int tile2d = (H % 16 == 0 && W % 16 == 0) ? 1 : 0;
+  const int tile2d = (Ho % 16 == 0 && Wo % 16 == 0) ? 1 : 0;
   const unsigned gx = tile2d
-      ? (unsigned)(B * (H / 16) * (W / 16))
+      ? (unsigned)(B * (Ho / 16) * (Wo / 16))
       : (unsigned)((M + GBM - 1) / GBM);
   dim3 grid(gx, (unsigned)((K + GBN - 1) / GBN));
   auto stream = at::hip::getCurrentHIPStream();
-  ConvGeo geo{(int)H, (int)W, (int)C, (int)rs, tile2d, xcd_swz};
+  ConvGeo geo{(int)H, (int)W, (int)C, (int)Ho, (int)Wo, (int)stride,
+              (int)rs, tile2d, xcd_swz};
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<true, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),

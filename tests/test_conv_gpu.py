@@ -142,7 +142,7 @@ def test_conv256_matches_fp32_conv():
             wkern = wkern[:, :, :1, :1].contiguous()
         wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
         b = torch.randn(K, device="cuda").to(torch.bfloat16)
-        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, rs, False)
+        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, rs, 1, False)
         xf = x.permute(0, 3, 1, 2).float()
         ref = torch.nn.functional.conv2d(
             xf, wkern.float(), b.float(), padding=1 if rs == 9 else 0)
@@ -150,3 +150,27 @@ def test_conv256_matches_fp32_conv():
         err = (y.float() - ref).abs().max().item()
         scale = ref.abs().max().item()
         assert err / scale < 0.02, f"conv {B}x{C}x{H}x{W}->{K} rs{rs}: {err/scale}"
+
+
+def test_conv256_stride2_matches_fp32_conv():
+    """Stride-2 3x3 (the UNet/VAE Downsample convs) on the 256-tile
+    kernel vs the fp32 torch reference."""
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    torch.manual_seed(3)
+    for (B, C, H, W, K) in [(2, 64, 32, 32, 96), (1, 128, 34, 34, 128),
+                            (2, 320, 20, 24, 320), (1, 64, 17, 19, 64)]:
+        x = (torch.randn(B, H, W, C, device="cuda") / 4).to(torch.bfloat16)
+        wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
+        wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
+        b = torch.randn(K, device="cuda").to(torch.bfloat16)
+        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, 9, 2, False)
+        xf = x.permute(0, 3, 1, 2).float()
+        ref = torch.nn.functional.conv2d(xf, wkern.float(), b.float(),
+                                         stride=2, padding=1)
+        ref = ref.permute(0, 2, 3, 1)
+        assert y.shape == ref.shape, (y.shape, ref.shape)
+        err = (y.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 0.02, f"stride2 {B}x{C}x{H}x{W}->{K}: {err/scale}"
