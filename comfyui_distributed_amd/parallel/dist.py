@@ -233,8 +233,14 @@ class ResultMailbox:
         self._seq += 1
 
     def flush(self) -> None:
-        """Wait for all outstanding sends (bounded)."""
-        deadline = datetime.timedelta(seconds=self.recv_timeout)
+        """Wait for all outstanding sends. The bound is deliberately
+        generous: a send only completes once rank 0 posts its receive, and
+        rank 0 may legitimately be inside a multi-minute first chunk
+        (MIOpen warmup, graph capture) before it drains — a short bound
+        here would crash healthy workers during warmup. A genuinely dead
+        rank 0 ends the whole job regardless."""
+        deadline = datetime.timedelta(
+            seconds=max(self.recv_timeout * 10.0, 900.0))
         for works, _p, _m in self._pending:
             for w in works:
                 w.wait(deadline)

@@ -16,6 +16,7 @@ tile assignment, requeues, or duplicated work after a crash.
 
 from __future__ import annotations
 
+import threading
 import time
 
 import torch
@@ -89,23 +90,45 @@ def run_distributed_usdu(
     mailbox = ResultMailbox(ctx, store, job_id, recv_timeout=recv_timeout)
 
     if not ctx.is_master:
+        # liveness heartbeat thread: a chunk's first pass can take MINUTES
+        # (MIOpen warmup, hipGraph capture) — per-chunk heartbeats alone
+        # would get a healthy rank dropped by the master's scheduler. A
+        # live process heartbeating == the reference's probe-alive grace
+        # (upscale/job_timeout.py:85-100); a SIGKILLed rank's thread dies
+        # with it, so crash detection is unaffected.
+        stop_hb = threading.Event()
+
+        def _hb_loop():
+            while not stop_hb.wait(5.0):
+                try:
+                    queue.heartbeat()
+                except Exception:
+                    return
+
+        hb_thread = threading.Thread(target=_hb_loop, daemon=True)
+        hb_thread.start()
         done = 0
-        while True:
-            ids = _pop_chunk(queue, ids_per_iter)
-            if not ids:
-                break
-            res = sample_tiles(stack, cond, uncond, params, canvas, plans, ids)
-            items = sorted(res.items())
-            # post the chunk's sends BEFORE marking done: a crash between
-            # the two leaves tiles un-done -> the scheduler requeues them
-            mailbox.send_chunk([img[0] for (_tb, img) in items],
-                               [tb for (tb, _img) in items])
-            for idx in ids:
-                queue.mark_done(idx)
+        try:
+            while True:
+                ids = _pop_chunk(queue, ids_per_iter)
+                if not ids:
+                    break
+                res = sample_tiles(stack, cond, uncond, params, canvas,
+                                   plans, ids)
+                items = sorted(res.items())
+                # post the chunk's sends BEFORE marking done: a crash
+                # between the two leaves tiles un-done -> the scheduler
+                # requeues them
+                mailbox.send_chunk([img[0] for (_tb, img) in items],
+                                   [tb for (tb, _img) in items])
+                for idx in ids:
+                    queue.mark_done(idx)
+                queue.heartbeat()
+                done += len(ids)
+            mailbox.finish()
             queue.heartbeat()
-            done += len(ids)
-        mailbox.finish()
-        queue.heartbeat()
+        finally:
+            stop_hb.set()
         debug_log(f"rank {ctx.rank}: processed {done} tiles")
         return None
 
