@@ -233,7 +233,7 @@ class StemConv(nn.Module):
 class Downsample(nn.Module):
     def __init__(self, channels):
         super().__init__()
-        self.conv = nn.Conv2d(channels, channels, 3, stride=2, padding=1)
+        self.conv = MfmaConv2d(channels, channels, 3, stride=2, padding=1)
 
     def forward(self, x):
         return self.conv(x)
@@ -337,7 +337,7 @@ class UNetModel(nn.Module):
                 self.output_blocks.append(_TimestepSequential(layers))
 
         self.out_norm = FusedGroupNorm(ch0, silu=True)
-        self.out_conv = nn.Conv2d(ch0, cfg.out_channels, 3, padding=1)
+        self.out_conv = MfmaConv2d(ch0, cfg.out_channels, 3, padding=1)
 
     def forward(self, x, timesteps, context, y=None):
         dtype = self.out_conv.weight.dtype

@@ -110,7 +110,7 @@ class VAEEncoder(nn.Module):
             [VAEResBlock(cin, cin), VAEAttention(cin), VAEResBlock(cin, cin)]
         )
         self.norm_out = FusedGroupNorm(cin, silu=True)
-        self.conv_out = nn.Conv2d(cin, cfg.latent_channels * 2, 3, padding=1)
+        self.conv_out = MfmaConv2d(cin, cfg.latent_channels * 2, 3, padding=1)
 
     def forward(self, x):
         h = self.conv_in(x)
@@ -140,7 +140,7 @@ class VAEDecoder(nn.Module):
                 ups.append(_DecoderUpsample(cin))
         self.up = nn.ModuleList(ups)
         self.norm_out = FusedGroupNorm(cin, silu=True)
-        self.conv_out = nn.Conv2d(cin, cfg.in_channels, 3, padding=1)
+        self.conv_out = MfmaConv2d(cin, cfg.in_channels, 3, padding=1)
 
     def forward(self, z):
         h = self.conv_in(z)
@@ -154,7 +154,7 @@ class VAEDecoder(nn.Module):
 class _DecoderUpsample(nn.Module):
     def __init__(self, channels):
         super().__init__()
-        self.conv = nn.Conv2d(channels, channels, 3, padding=1)
+        self.conv = MfmaConv2d(channels, channels, 3, padding=1)
 
     def forward(self, x):
         return self.conv(F.interpolate(x, scale_factor=2, mode="nearest"))

@@ -97,10 +97,15 @@ def conv_supported(conv) -> bool:
         # keeping it off MIOpen also kills its exhaustive-find warmup
         conv.stride == (2, 2) and k == (3, 3) and conv.in_channels % 64 == 0
     )
+    # the 256-tile kernel bounds-masks any K_out (the UNet's K=4 out_conv
+    # and the VAE's K=3/K=8 edge convs included — MIOpen's heuristic picks
+    # for those are pathological); the v1/v2 kernels need K % 16
+    kout_ok = (conv.out_channels % 16 == 0
+               or (_CONV256 and conv.in_channels % 64 == 0))
     return (
         stride_ok
         and conv.in_channels % 32 == 0
-        and conv.out_channels % 16 == 0
+        and kout_ok
         and ((k == (3, 3) and conv.padding == (1, 1))
              or (k == (1, 1) and conv.padding == (0, 0)))
         and conv.dilation == (1, 1)
