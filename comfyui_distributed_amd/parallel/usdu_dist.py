@@ -11,7 +11,11 @@ takes over any tile whose result never arrived (reference master-takeover,
 upscale/modes/static.py:354-363,469-513). Determinism: extraction reads the
 original canvas and blending runs in ascending (tile, batch) order on
 rank 0, so the result is bit-identical to the single-GPU run regardless of
-tile assignment, requeues, or duplicated work after a crash.
+tile assignment, requeues, or duplicated work after a crash. (On GPU,
+"bit-identical" is modulo one source of run-to-run noise that is
+independent of assignment: the GroupNorm stats kernel's fp32 atomic
+reduction order, ~1e-2 through a bf16 sampler chain — see
+tests/test_distributed_gpu.py. The CPU path is exactly deterministic.)
 """
 
 from __future__ import annotations
