@@ -290,10 +290,13 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
     compute_quadrant(p, 0, 0);
     if (more) stage_piece(kt + 1, 1 - p, 1);
     compute_quadrant(p, 0, 1);
-    if (more) {
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      stage_piece(kt + 1, 1 - p, 2);
-    }
+    // A1's landing must be COLLECTIVE before any wave reads it: each
+    // wave's counted vmcnt only drains its OWN staging DMA, so the
+    // barrier after the wait is what guarantees every wave's chunks are
+    // in LDS (the guide's template always pairs vmcnt with a barrier)
+    if (more) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (more) stage_piece(kt + 1, 1 - p, 2);
     compute_quadrant(p, 1, 0);
     if (more) stage_piece(kt + 1, 1 - p, 3);
     compute_quadrant(p, 1, 1);
