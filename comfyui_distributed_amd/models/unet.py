@@ -248,6 +248,12 @@ class Upsample(nn.Module):
         # odd latent sizes (e.g. 68->34->17->9) need the exact skip shape on
         # the way back up, not blind 2x (ComfyUI passes output_shape the
         # same way)
+        exact_2x = (output_shape is None
+                    or tuple(output_shape) == (2 * x.shape[2], 2 * x.shape[3]))
+        if (exact_2x and x.is_cuda
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and ops.conv_supported(self.conv)):
+            return ops.conv2d_mfma(x, self.conv, up2=True)
         if output_shape is not None:
             y = F.interpolate(x, size=tuple(output_shape), mode="nearest")
         else:

@@ -157,6 +157,12 @@ class _DecoderUpsample(nn.Module):
         self.conv = MfmaConv2d(channels, channels, 3, padding=1)
 
     def forward(self, x):
+        if (x.is_cuda and x.is_contiguous(memory_format=torch.channels_last)
+                and ops.conv_supported(self.conv)):
+            # nearest-2x fused into the conv's tap addressing: the 4x-sized
+            # upsampled intermediate (the decoder's largest tensors) never
+            # touches HBM
+            return ops.conv2d_mfma(x, self.conv, up2=True)
         return self.conv(F.interpolate(x, scale_factor=2, mode="nearest"))
 
 

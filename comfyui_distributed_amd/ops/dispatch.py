@@ -131,10 +131,13 @@ _CONV256 = os.environ.get("DISTGPU_CONV256", "1") == "1"
 _GEMM256 = os.environ.get("DISTGPU_GEMM256", "0") == "1"
 
 
-def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
+def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
+                up2: bool = False) -> torch.Tensor:
     """x: NCHW tensor in channels_last memory format (bf16, on GPU) ->
     same layout. Dispatches to the implicit-GEMM NHWC kernel: the 256-tile
-    glds template (gemm.hip) when shapes allow, else the v1/v2 kernels."""
+    glds template (gemm.hip) when shapes allow, else the v1/v2 kernels.
+    ``up2`` fuses a nearest-2x upsample into the conv's tap addressing
+    (the upsampled tensor never materializes)."""
     assert x.is_cuda
     b, c, h, w = x.shape
     nhwc = x.permute(0, 2, 3, 1)  # view: contiguous when x is channels_last
@@ -144,13 +147,13 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tensor:
     rs = 9 if conv.kernel_size == (3, 3) else 1
     bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
     stride = conv.stride[0]
-    if _CONV256 and c % 64 == 0 and b * h * w >= 256:
+    if _CONV256 and c % 64 == 0 and (b * h * w >= 256 or up2):
         y = ext.get_ext(True).conv256_nhwc(
             nhwc.to(torch.bfloat16), wt, bias, b, h, w, c,
-            conv.out_channels, rs, stride, fuse_silu,
+            conv.out_channels, rs, stride, up2, fuse_silu,
         )
         return y.permute(0, 3, 1, 2)
-    assert stride == 1, "strided conv needs the conv256 path"
+    assert stride == 1 and not up2, "strided/up2 conv needs the conv256 path"
     y = ext.get_ext(True).conv_nhwc(
         nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
         fuse_silu,

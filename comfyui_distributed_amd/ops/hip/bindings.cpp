@@ -36,7 +36,7 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
                            torch::Tensor bias, int64_t B, int64_t H,
                            int64_t W, int64_t C, int64_t K, int64_t rs,
-                           int64_t stride, bool fuse_silu);
+                           int64_t stride, bool up2, bool fuse_silu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_fused", &group_norm_fused,

@@ -58,6 +58,11 @@ struct ConvGeo {
   int H, W, C;     // INPUT spatial + channels
   int Ho, Wo;      // output spatial (Ho = H/stride for 3x3 pad1)
   int stride;      // 1 or 2 (the UNet/VAE Downsample convs)
+  int up2;         // 1: conv consumes a VIRTUAL nearest-2x upsample of x
+                   // (Ho = 2H) — the upsampled tensor never exists, the
+                   // tap address just halves: the VAE decoder's biggest
+                   // intermediates (F.interpolate output + its re-read)
+                   // disappear entirely
   int RS;          // 9 for 3x3 (pad 1), 1 for 1x1
   int tile2d;      // 1: blocks cover 16x16 pixel tiles (H,W % 16 == 0) —
                    // a 3x3 tap re-reads an 18x18 halo (1.27x) instead of
@@ -213,10 +218,21 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
       } else {
         const AChunk& d = a_desc[pc][i];
         if (IS_CONV) {
-          const int sy = d.px_y * geo.stride + tap_dy;
-          const int sx = d.px_x * geo.stride + tap_dx;
-          const bool ok = d.ok && sy >= 0 && sy < geo.H && sx >= 0 &&
-                          sx < geo.W;
+          int sy, sx;
+          bool ok;
+          if (geo.up2) {
+            // pad is applied on the VIRTUAL upsampled canvas [2H, 2W]
+            const int uy = d.px_y + tap_dy;
+            const int ux = d.px_x + tap_dx;
+            ok = d.ok && uy >= 0 && uy < 2 * geo.H && ux >= 0 &&
+                 ux < 2 * geo.W;
+            sy = uy >> 1;
+            sx = ux >> 1;
+          } else {
+            sy = d.px_y * geo.stride + tap_dy;
+            sx = d.px_x * geo.stride + tap_dx;
+            ok = d.ok && sy >= 0 && sy < geo.H && sx >= 0 && sx < geo.W;
+          }
           src = ok ? d.row_base + ((long long)sy * geo.W + sx) * geo.C +
                          cbase + c_local[pc][i]
                    : zero_page;
@@ -362,7 +378,7 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
   }
   dim3 grid((unsigned)((M + GBM - 1) / GBM), (unsigned)((N + GBN - 1) / GBN));
   auto stream = at::hip::getCurrentHIPStream();
-  ConvGeo geo{0, 0, 0, 0, 0, 1, 0, 0, 0};
+  ConvGeo geo{0, 0, 0, 0, 0, 1, 0, 0, 0, 0};
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<false, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
@@ -382,18 +398,19 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
                            torch::Tensor bias, int64_t B, int64_t H,
                            int64_t W, int64_t C, int64_t K, int64_t rs,
-                           int64_t stride, bool fuse_silu) {
+                           int64_t stride, bool up2, bool fuse_silu) {
   // x [B,H,W,C] bf16 NHWC, wt [K_out, rs*C] repacked -> y [B,Ho,Wo,K]
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
               x.is_contiguous());
   TORCH_CHECK(wt.is_cuda() && wt.is_contiguous());
   TORCH_CHECK(rs == 9 || rs == 1);
   TORCH_CHECK(stride == 1 || (stride == 2 && rs == 9));
+  TORCH_CHECK(!up2 || (stride == 1 && rs == 9));
   TORCH_CHECK(C % GBK == 0, "C must be a multiple of 64");
   // 3x3 always pad 1: Ho = ceil(H/stride) (stride 2 needs even dims for
-  // the torch formula (H+2-3)/2+1 = H/2 when H even)
-  const int64_t Ho = stride == 1 ? H : (H + 2 - 3) / 2 + 1;
-  const int64_t Wo = stride == 1 ? W : (W + 2 - 3) / 2 + 1;
+  // the torch formula (H+2-3)/2+1 = H/2 when H even); up2 doubles
+  const int64_t Ho = up2 ? 2 * H : (stride == 1 ? H : (H + 2 - 3) / 2 + 1);
+  const int64_t Wo = up2 ? 2 * W : (stride == 1 ? W : (W + 2 - 3) / 2 + 1);
   const long long M = B * Ho * Wo;
   const int Kdim = (int)(rs * C);
   auto y = torch::empty({B, Ho, Wo, K}, x.options());
@@ -414,7 +431,7 @@ torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
   dim3 grid(gx, (unsigned)((K + GBN - 1) / GBN));
   auto stream = at::hip::getCurrentHIPStream();
   ConvGeo geo{(int)H, (int)W, (int)C, (int)Ho, (int)Wo, (int)stride,
-              (int)rs, tile2d, xcd_swz};
+              up2 ? 1 : 0, (int)rs, tile2d, xcd_swz};
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<true, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),

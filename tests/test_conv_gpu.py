@@ -142,7 +142,7 @@ def test_conv256_matches_fp32_conv():
             wkern = wkern[:, :, :1, :1].contiguous()
         wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
         b = torch.randn(K, device="cuda").to(torch.bfloat16)
-        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, rs, 1, False)
+        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, rs, 1, False, False)
         xf = x.permute(0, 3, 1, 2).float()
         ref = torch.nn.functional.conv2d(
             xf, wkern.float(), b.float(), padding=1 if rs == 9 else 0)
@@ -165,7 +165,7 @@ def test_conv256_stride2_matches_fp32_conv():
         wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
         wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
         b = torch.randn(K, device="cuda").to(torch.bfloat16)
-        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, 9, 2, False)
+        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, 9, 2, False, False)
         xf = x.permute(0, 3, 1, 2).float()
         ref = torch.nn.functional.conv2d(xf, wkern.float(), b.float(),
                                          stride=2, padding=1)
@@ -174,3 +174,28 @@ def test_conv256_stride2_matches_fp32_conv():
         err = (y.float() - ref).abs().max().item()
         scale = ref.abs().max().item()
         assert err / scale < 0.02, f"stride2 {B}x{C}x{H}x{W}->{K}: {err/scale}"
+
+
+def test_conv256_fused_upsample_matches_interpolate_conv():
+    """up2 mode: conv of a VIRTUAL nearest-2x upsample vs
+    F.interpolate + conv fp32 reference (VAE decoder upsample fusion)."""
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    torch.manual_seed(5)
+    for (B, C, H, W, K) in [(2, 64, 16, 16, 96), (1, 128, 17, 19, 128),
+                            (2, 128, 24, 24, 64)]:
+        x = (torch.randn(B, H, W, C, device="cuda") / 4).to(torch.bfloat16)
+        wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
+        wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
+        b = torch.randn(K, device="cuda").to(torch.bfloat16)
+        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, 9, 1, True, False)
+        xf = x.permute(0, 3, 1, 2).float()
+        up = torch.nn.functional.interpolate(xf, scale_factor=2,
+                                             mode="nearest")
+        ref = torch.nn.functional.conv2d(up, wkern.float(), b.float(),
+                                         padding=1).permute(0, 2, 3, 1)
+        assert y.shape == ref.shape
+        err = (y.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 0.02, f"up2 {B}x{C}x{H}x{W}->{K}: {err/scale}"
