@@ -297,21 +297,18 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
   for (int kt = 0; kt < n_ktiles; ++kt) {
     const int p = kt & 1;
     const bool more = kt + 1 < n_ktiles;
-    if (more)
-      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    // Full drain at tile start: the tile's 8 glds were issued a whole
+    // K-tile ago, so vmcnt(0) is cheap here — and the barrier AFTER the
+    // drain makes every wave's staging collectively visible (a counted
+    // per-wave vmcnt alone cannot order OTHER waves' DMA against this
+    // wave's ds_reads; the earlier staggered vmcnt(2)/(4) scheme needed a
+    // second barrier for that and measured slower)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     if (more) stage_piece(kt + 1, 1 - p, 0);
     compute_quadrant(p, 0, 0);
     if (more) stage_piece(kt + 1, 1 - p, 1);
     compute_quadrant(p, 0, 1);
-    // A1's landing must be COLLECTIVE before any wave reads it: each
-    // wave's counted vmcnt only drains its OWN staging DMA, so the
-    // barrier after the wait is what guarantees every wave's chunks are
-    // in LDS (the guide's template always pairs vmcnt with a barrier)
-    if (more) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
     if (more) stage_piece(kt + 1, 1 - p, 2);
     compute_quadrant(p, 1, 0);
     if (more) stage_piece(kt + 1, 1 - p, 3);
