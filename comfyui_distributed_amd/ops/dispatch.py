@@ -25,6 +25,22 @@ def hip_available() -> bool:
     return ext.get_ext(required=False) is not None
 
 
+def _f32(t: torch.Tensor | None):
+    """fp32 view of a (usually bf16) parameter, cached ON the tensor object
+    — the GN/conv launchers used to re-cast weight+bias on every call:
+    ~27k bf16->f32 copy kernels per flagship canvas (rocprof r02)."""
+    if t is None or t.numel() == 0 or t.dtype == torch.float32:
+        return t
+    c = getattr(t, "_distgpu_f32", None)
+    if c is None or c.device != t.device:
+        c = t.detach().float().contiguous()
+        try:
+            t._distgpu_f32 = c
+        except Exception:  # non-writable tensor subclass: fall back
+            pass
+    return c
+
+
 def _on_gpu(*tensors: torch.Tensor) -> bool:
     return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
 
@@ -50,7 +66,7 @@ def group_norm_silu(
         ):
             return group_norm_silu_cl(x, groups, weight, bias, eps, silu)
         return ext.get_ext(True).group_norm_fused(
-            x.to(torch.bfloat16), groups, weight, bias, eps, silu
+            x.to(torch.bfloat16), groups, _f32(weight), _f32(bias), eps, silu
         )
     # manual GN (F.group_norm rejects 1-value-per-group shapes, e.g. a
     # batch-1 tensor at 1x1 spatial in deep tiny-config levels)
@@ -145,7 +161,7 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
         nhwc = nhwc.contiguous()
     wt = _repacked_weight(conv)
     rs = 9 if conv.kernel_size == (3, 3) else 1
-    bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
+    bias = _f32(conv.bias) if conv.bias is not None else torch.empty(0, device=x.device)
     stride = conv.stride[0]
     if _CONV256 and c % 64 == 0 and (b * h * w >= 256 or up2):
         y = ext.get_ext(True).conv256_nhwc(
@@ -196,7 +212,7 @@ def conv2d_smallc(x: torch.Tensor, conv, fuse_silu: bool = False) -> torch.Tenso
         nhwc = nhwc.contiguous()
     wt = _repacked_weight(conv)
     rs = 9 if conv.kernel_size == (3, 3) else 1
-    bias = conv.bias if conv.bias is not None else torch.empty(0, device=x.device)
+    bias = _f32(conv.bias) if conv.bias is not None else torch.empty(0, device=x.device)
     y = ext.get_ext(True).conv_smallc(
         nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
         fuse_silu,
@@ -212,7 +228,7 @@ def group_norm_silu_cl(x: torch.Tensor, groups: int, weight, bias,
     if not nhwc.is_contiguous():
         nhwc = nhwc.contiguous()
     y = ext.get_ext(True).group_norm_nhwc(
-        nhwc.to(torch.bfloat16), groups, weight, bias, eps, silu
+        nhwc.to(torch.bfloat16), groups, _f32(weight), _f32(bias), eps, silu
     )
     return y.permute(0, 3, 1, 2)
 
