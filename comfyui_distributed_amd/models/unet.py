@@ -124,8 +124,12 @@ class ResBlock(nn.Module):
     def forward(self, x, emb):
         h = self.conv1(self.norm1(x))
         h = h + self.emb_proj(F.silu(emb))[:, :, None, None]
-        h = self.conv2(self.norm2(h))
-        return h + self.skip(x)
+        h2 = self.norm2(h)
+        skip = self.skip(x)
+        if (h2.is_cuda and h2.is_contiguous(memory_format=torch.channels_last)
+                and ops.conv_supported(self.conv2)):
+            return ops.conv2d_mfma(h2, self.conv2, residual=skip)
+        return self.conv2(h2) + skip
 
 
 class CrossAttention(nn.Module):

@@ -60,9 +60,9 @@ class VAEResBlock(nn.Module):
             h = ops.conv2d_mfma(h, self.conv1)
             h = ops.group_norm_silu_cl(h, self.norm2.groups, self.norm2.weight,
                                        self.norm2.bias, 1e-5, True)
-            h = ops.conv2d_mfma(h, self.conv2)
             skip = x if isinstance(self.skip, nn.Identity) else ops.conv2d_mfma(x, self.skip)
-            return h + skip
+            # skip add fused into conv2's epilogue (no separate add pass)
+            return ops.conv2d_mfma(h, self.conv2, residual=skip)
         h = self.conv2(self.norm2(self.conv1(self.norm1(x))))
         return h + self.skip(x)
 

@@ -148,12 +148,15 @@ _GEMM256 = os.environ.get("DISTGPU_GEMM256", "0") == "1"
 
 
 def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
-                up2: bool = False) -> torch.Tensor:
+                up2: bool = False,
+                residual: torch.Tensor | None = None) -> torch.Tensor:
     """x: NCHW tensor in channels_last memory format (bf16, on GPU) ->
     same layout. Dispatches to the implicit-GEMM NHWC kernel: the 256-tile
     glds template (gemm.hip) when shapes allow, else the v1/v2 kernels.
     ``up2`` fuses a nearest-2x upsample into the conv's tap addressing
-    (the upsampled tensor never materializes)."""
+    (the upsampled tensor never materializes). ``residual`` (a
+    channels_last NCHW tensor of the OUTPUT shape) is added in the
+    epilogue — the ResBlock skip connection without its own kernel."""
     assert x.is_cuda
     b, c, h, w = x.shape
     nhwc = x.permute(0, 2, 3, 1)  # view: contiguous when x is channels_last
@@ -164,12 +167,21 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
     bias = _f32(conv.bias) if conv.bias is not None else torch.empty(0, device=x.device)
     stride = conv.stride[0]
     if _CONV256 and c % 64 == 0 and (b * h * w >= 256 or up2):
+        if residual is not None:
+            res = residual.permute(0, 2, 3, 1)
+            if not res.is_contiguous():
+                res = res.contiguous()
+            res = res.to(torch.bfloat16)
+        else:
+            res = torch.empty(0, device=x.device, dtype=torch.bfloat16)
         y = ext.get_ext(True).conv256_nhwc(
-            nhwc.to(torch.bfloat16), wt, bias, b, h, w, c,
+            nhwc.to(torch.bfloat16), wt, bias, res, b, h, w, c,
             conv.out_channels, rs, stride, up2, fuse_silu,
         )
         return y.permute(0, 3, 1, 2)
     assert stride == 1 and not up2, "strided/up2 conv needs the conv256 path"
+    if residual is not None:
+        return conv2d_mfma(x, conv, fuse_silu) + residual
     y = ext.get_ext(True).conv_nhwc(
         nhwc.to(torch.bfloat16), wt, bias, b, h, w, c, conv.out_channels, rs,
         fuse_silu,

@@ -34,7 +34,8 @@ torch::Tensor conv_smallc(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
 torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
                            torch::Tensor bias, bool fuse_silu);
 torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
-                           torch::Tensor bias, int64_t B, int64_t H,
+                           torch::Tensor bias, torch::Tensor residual,
+                           int64_t B, int64_t H,
                            int64_t W, int64_t C, int64_t K, int64_t rs,
                            int64_t stride, bool up2, bool fuse_silu);
 

@@ -93,6 +93,8 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
     const uint16_t* __restrict__ Bw,  // [N, K] (conv: repacked [K_out][RS*C])
     const float* __restrict__ bias,   // [N] or nullptr
     const uint16_t* __restrict__ zero_page,  // >=16B of zeros
+    const uint16_t* __restrict__ residual,   // [M, N] bf16 or nullptr:
+                                             // fused skip-connection add
     uint16_t* __restrict__ Y,         // [M, N] bf16
     long long M, int N, int Kdim, ConvGeo geo) {
   const int tid = threadIdx.x;
@@ -338,6 +340,7 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
         }
         float v = acc[mf][nf][rr] + bval;
         if (FUSE_SILU) v = silu_f(v);
+        if (residual) v += bf16_bits_to_f32(residual[m * N + n]);
         Y[m * N + n] = f32_to_bf16_bits(v);
       }
     }
@@ -358,6 +361,7 @@ static const uint16_t* zero_page_ptr(const torch::Tensor& like) {
 
 torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
                            torch::Tensor bias, bool fuse_silu) {
+  // (plain-GEMM entry has no residual fusion)
   // x [M, K] bf16, w [N, K] bf16 (torch Linear layout) -> y [M, N] bf16
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
               x.is_contiguous());
@@ -380,20 +384,21 @@ torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
     hipLaunchKernelGGL((gemm256_kernel<false, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
                        (const uint16_t*)w.data_ptr(), bptr,
-                       zero_page_ptr(x), (uint16_t*)y.data_ptr(), M, N, K,
-                       geo);
+                       zero_page_ptr(x), (const uint16_t*)nullptr,
+                       (uint16_t*)y.data_ptr(), M, N, K, geo);
   else
     hipLaunchKernelGGL((gemm256_kernel<false, false>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
                        (const uint16_t*)w.data_ptr(), bptr,
-                       zero_page_ptr(x), (uint16_t*)y.data_ptr(), M, N, K,
-                       geo);
+                       zero_page_ptr(x), (const uint16_t*)nullptr,
+                       (uint16_t*)y.data_ptr(), M, N, K, geo);
   HIP_CHECK_LAUNCH();
   return y;
 }
 
 torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
-                           torch::Tensor bias, int64_t B, int64_t H,
+                           torch::Tensor bias, torch::Tensor residual,
+                           int64_t B, int64_t H,
                            int64_t W, int64_t C, int64_t K, int64_t rs,
                            int64_t stride, bool up2, bool fuse_silu) {
   // x [B,H,W,C] bf16 NHWC, wt [K_out, rs*C] repacked -> y [B,Ho,Wo,K]
@@ -429,18 +434,26 @@ torch::Tensor conv256_nhwc(torch::Tensor x, torch::Tensor wt,
   auto stream = at::hip::getCurrentHIPStream();
   ConvGeo geo{(int)H, (int)W, (int)C, (int)Ho, (int)Wo, (int)stride,
               up2 ? 1 : 0, (int)rs, tile2d, xcd_swz};
+  const uint16_t* resptr = nullptr;
+  if (residual.defined() && residual.numel() > 0) {
+    TORCH_CHECK(residual.is_cuda() && residual.is_contiguous() &&
+                residual.scalar_type() == at::kBFloat16 &&
+                residual.numel() == M * K,
+                "residual must be a contiguous bf16 [B,Ho,Wo,K] tensor");
+    resptr = (const uint16_t*)residual.data_ptr();
+  }
   if (fuse_silu)
     hipLaunchKernelGGL((gemm256_kernel<true, true>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
                        (const uint16_t*)wt.data_ptr(), bptr,
-                       zero_page_ptr(x), (uint16_t*)y.data_ptr(), M, (int)K,
-                       Kdim, geo);
+                       zero_page_ptr(x), resptr,
+                       (uint16_t*)y.data_ptr(), M, (int)K, Kdim, geo);
   else
     hipLaunchKernelGGL((gemm256_kernel<true, false>), grid, dim3(512), 0,
                        stream, (const uint16_t*)x.data_ptr(),
                        (const uint16_t*)wt.data_ptr(), bptr,
-                       zero_page_ptr(x), (uint16_t*)y.data_ptr(), M, (int)K,
-                       Kdim, geo);
+                       zero_page_ptr(x), resptr,
+                       (uint16_t*)y.data_ptr(), M, (int)K, Kdim, geo);
   HIP_CHECK_LAUNCH();
   return y;
 }

@@ -7,6 +7,10 @@ import torch.nn.functional as F
 pytestmark = pytest.mark.gpu
 
 
+def _nores():
+    return torch.empty(0, device="cuda", dtype=torch.bfloat16)
+
+
 @pytest.fixture(scope="module")
 def extmod():
     from comfyui_distributed_amd.ops import ext
@@ -142,7 +146,7 @@ def test_conv256_matches_fp32_conv():
             wkern = wkern[:, :, :1, :1].contiguous()
         wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
         b = torch.randn(K, device="cuda").to(torch.bfloat16)
-        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, rs, 1, False, False)
+        y = mod.conv256_nhwc(x, wt, b, _nores(), B, H, W, C, K, rs, 1, False, False)
         xf = x.permute(0, 3, 1, 2).float()
         ref = torch.nn.functional.conv2d(
             xf, wkern.float(), b.float(), padding=1 if rs == 9 else 0)
@@ -165,7 +169,7 @@ def test_conv256_stride2_matches_fp32_conv():
         wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
         wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
         b = torch.randn(K, device="cuda").to(torch.bfloat16)
-        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, 9, 2, False, False)
+        y = mod.conv256_nhwc(x, wt, b, _nores(), B, H, W, C, K, 9, 2, False, False)
         xf = x.permute(0, 3, 1, 2).float()
         ref = torch.nn.functional.conv2d(xf, wkern.float(), b.float(),
                                          stride=2, padding=1)
@@ -189,7 +193,7 @@ def test_conv256_fused_upsample_matches_interpolate_conv():
         wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
         wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
         b = torch.randn(K, device="cuda").to(torch.bfloat16)
-        y = mod.conv256_nhwc(x, wt, b, B, H, W, C, K, 9, 1, True, False)
+        y = mod.conv256_nhwc(x, wt, b, _nores(), B, H, W, C, K, 9, 1, True, False)
         xf = x.permute(0, 3, 1, 2).float()
         up = torch.nn.functional.interpolate(xf, scale_factor=2,
                                              mode="nearest")
@@ -199,3 +203,24 @@ def test_conv256_fused_upsample_matches_interpolate_conv():
         err = (y.float() - ref).abs().max().item()
         scale = ref.abs().max().item()
         assert err / scale < 0.02, f"up2 {B}x{C}x{H}x{W}->{K}: {err/scale}"
+
+
+def test_conv256_fused_residual_add():
+    """Epilogue residual fusion (the ResBlock skip add) vs conv + add."""
+    from comfyui_distributed_amd.ops import ext
+
+    mod = ext.get_ext(True)
+    torch.manual_seed(9)
+    B, C, H, W, K = 2, 64, 20, 20, 96
+    x = (torch.randn(B, H, W, C, device="cuda") / 4).to(torch.bfloat16)
+    wkern = (torch.randn(K, C, 3, 3, device="cuda") / 8).to(torch.bfloat16)
+    wt = wkern.permute(0, 2, 3, 1).reshape(K, -1).contiguous()
+    b = torch.randn(K, device="cuda").to(torch.bfloat16)
+    res = torch.randn(B, H, W, K, device="cuda").to(torch.bfloat16)
+    y = mod.conv256_nhwc(x, wt, b, res.contiguous(), B, H, W, C, K, 9, 1,
+                         False, False)
+    ref = torch.nn.functional.conv2d(
+        x.permute(0, 3, 1, 2).float(), wkern.float(), b.float(), padding=1
+    ).permute(0, 2, 3, 1) + res.float()
+    err = (y.float() - ref).abs().max().item()
+    assert err / ref.abs().max().item() < 0.03, err
