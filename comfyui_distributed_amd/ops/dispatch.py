@@ -145,6 +145,8 @@ _CONV256 = os.environ.get("DISTGPU_CONV256", "1") == "1"
 # TF vs our 205-1009) — the 256-tile GEMM stays available for fused uses and
 # future skinny-N tiles but is opt-in for nn.Linear routing
 _GEMM256 = os.environ.get("DISTGPU_GEMM256", "0") == "1"
+# epilogue residual fusion (ResBlock skip add) — A/B gate
+_FUSE_RES = os.environ.get("DISTGPU_FUSE_RESIDUAL", "1") == "1"
 
 
 def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
@@ -166,6 +168,8 @@ def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
     rs = 9 if conv.kernel_size == (3, 3) else 1
     bias = _f32(conv.bias) if conv.bias is not None else torch.empty(0, device=x.device)
     stride = conv.stride[0]
+    if residual is not None and not _FUSE_RES:
+        return conv2d_mfma(x, conv, fuse_silu, up2) + residual
     if _CONV256 and c % 64 == 0 and (b * h * w >= 256 or up2):
         if residual is not None:
             res = residual.permute(0, 2, 3, 1)
