@@ -145,8 +145,12 @@ _CONV256 = os.environ.get("DISTGPU_CONV256", "1") == "1"
 # TF vs our 205-1009) — the 256-tile GEMM stays available for fused uses and
 # future skinny-N tiles but is opt-in for nn.Linear routing
 _GEMM256 = os.environ.get("DISTGPU_GEMM256", "0") == "1"
-# epilogue residual fusion (ResBlock skip add) — A/B gate
-_FUSE_RES = os.environ.get("DISTGPU_FUSE_RESIDUAL", "1") == "1"
+# epilogue residual fusion (ResBlock skip add): measured 1.5% SLOWER than
+# the separate eager add on the flagship (same-box A/B, gpurun_out/call17:
+# 7.915 vs 8.032 tiles/s — the dependent per-element load makes the
+# store-issue-bound epilogue longer than the well-overlapped add pass).
+# Kept available for shapes where it may win; off by default.
+_FUSE_RES = os.environ.get("DISTGPU_FUSE_RESIDUAL", "0") == "1"
 
 
 def conv2d_mfma(x: torch.Tensor, conv, fuse_silu: bool = False,
