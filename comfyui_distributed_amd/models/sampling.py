@@ -104,6 +104,18 @@ class NoiseSchedule:
         return s[-(steps + 1):]
 
 
+#: cached graphs per wrapper before the least-recently-used is evicted —
+#: each entry pins static I/O buffers plus a private memory pool, so an
+#: unbounded dict would grow without limit on a long-running server that
+#: sees many distinct shapes/configs
+GRAPH_CACHE_CAP = 8
+
+
+def _evict_lru(graphs: dict):
+    if len(graphs) > GRAPH_CACHE_CAP:
+        graphs.pop(next(iter(graphs)))
+
+
 class GraphedModel:
     """hipGraph-captured model wrapper: one graph per input-shape set.
 
@@ -128,7 +140,7 @@ class GraphedModel:
             tuple(x.shape), tuple(ctx.shape),
             None if y is None else tuple(y.shape), x.dtype,
         )
-        entry = self.graphs.get(key)
+        entry = self.graphs.pop(key, None)
         if entry is None:
             try:
                 entry = self._capture(x, t, ctx, y)
@@ -138,7 +150,8 @@ class GraphedModel:
                 warnings.warn(f"hipGraph capture failed, running eager: {exc!r}")
                 self._failed = True
                 return self.model(x, t, ctx, y=y) if y is not None else self.model(x, t, ctx)
-            self.graphs[key] = entry
+        self.graphs[key] = entry  # re-insert = most recently used
+        _evict_lru(self.graphs)
         sx, st, sc, sy, sout, g = entry
         sx.copy_(x)
         st.copy_(t)
@@ -285,7 +298,7 @@ class GraphedSamplerLoop:
             None if y_c is None else tuple(y_c.shape),
             start_from_latent is not None,
         )
-        entry = self.graphs.get(key)
+        entry = self.graphs.pop(key, None)
         if entry is None:
             try:
                 entry = self._capture(key, cond, uncond, cfg_scale, noise,
@@ -297,7 +310,8 @@ class GraphedSamplerLoop:
                     f"sampler-loop hipGraph capture failed, eager: {exc!r}")
                 self._failed = True
                 return None
-            self.graphs[key] = entry
+        self.graphs[key] = entry  # re-insert = most recently used
+        _evict_lru(self.graphs)
         (s_noise, s_lat, s_ctx, s_y, s_out, g) = entry
         s_noise.copy_(noise)
         if s_lat is not None:

@@ -278,14 +278,15 @@ __global__ __launch_bounds__(NWAVES * 64) void attn_fwd_kernel(
   }
 }
 
-static torch::Tensor g_attn_zero_page;
+// intentionally leaked (see gemm.hip zero_page_ptr): exit-time tensor
+// destruction would outlive the HIP context
+static torch::Tensor* g_attn_zero_page = nullptr;
 
 static const uint16_t* attn_zero_page(const torch::Tensor& like) {
-  if (!g_attn_zero_page.defined() ||
-      g_attn_zero_page.device() != like.device())
-    g_attn_zero_page =
-        torch::zeros({64}, like.options().dtype(at::kBFloat16));
-  return (const uint16_t*)g_attn_zero_page.data_ptr();
+  if (!g_attn_zero_page || g_attn_zero_page->device() != like.device())
+    g_attn_zero_page = new torch::Tensor(
+        torch::zeros({64}, like.options().dtype(at::kBFloat16)));
+  return (const uint16_t*)g_attn_zero_page->data_ptr();
 }
 
 static torch::Tensor launch_attn_raw(const uint16_t* qp, const uint16_t* kp,

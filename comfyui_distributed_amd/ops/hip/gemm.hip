@@ -351,12 +351,15 @@ __global__ __launch_bounds__(512, 1) void gemm256_kernel(
 // launchers
 // ---------------------------------------------------------------------------
 
-static torch::Tensor g_zero_page;
+// intentionally leaked: a static torch::Tensor's destructor would run at
+// process exit AFTER the HIP context is torn down
+static torch::Tensor* g_zero_page = nullptr;
 
 static const uint16_t* zero_page_ptr(const torch::Tensor& like) {
-  if (!g_zero_page.defined() || g_zero_page.device() != like.device())
-    g_zero_page = torch::zeros({64}, like.options().dtype(at::kBFloat16));
-  return (const uint16_t*)g_zero_page.data_ptr();
+  if (!g_zero_page || g_zero_page->device() != like.device())
+    g_zero_page =
+        new torch::Tensor(torch::zeros({64}, like.options().dtype(at::kBFloat16)));
+  return (const uint16_t*)g_zero_page->data_ptr();
 }
 
 torch::Tensor gemm256_bf16(torch::Tensor x, torch::Tensor w,
