@@ -315,3 +315,16 @@ def test_sampler_and_flow_loops_abort_on_interrupt():
             sample_flow(lambda x, t: x, torch.randn(1, 4, 2, 2, 2), steps=2)
     finally:
         rt.clear_interrupt()
+
+
+def test_graph_cache_lru_eviction():
+    """hipGraph caches are bounded: oldest entry evicted past the cap."""
+    from comfyui_distributed_amd.models import sampling
+
+    graphs = {}
+    for i in range(sampling.GRAPH_CACHE_CAP + 3):
+        graphs[("key", i)] = i
+        sampling._evict_lru(graphs)
+    assert len(graphs) == sampling.GRAPH_CACHE_CAP
+    assert ("key", 0) not in graphs  # oldest gone
+    assert ("key", sampling.GRAPH_CACHE_CAP + 2) in graphs  # newest kept

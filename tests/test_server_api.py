@@ -579,3 +579,29 @@ def test_view_content_types(client, tmp_path):
             assert r.status == 200 and r.content_type == ctype, name
 
     run(loop, go())
+
+
+def test_static_js_served_and_confined(tmp_config):
+    """The panel's shared JS module is served from /static/, and path
+    escapes are rejected."""
+    import asyncio
+
+    from aiohttp.test_utils import TestClient, TestServer
+
+    from comfyui_distributed_amd.server.app import DistributedServer
+
+    async def go():
+        cl = TestClient(TestServer(DistributedServer().build_app()))
+        await cl.start_server()
+        try:
+            r = await cl.get("/static/js/distributed.js")
+            assert r.status == 200
+            body = await r.text()
+            assert "decideSubmission" in body and "valueWidgetModel" in body
+            assert r.headers["Content-Type"].startswith("application/javascript")
+            r = await cl.get("/static/../app.py")
+            assert r.status == 404
+        finally:
+            await cl.close()
+
+    asyncio.run(go())
