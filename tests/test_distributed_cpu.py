@@ -260,3 +260,48 @@ def test_wan_seed_parallel_frame_gather_world2():
     res = out[0]
     assert res["n"] == 10
     assert res["meta"] == [(r, i) for r in range(2) for i in range(5)]
+
+
+def _body_mailbox(ctx, port):
+    """ResultMailbox contract: multi-chunk streaming, interleaved drains,
+    done-marker termination, payload dtype preserved on the wire."""
+    import torch.distributed as tdist
+
+    from comfyui_distributed_amd.parallel.dist import ResultMailbox
+
+    store = tdist.TCPStore("127.0.0.1", port + 1000, ctx.world_size,
+                           ctx.is_master)
+    mb = ResultMailbox(ctx, store, "mbjob", recv_timeout=30.0)
+    if not ctx.is_master:
+        for seq in range(3):
+            tensors = [torch.full((4, 4), float(seq * 10 + i),
+                                  dtype=torch.bfloat16) for i in range(2)]
+            mb.send_chunk(tensors, [(seq, i) for i in range(2)])
+        mb.finish()
+        return None
+    got_t, got_m = [], []
+    import time as _t
+
+    deadline = _t.monotonic() + 60
+    while not mb.rank_finished(1):
+        t, m = mb.drain()
+        got_t += t
+        got_m += m
+        if _t.monotonic() > deadline:
+            break
+        _t.sleep(0.01)
+    t, m = mb.drain()
+    got_t += t
+    got_m += m
+    return {"meta": got_m,
+            "vals": [float(x[0, 0]) for x in got_t],
+            "dtypes": [str(x.dtype) for x in got_t]}
+
+
+def test_result_mailbox_streams_chunks_in_order():
+    out = _spawn("_body_mailbox", port=PORT_BASE + 10)
+    res = out[0]
+    assert res["meta"] == [(s, i) for s in range(3) for i in range(2)]
+    assert res["vals"] == [0.0, 1.0, 10.0, 11.0, 20.0, 21.0]
+    # source dtype preserved on the wire (bf16, not fp32-upcast)
+    assert all(d == "torch.bfloat16" for d in res["dtypes"])
