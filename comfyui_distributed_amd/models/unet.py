@@ -256,6 +256,7 @@ class Upsample(nn.Module):
                     or tuple(output_shape) == (2 * x.shape[2], 2 * x.shape[3]))
         if (exact_2x and x.is_cuda
                 and x.is_contiguous(memory_format=torch.channels_last)
+                and x.shape[1] % 64 == 0  # up2 needs the conv256 path
                 and ops.conv_supported(self.conv)):
             return ops.conv2d_mfma(x, self.conv, up2=True)
         if output_shape is not None:

@@ -158,6 +158,7 @@ class _DecoderUpsample(nn.Module):
 
     def forward(self, x):
         if (x.is_cuda and x.is_contiguous(memory_format=torch.channels_last)
+                and x.shape[1] % 64 == 0  # up2 needs the conv256 path
                 and ops.conv_supported(self.conv)):
             # nearest-2x fused into the conv's tap addressing: the 4x-sized
             # upsampled intermediate (the decoder's largest tensors) never
