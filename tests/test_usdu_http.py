@@ -198,3 +198,30 @@ def test_master_static_cleans_job_on_failure(monkeypatch):
             enabled_workers=[], worker_id="",
         )
     assert asyncio.run(rt.job_state.get_tile_job("doomed")) is None
+
+
+def test_requeue_scales_to_1024_tiles():
+    """SDXL-8K scale guard (BASELINE config 4 has 64 tiles; an 8K canvas
+    with 256px tiles has 1024): the HTTP requeue path is in-memory dicts,
+    so a dead worker holding hundreds of assignments requeues in
+    milliseconds, not store-RTT-bound seconds (the round-1 advisor's
+    O(total) concern on the RCCL path's assigned_incomplete)."""
+    async def go():
+        job = TileJobState(job_id="big", total_tasks=1024, batch_size=2)
+        job.worker_status["w1"] = time.time() - 120
+        for t in range(512):
+            job.assigned_to_workers[t] = "w1"
+            if t % 2 == 0:  # half its tiles fully done
+                job.completed_tasks[(t, 0)] = True
+                job.completed_tasks[(t, 1)] = True
+        t0 = time.monotonic()
+        await usdu_http.check_and_requeue_timed_out_workers(job, timeout=60)
+        elapsed = time.monotonic() - t0
+        requeued = []
+        while not job.pending_tasks.empty():
+            requeued.append(job.pending_tasks.get_nowait())
+        return requeued, elapsed
+
+    requeued, elapsed = asyncio.run(go())
+    assert sorted(requeued) == list(range(1, 512, 2))  # odd = incomplete
+    assert elapsed < 1.0, f"requeue took {elapsed:.2f}s at 1024-tile scale"
