@@ -181,3 +181,44 @@ def test_concurrent_requeue_no_duplication():
     for t in threads:
         t.join()
     assert sorted(seen) == sorted(dead + [i for i in range(100) if i not in dead])
+
+
+def test_assigned_incomplete_store_ops_bounded_at_1024_tiles():
+    """Round-1 advisor: assigned_incomplete did one store round-trip per
+    task in the JOB (1024 at 8K canvases). Now it is two list reads plus
+    one done-probe per *candidate* — count the ops to pin the contract."""
+
+    class CountingStore(LocalStore):
+        def __init__(self):
+            super().__init__()
+            self.ops = 0
+
+        def get(self, key):
+            self.ops += 1
+            return super().get(key)
+
+        def add(self, key, amount):
+            self.ops += 1
+            return super().add(key, amount)
+
+        def compare_set(self, key, expected, desired):
+            self.ops += 1
+            return super().compare_set(key, expected, desired)
+
+        def set(self, key, value):
+            self.ops += 1
+            return super().set(key, value)
+
+    store = CountingStore()
+    master = TileQueue(store, "big", 0)
+    master.init_job(1024)
+    worker = TileQueue(store, "big", 1)
+    pulled = [worker.pop() for _ in range(64)]
+    for idx in pulled[:32]:
+        worker.mark_done(idx)
+    store.ops = 0
+    incomplete = master.assigned_incomplete(1)
+    assert sorted(incomplete) == sorted(pulled[32:])
+    # 2 list reads + <= 1 done-probe per incomplete candidate (32) — far
+    # from the old 2*1024 scan
+    assert store.ops <= 2 + 40, store.ops
